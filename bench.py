@@ -1,0 +1,214 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: FLAN-T5-base fine-tune step throughput (samples/sec).
+
+Measures the reference workshop's headline workload (flan-t5-base,
+per-device batch 2, seq len 512 with max-length padding, labels = input_ids
+copy per NLP_workloads/Anyscale_job/utils.py:31, AdamW lr 2e-5 wd 0.01,
+grad-clip 1.0, bf16) on synthetic Alpaca-shaped data with random-init weights.
+
+Launched by the driver as
+  python bench.py --gpus 1 ...                                (single GPU)
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N ...           (one rank/GPU)
+
+Rank 0 prints exactly one JSON line.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from mi355.models.t5.configuration import t5_config
+from mi355.models.t5.modeling import T5ForConditionalGeneration
+from mi355.parallel import (FlatAdamW, FlatReplica, barrier, get_rank,
+                            get_world_size, init_distributed)
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--model", default="flan-t5-base")
+    ap.add_argument("--batch", type=int, default=2, help="per-device batch (reference config)")
+    ap.add_argument("--seq", type=int, default=512)
+    ap.add_argument("--device", default=None, help="cpu override for local testing")
+    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--graphs", default="auto", choices=["auto", "on", "off"],
+                    help="hipGraph-capture the train step")
+    ap.add_argument("--mode", default="train", choices=["train", "infer"])
+    return ap.parse_args()
+
+
+def build_model(args, device, dtype):
+    torch.manual_seed(1234)  # identical random init on every rank
+    cfg = t5_config(args.model, dropout_rate=0.0)  # dropout off: identical work per step
+    model = T5ForConditionalGeneration(cfg).to(device=device, dtype=dtype)
+    model.train()
+    return model
+
+
+def synthetic_batch(args, device, seed):
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    input_ids = torch.randint(5, 32000, (args.batch, args.seq), generator=g)
+    # Workshop quirk preserved for parity: labels are a copy of input_ids
+    # (Anyscale_job/utils.py:31 — the workshop trains input reconstruction).
+    labels = input_ids.clone()
+    return input_ids.to(device), labels.to(device)
+
+
+def train_step(replica, optimizer, input_ids, labels):
+    optimizer.zero_grad()
+    out = replica(input_ids=input_ids, labels=labels)
+    out.loss.backward()
+    replica.finish_grad_sync()
+    optimizer.step()
+    return out.loss
+
+
+def run_train(args, device, dtype):
+    model = build_model(args, device, dtype)
+    replica = FlatReplica(model)
+    optimizer = FlatAdamW(replica, lr=2e-5, weight_decay=0.01, max_grad_norm=1.0)
+    input_ids, labels = synthetic_batch(args, device, seed=1000 + get_rank())
+
+    use_graphs = False
+    if device.type == "cuda" and args.graphs in ("auto", "on"):
+        use_graphs = True
+
+    step_fn = lambda: train_step(replica, optimizer, input_ids, labels)
+
+    for _ in range(max(args.warmup, 1)):
+        step_fn()
+
+    graph = None
+    if use_graphs:
+        try:
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            # capture one full step (fwd+bwd+allreduce+adamw) as a hipGraph
+            with torch.cuda.graph(g):
+                step_fn()
+            graph = g
+            graph.replay()  # warm replay
+            torch.cuda.synchronize()
+        except Exception as e:
+            if args.graphs == "on":
+                raise
+            if get_rank() == 0:
+                print(f"# graph capture unavailable, eager fallback: {e!r}")
+            graph = None
+
+    def timed_step():
+        if graph is not None:
+            graph.replay()
+        else:
+            step_fn()
+
+    timed_step()  # one more warm step after potential capture
+    barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        timed_step()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    barrier()
+    elapsed = time.perf_counter() - t0
+
+    elapsed = max_over_ranks(elapsed, device)
+    return elapsed, {"graphs": graph is not None}
+
+
+def run_infer(args, device, dtype):
+    """Batch text-generation inference (BASELINE config 3): batch 256 fp16,
+    max_new_tokens 128 — seq/sec."""
+    torch.manual_seed(1234)
+    cfg = t5_config(args.model, dropout_rate=0.0)
+    model = T5ForConditionalGeneration(cfg).to(device=device, dtype=dtype).eval()
+    b = args.batch
+    g = torch.Generator(device="cpu").manual_seed(7 + get_rank())
+    input_ids = torch.randint(5, 32000, (b, args.seq), generator=g).to(device)
+
+    def step_fn():
+        model.generate(input_ids=input_ids, max_new_tokens=128, min_new_tokens=128)
+
+    for _ in range(max(args.warmup, 1)):
+        step_fn()
+    barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step_fn()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    barrier()
+    elapsed = time.perf_counter() - t0
+    return max_over_ranks(elapsed, device), {}
+
+
+def max_over_ranks(elapsed, device):
+    if get_world_size() > 1:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], device=device if device.type == "cuda" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    return elapsed
+
+
+def main():
+    args = parse_args()
+    rank = init_distributed()
+    n = get_world_size() if get_world_size() > 1 else args.gpus
+    if args.device:
+        device = torch.device(args.device)
+    elif torch.cuda.is_available():
+        device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", "0")))
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+    dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+
+    if args.mode == "train":
+        elapsed, extra = run_train(args, device, dtype)
+        metric = "train_samples_per_s"
+    else:
+        elapsed, extra = run_infer(args, device, dtype)
+        metric = "infer_seq_per_s"
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    value = args.batch * n * args.steps / elapsed
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": metric,
+            "value": round(value, 3),
+            "unit": "samples/s" if args.mode == "train" else "seq/s",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * n,
+                "seq_len": args.seq,
+                "parallelism": f"dp{n}",
+                **extra,
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

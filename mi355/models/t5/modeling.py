@@ -1,0 +1,414 @@
+"""MI355X-native T5 / FLAN-T5 encoder-decoder.
+
+Re-implements the model family the reference workshop fine-tunes and serves
+(``T5ForConditionalGeneration`` — reference call sites
+Model_finetuning_and_batch_inference.ipynb:cell-25/cell-46/cell-49,
+NLP_workloads/Anyscale_job/predictor.py:102) from scratch on top of
+``mi355.ops``: every hot op (RMSNorm, fused attention with T5's additive
+relative-position bias, gated-GELU, vocab cross-entropy) dispatches to
+hand-written HIP/CDNA4 kernels on GPU; plain GEMMs go through torch's
+hipBLASLt path. Parameter names and the ``save_pretrained`` checkpoint layout
+match the HF T5 convention so checkpoints are drop-in.
+"""
+from __future__ import annotations
+
+import math
+import os
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ... import ops
+from .configuration import T5Config, t5_config
+
+NEG_INF = -1e9
+
+
+class T5LayerNorm(nn.Module):
+    def __init__(self, d: int, eps: float):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(d))
+        self.variance_epsilon = eps
+
+    def forward(self, x):
+        return ops.rms_norm(x, self.weight, self.variance_epsilon)
+
+
+def relative_position_bucket(rel_pos: torch.Tensor, bidirectional: bool,
+                             num_buckets: int, max_distance: int) -> torch.Tensor:
+    """T5 bucketing of (key_pos - query_pos) — SURVEY.md K4."""
+    ret = torch.zeros_like(rel_pos)
+    n = -rel_pos
+    if bidirectional:
+        num_buckets //= 2
+        ret = ret + (n < 0).long() * num_buckets
+        n = n.abs()
+    else:
+        n = torch.clamp(n, min=0)
+    max_exact = num_buckets // 2
+    is_small = n < max_exact
+    val_large = max_exact + (
+        torch.log(n.float() / max_exact + 1e-20) / math.log(max_distance / max_exact)
+        * (num_buckets - max_exact)
+    ).long()
+    val_large = torch.clamp(val_large, max=num_buckets - 1)
+    return ret + torch.where(is_small, n, val_large)
+
+
+class T5Attention(nn.Module):
+    def __init__(self, config: T5Config, has_relative_attention_bias: bool, is_decoder: bool):
+        super().__init__()
+        self.is_decoder = is_decoder
+        self.has_relative_attention_bias = has_relative_attention_bias
+        self.num_buckets = config.relative_attention_num_buckets
+        self.max_distance = config.relative_attention_max_distance
+        self.d_model = config.d_model
+        self.d_kv = config.d_kv
+        self.n_heads = config.num_heads
+        self.inner_dim = self.n_heads * self.d_kv
+        self.q = nn.Linear(self.d_model, self.inner_dim, bias=False)
+        self.k = nn.Linear(self.d_model, self.inner_dim, bias=False)
+        self.v = nn.Linear(self.d_model, self.inner_dim, bias=False)
+        self.o = nn.Linear(self.inner_dim, self.d_model, bias=False)
+        if has_relative_attention_bias:
+            self.relative_attention_bias = nn.Embedding(self.num_buckets, self.n_heads)
+        self.dropout = config.dropout_rate
+
+    def compute_bias(self, sq: int, sk: int, device) -> torch.Tensor:
+        """(1, H, sq, sk) additive relative-position bias."""
+        cpos = torch.arange(sq, dtype=torch.long, device=device)[:, None]
+        mpos = torch.arange(sk, dtype=torch.long, device=device)[None, :]
+        buckets = relative_position_bucket(
+            mpos - cpos, bidirectional=not self.is_decoder,
+            num_buckets=self.num_buckets, max_distance=self.max_distance,
+        )
+        vals = self.relative_attention_bias(buckets)  # (sq, sk, H)
+        return vals.permute(2, 0, 1).unsqueeze(0)
+
+    def _shape(self, x, b):
+        return x.view(b, -1, self.n_heads, self.d_kv).transpose(1, 2)
+
+    def forward(self, hidden, kv_hidden=None, position_bias=None, kpad=None,
+                causal=False, kv_cache=None):
+        """kv_cache: optional dict used by generate() —
+        {"k","v": (B,H,Tmax,D) preallocated, "len": int} for self-attn append,
+        or {"k","v" precomputed static} for cross-attn."""
+        b, sq, _ = hidden.shape
+        qs = self._shape(self.q(hidden), b)
+        if kv_cache is not None and kv_cache.get("static", False):
+            ks, vs = kv_cache["k"], kv_cache["v"]
+        else:
+            src = hidden if kv_hidden is None else kv_hidden
+            ks = self._shape(self.k(src), b)
+            vs = self._shape(self.v(src), b)
+            if kv_cache is not None:
+                t0 = kv_cache["len"]
+                kv_cache["k"][:, :, t0:t0 + sq] = ks
+                kv_cache["v"][:, :, t0:t0 + sq] = vs
+                kv_cache["len"] = t0 + sq
+                ks = kv_cache["k"][:, :, : t0 + sq]
+                vs = kv_cache["v"][:, :, : t0 + sq]
+        if kv_cache is not None and sq == 1 and not kv_cache.get("static", False):
+            out = ops.decode_attention(qs, kv_cache["k"], kv_cache["v"],
+                                       kv_cache["len"], bias=position_bias)
+        else:
+            out = ops.attention(qs, ks, vs, bias=position_bias, causal=causal, kpad=kpad)
+        out = out.transpose(1, 2).reshape(b, sq, self.inner_dim)
+        return self.o(out)
+
+
+class T5LayerSelfAttention(nn.Module):
+    def __init__(self, config, has_relative_attention_bias, is_decoder):
+        super().__init__()
+        self.SelfAttention = T5Attention(config, has_relative_attention_bias, is_decoder)
+        self.layer_norm = T5LayerNorm(config.d_model, config.layer_norm_epsilon)
+        self.dropout = nn.Dropout(config.dropout_rate)
+
+    def forward(self, hidden, position_bias=None, kpad=None, causal=False, kv_cache=None):
+        normed = self.layer_norm(hidden)
+        att = self.SelfAttention(normed, position_bias=position_bias, kpad=kpad,
+                                 causal=causal, kv_cache=kv_cache)
+        return hidden + self.dropout(att)
+
+
+class T5LayerCrossAttention(nn.Module):
+    def __init__(self, config):
+        super().__init__()
+        self.EncDecAttention = T5Attention(config, False, is_decoder=True)
+        self.layer_norm = T5LayerNorm(config.d_model, config.layer_norm_epsilon)
+        self.dropout = nn.Dropout(config.dropout_rate)
+
+    def forward(self, hidden, enc_hidden, kpad=None, kv_cache=None):
+        normed = self.layer_norm(hidden)
+        att = self.EncDecAttention(normed, kv_hidden=enc_hidden, kpad=kpad, kv_cache=kv_cache)
+        return hidden + self.dropout(att)
+
+
+class T5DenseGatedActDense(nn.Module):
+    def __init__(self, config):
+        super().__init__()
+        self.wi_0 = nn.Linear(config.d_model, config.d_ff, bias=False)
+        self.wi_1 = nn.Linear(config.d_model, config.d_ff, bias=False)
+        self.wo = nn.Linear(config.d_ff, config.d_model, bias=False)
+        self.dropout = nn.Dropout(config.dropout_rate)
+
+    def forward(self, x):
+        h = ops.geglu(self.wi_0(x), self.wi_1(x))
+        return self.wo(self.dropout(h))
+
+
+class T5LayerFF(nn.Module):
+    def __init__(self, config):
+        super().__init__()
+        self.DenseReluDense = T5DenseGatedActDense(config)
+        self.layer_norm = T5LayerNorm(config.d_model, config.layer_norm_epsilon)
+        self.dropout = nn.Dropout(config.dropout_rate)
+
+    def forward(self, hidden):
+        return hidden + self.dropout(self.DenseReluDense(self.layer_norm(hidden)))
+
+
+class T5Block(nn.Module):
+    def __init__(self, config, has_relative_attention_bias, is_decoder):
+        super().__init__()
+        self.is_decoder = is_decoder
+        self.layer = nn.ModuleList()
+        self.layer.append(T5LayerSelfAttention(config, has_relative_attention_bias, is_decoder))
+        if is_decoder:
+            self.layer.append(T5LayerCrossAttention(config))
+        self.layer.append(T5LayerFF(config))
+
+    def forward(self, hidden, position_bias=None, kpad=None, enc_hidden=None,
+                enc_kpad=None, causal=False, self_cache=None, cross_cache=None):
+        hidden = self.layer[0](hidden, position_bias=position_bias, kpad=kpad,
+                               causal=causal, kv_cache=self_cache)
+        if self.is_decoder:
+            hidden = self.layer[1](hidden, enc_hidden, kpad=enc_kpad, kv_cache=cross_cache)
+        hidden = self.layer[-1](hidden)
+        return hidden
+
+
+class T5Stack(nn.Module):
+    def __init__(self, config: T5Config, embed_tokens: nn.Embedding, is_decoder: bool):
+        super().__init__()
+        self.config = config
+        self.is_decoder = is_decoder
+        self.embed_tokens = embed_tokens
+        n = config.num_decoder_layers if is_decoder else config.num_layers
+        self.block = nn.ModuleList(
+            [T5Block(config, has_relative_attention_bias=(i == 0), is_decoder=is_decoder)
+             for i in range(n)]
+        )
+        self.final_layer_norm = T5LayerNorm(config.d_model, config.layer_norm_epsilon)
+        self.dropout = nn.Dropout(config.dropout_rate)
+
+    def forward(self, input_ids=None, inputs_embeds=None, kpad=None, enc_hidden=None,
+                enc_kpad=None, caches=None, past_len: int = 0):
+        x = self.embed_tokens(input_ids) if inputs_embeds is None else inputs_embeds
+        x = self.dropout(x)
+        sq = x.shape[1]
+        sk = past_len + sq
+        # The relative-position bias is computed once by block 0 and shared by
+        # every layer (T5 semantics).
+        attn0 = self.block[0].layer[0].SelfAttention
+        position_bias = attn0.compute_bias(sk, sk, x.device)[:, :, past_len:].to(x.dtype) \
+            if attn0.has_relative_attention_bias else None
+        causal = self.is_decoder and sq > 1
+        for i, blk in enumerate(self.block):
+            self_cache = caches[i]["self"] if caches is not None else None
+            cross_cache = caches[i]["cross"] if (caches is not None and self.is_decoder) else None
+            x = blk(x, position_bias=position_bias, kpad=kpad, enc_hidden=enc_hidden,
+                    enc_kpad=enc_kpad, causal=causal, self_cache=self_cache,
+                    cross_cache=cross_cache)
+        return self.dropout(self.final_layer_norm(x))
+
+
+@dataclass
+class Seq2SeqOutput:
+    loss: torch.Tensor | None = None
+    logits: torch.Tensor | None = None
+    encoder_last_hidden_state: torch.Tensor | None = None
+
+    def __getitem__(self, k):
+        return getattr(self, k)
+
+
+def _kpad_from_mask(attention_mask: torch.Tensor | None):
+    """(B,S) 1/0 mask -> additive fp32 (B,S) or None when nothing is padded."""
+    if attention_mask is None:
+        return None
+    if bool(attention_mask.all()):
+        return None
+    return (1.0 - attention_mask.float()) * NEG_INF
+
+
+class T5ForConditionalGeneration(nn.Module):
+    config_class = T5Config
+
+    def __init__(self, config: T5Config):
+        super().__init__()
+        self.config = config
+        self.model_dim = config.d_model
+        self.shared = nn.Embedding(config.vocab_size, config.d_model)
+        self.encoder = T5Stack(config, self.shared, is_decoder=False)
+        self.decoder = T5Stack(config, self.shared, is_decoder=True)
+        self.lm_head = nn.Linear(config.d_model, config.vocab_size, bias=False)
+        self.init_weights()
+        if config.tie_word_embeddings:
+            self.lm_head.weight = self.shared.weight
+
+    # -- init (T5 scheme, scaled by initializer_factor) --------------------- #
+    def init_weights(self):
+        f = self.config.initializer_factor
+        d_model, d_kv, d_ff = self.config.d_model, self.config.d_kv, self.config.d_ff
+        n_heads = self.config.num_heads
+        self.shared.weight.data.normal_(0.0, f * 1.0)
+        self.lm_head.weight.data.normal_(0.0, f * d_model**-0.5)
+        for m in self.modules():
+            if isinstance(m, T5LayerNorm):
+                m.weight.data.fill_(f * 1.0)
+            elif isinstance(m, T5Attention):
+                m.q.weight.data.normal_(0.0, f * (d_model * d_kv) ** -0.5)
+                m.k.weight.data.normal_(0.0, f * d_model**-0.5)
+                m.v.weight.data.normal_(0.0, f * d_model**-0.5)
+                m.o.weight.data.normal_(0.0, f * (n_heads * d_kv) ** -0.5)
+                if m.has_relative_attention_bias:
+                    m.relative_attention_bias.weight.data.normal_(0.0, f * d_model**-0.5)
+            elif isinstance(m, T5DenseGatedActDense):
+                m.wi_0.weight.data.normal_(0.0, f * d_model**-0.5)
+                m.wi_1.weight.data.normal_(0.0, f * d_model**-0.5)
+                m.wo.weight.data.normal_(0.0, f * d_ff**-0.5)
+
+    # -- training / scoring forward ----------------------------------------- #
+    def _shift_right(self, labels):
+        start = self.config.decoder_start_token_id
+        shifted = labels.new_zeros(labels.shape)
+        shifted[:, 1:] = labels[:, :-1].clone()
+        shifted[:, 0] = start
+        shifted.masked_fill_(shifted == -100, self.config.pad_token_id)
+        return shifted
+
+    @property
+    def device(self):
+        return self.shared.weight.device
+
+    def forward(self, input_ids=None, attention_mask=None, decoder_input_ids=None,
+                decoder_attention_mask=None, labels=None, encoder_outputs=None):
+        kpad = _kpad_from_mask(attention_mask)
+        if encoder_outputs is None:
+            encoder_outputs = self.encoder(input_ids=input_ids, kpad=kpad)
+        if decoder_input_ids is None and labels is not None:
+            decoder_input_ids = self._shift_right(labels)
+        dec = self.decoder(input_ids=decoder_input_ids, kpad=_kpad_from_mask(decoder_attention_mask),
+                           enc_hidden=encoder_outputs, enc_kpad=kpad)
+        if self.config.tie_word_embeddings:
+            dec = dec * (self.model_dim**-0.5)
+        logits = self.lm_head(dec)
+        loss = None
+        if labels is not None:
+            loss = ops.cross_entropy(logits, labels, ignore_index=-100)
+        return Seq2SeqOutput(loss=loss, logits=logits, encoder_last_hidden_state=encoder_outputs)
+
+    # -- greedy generate with preallocated KV cache -------- #
+    @torch.no_grad()
+    def generate(self, input_ids=None, attention_mask=None, max_new_tokens: int = 128,
+                 min_new_tokens: int = 0, eos_token_id=None, **_ignored):
+        cfg = self.config
+        eos = cfg.eos_token_id if eos_token_id is None else eos_token_id
+        b = input_ids.shape[0]
+        dev = input_ids.device
+        dt = self.shared.weight.dtype
+        kpad = _kpad_from_mask(attention_mask)
+        enc = self.encoder(input_ids=input_ids, kpad=kpad)
+        h, dkv = cfg.num_heads, cfg.d_kv
+        caches = []
+        for blk in self.decoder.block:
+            att = blk.layer[1].EncDecAttention
+            ck = att._shape(att.k(enc), b)
+            cv = att._shape(att.v(enc), b)
+            caches.append({
+                "self": {"k": torch.zeros(b, h, max_new_tokens, dkv, device=dev, dtype=dt),
+                         "v": torch.zeros(b, h, max_new_tokens, dkv, device=dev, dtype=dt),
+                         "len": 0},
+                "cross": {"k": ck, "v": cv, "static": True},
+            })
+        tokens = torch.full((b, 1), cfg.decoder_start_token_id, device=dev, dtype=torch.long)
+        out_tokens = []
+        finished = torch.zeros(b, dtype=torch.bool, device=dev)
+        attn0 = self.decoder.block[0].layer[0].SelfAttention
+        full_bias = attn0.compute_bias(max_new_tokens, max_new_tokens, dev).to(dt)
+        for t in range(max_new_tokens):
+            bias_t = full_bias[:, :, t : t + 1, : t + 1]
+            dec = self._decode_step(tokens, enc, kpad, caches, t, bias_t)
+            if cfg.tie_word_embeddings:
+                dec = dec * (self.model_dim**-0.5)
+            logits = self.lm_head(dec[:, -1])
+            if t < min_new_tokens:
+                logits[:, eos] = float("-inf")
+            nxt = logits.argmax(dim=-1)
+            nxt = torch.where(finished, torch.full_like(nxt, cfg.pad_token_id), nxt)
+            out_tokens.append(nxt)
+            finished |= nxt == eos
+            if bool(finished.all()):
+                break
+            tokens = nxt.unsqueeze(1)
+        return torch.stack(out_tokens, dim=1)
+
+    def _decode_step(self, tokens, enc, kpad, caches, t, bias_t):
+        x = self.decoder.dropout(self.decoder.embed_tokens(tokens))
+        for i, blk in enumerate(self.decoder.block):
+            x = blk(x, position_bias=bias_t, enc_hidden=enc, enc_kpad=kpad, causal=False,
+                    self_cache=caches[i]["self"], cross_cache=caches[i]["cross"])
+        return self.decoder.final_layer_norm(x)
+
+    # -- HF-layout checkpoint I/O --------------------------- #
+    def save_pretrained(self, save_dir: str, safe_serialization: bool = True):
+        os.makedirs(save_dir, exist_ok=True)
+        self.config.save_pretrained(save_dir)
+        state = {k: v for k, v in self.state_dict().items()}
+        # encoder/decoder embed_tokens alias the shared embedding (HF layout
+        # stores only "shared.weight"); a tied lm_head is also dropped.
+        state.pop("encoder.embed_tokens.weight", None)
+        state.pop("decoder.embed_tokens.weight", None)
+        if self.config.tie_word_embeddings:
+            state.pop("lm_head.weight", None)
+        if safe_serialization:
+            from safetensors.torch import save_file
+
+            save_file({k: v.contiguous().cpu() for k, v in state.items()},
+                      os.path.join(save_dir, "model.safetensors"),
+                      metadata={"format": "pt"})
+        else:
+            torch.save(state, os.path.join(save_dir, "pytorch_model.bin"))
+
+    @classmethod
+    def from_pretrained(cls, path: str, torch_dtype=None, config: T5Config | None = None,
+                        **_ignored):
+        if config is None and os.path.isdir(path):
+            config = T5Config.from_pretrained(path)
+        elif config is None:
+            config = t5_config(path)
+        model = cls(config)
+        st_file = os.path.join(path, "model.safetensors") if os.path.isdir(path) else None
+        if st_file and os.path.exists(st_file):
+            from safetensors.torch import load_file
+
+            state = load_file(st_file)
+            missing, unexpected = model.load_state_dict(state, strict=False)
+            aliases = {"encoder.embed_tokens.weight", "decoder.embed_tokens.weight"}
+            if model.config.tie_word_embeddings:
+                aliases.add("lm_head.weight")
+            missing = [k for k in missing if k not in aliases]
+            if missing:
+                raise RuntimeError(f"missing keys in checkpoint: {missing}")
+            if model.config.tie_word_embeddings:
+                model.lm_head.weight = model.shared.weight
+        elif os.path.isdir(path) and os.path.exists(os.path.join(path, "pytorch_model.bin")):
+            state = torch.load(os.path.join(path, "pytorch_model.bin"), map_location="cpu",
+                               weights_only=True)
+            model.load_state_dict(state, strict=False)
+        if torch_dtype is not None:
+            model = model.to(torch_dtype)
+        return model

@@ -1,0 +1,235 @@
+"""mi355.ops — dispatch layer between hand-written HIP/CDNA4 kernels and the
+plain-PyTorch reference implementations.
+
+Policy:
+  * on a ROCm GPU tensor the HIP extension (``mi355._C``, built in-tree for
+    gfx950) is the ONLY path — if it is missing we raise instead of silently
+    falling back to eager PyTorch;
+  * on CPU tensors we run ``torch_ref`` (differentiable plain torch), which is
+    also the numerics reference the kernels are tested against;
+  * ``MI355_FORCE_REF=1`` forces the reference path everywhere (used by the
+    numerics tests to produce same-device baselines).
+"""
+from __future__ import annotations
+
+import os
+
+import torch
+
+from . import torch_ref
+
+_EXT = None
+_EXT_ERR: Exception | None = None
+
+
+def _load_ext():
+    global _EXT, _EXT_ERR
+    if _EXT is not None or _EXT_ERR is not None:
+        return _EXT
+    try:
+        from mi355 import _C  # built by tools/build_ext.py (hipcc, gfx950)
+
+        _EXT = _C
+    except Exception as e:  # pragma: no cover - exercised only on GPU boxes
+        _EXT_ERR = e
+    return _EXT
+
+
+def hip_available() -> bool:
+    return torch.cuda.is_available() and _load_ext() is not None
+
+
+def _use_hip(*tensors: torch.Tensor) -> bool:
+    if os.environ.get("MI355_FORCE_REF") == "1":
+        return False
+    if not tensors or not tensors[0].is_cuda:
+        return False
+    ext = _load_ext()
+    if ext is None:
+        raise RuntimeError(
+            "mi355: tensor is on GPU but the HIP extension mi355._C is not "
+            "importable (build it with `python tools/build_ext.py`). Refusing "
+            f"to fall back to eager PyTorch on the hot path. Cause: {_EXT_ERR!r}"
+        )
+    return True
+
+
+# --------------------------------------------------------------------------- #
+# RMSNorm (T5LayerNorm)                                                       #
+# --------------------------------------------------------------------------- #
+class _RMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        ext = _load_ext()
+        x2d = x.contiguous().view(-1, x.shape[-1])
+        y, inv_rms = ext.rmsnorm_fwd(x2d, weight, eps)
+        ctx.save_for_backward(x2d, weight, inv_rms)
+        return y.view_as(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2d, weight, inv_rms = ctx.saved_tensors
+        dx, dw = _load_ext().rmsnorm_bwd(dy.contiguous().view_as(x2d), x2d, weight, inv_rms)
+        return dx.view_as(dy), dw.to(weight.dtype), None
+
+
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6) -> torch.Tensor:
+    if _use_hip(x):
+        return _RMSNorm.apply(x, weight, eps)
+    return torch_ref.rms_norm(x, weight, eps)
+
+
+# --------------------------------------------------------------------------- #
+# Gated-GELU elementwise combine                                              #
+# --------------------------------------------------------------------------- #
+class _GeGLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gate, up):
+        gate = gate.contiguous()
+        up = up.contiguous()
+        out = _load_ext().geglu_fwd(gate, up)
+        ctx.save_for_backward(gate, up)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        gate, up = ctx.saved_tensors
+        dgate, dup = _load_ext().geglu_bwd(dout.contiguous(), gate, up)
+        return dgate, dup
+
+
+def geglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    if _use_hip(gate):
+        return _GeGLU.apply(gate, up)
+    return torch_ref.geglu(gate, up)
+
+
+# --------------------------------------------------------------------------- #
+# Fused attention with additive bias (T5 relative-position bias / masks)      #
+# --------------------------------------------------------------------------- #
+class _Attention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, bias, causal, kpad):
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        bias_c = bias.contiguous() if bias is not None else None
+        kpad_c = kpad.contiguous().float() if kpad is not None else None
+        o, lse = _load_ext().attn_fwd(q, k, v, bias_c, kpad_c, causal)
+        ctx.save_for_backward(q, k, v, bias_c, kpad_c, o, lse)
+        ctx.causal = causal
+        ctx.bias_requires_grad = bias is not None and bias.requires_grad
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, bias, kpad, o, lse = ctx.saved_tensors
+        do = do.contiguous()
+        ext = _load_ext()
+        if hasattr(ext, "attn_bwd"):
+            dq, dk, dv, dbias = ext.attn_bwd(
+                do, q, k, v, o, lse, bias, kpad, ctx.causal, ctx.bias_requires_grad
+            )
+        else:  # composed bwd via recompute (hipBLASLt batched GEMMs)
+            dq, dk, dv, dbias = _attn_bwd_composed(
+                do, q, k, v, lse, bias, kpad, ctx.causal, ctx.bias_requires_grad
+            )
+        if dbias is not None and bias is not None and dbias.shape != bias.shape:
+            # reduce over broadcast dims (bias is usually (1,H,Sq,Sk))
+            for d in range(dbias.dim()):
+                if bias.shape[d] == 1 and dbias.shape[d] != 1:
+                    dbias = dbias.sum(dim=d, keepdim=True)
+        return dq, dk, dv, dbias, None, None
+
+
+def _attn_bwd_composed(do, q, k, v, lse, bias, kpad, causal, need_dbias):
+    """Flash backward composed from batched GEMMs + the saved LSE."""
+    qf, kf, vf, dof = q.float(), k.float(), v.float(), do.float()
+    s = torch.matmul(qf, kf.transpose(-1, -2))
+    if bias is not None:
+        s = s + bias.float()
+    if kpad is not None:
+        s = s + kpad.view(kpad.shape[0], 1, 1, kpad.shape[-1])
+    if causal:
+        sq, sk = s.shape[-2], s.shape[-1]
+        iq = torch.arange(sq, device=s.device).unsqueeze(-1) + (sk - sq)
+        ik = torch.arange(sk, device=s.device).unsqueeze(0)
+        s = s.masked_fill(ik > iq, float("-inf"))
+    p = torch.exp(s - lse.unsqueeze(-1))
+    dv = torch.matmul(p.transpose(-1, -2), dof)
+    dp = torch.matmul(dof, vf.transpose(-1, -2))
+    delta = (dof * torch.matmul(p, vf)).sum(-1, keepdim=True)
+    ds = p * (dp - delta)
+    dq = torch.matmul(ds, kf)
+    dk = torch.matmul(ds.transpose(-1, -2), qf)
+    dbias = ds if need_dbias else None
+    return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), dbias
+
+
+def attention(q, k, v, bias=None, causal=False, scale: float = 1.0, kpad=None):
+    """kpad: optional (B, Sk) ADDITIVE fp32 key-padding mask (0 keep / -1e9
+    drop), kept separate from ``bias`` so a (1,H,Sq,Sk) relative-position bias
+    never has to be materialized per batch element."""
+    if _use_hip(q) and scale == 1.0:
+        return _Attention.apply(q, k, v, bias, causal, kpad)
+    return torch_ref.attention(q, k, v, bias=bias, causal=causal, scale=scale, kpad=kpad)
+
+
+# --------------------------------------------------------------------------- #
+# Vocab cross-entropy (fused log-softmax + NLL, mean over non-ignored)        #
+# --------------------------------------------------------------------------- #
+class _CrossEntropy(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, labels, ignore_index):
+        logits2d = logits.contiguous().view(-1, logits.shape[-1])
+        labels1d = labels.contiguous().view(-1)
+        loss_sum, lse, nvalid = _load_ext().ce_fwd(logits2d, labels1d, ignore_index)
+        ctx.save_for_backward(logits2d, labels1d, lse)
+        ctx.ignore_index = ignore_index
+        ctx.nvalid = max(int(nvalid), 1)
+        return loss_sum / ctx.nvalid
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits2d, labels1d, lse = ctx.saved_tensors
+        dlogits = _load_ext().ce_bwd(
+            logits2d, labels1d, lse, ctx.ignore_index, float(dloss) / ctx.nvalid
+        )
+        return dlogits.view(logits2d.shape), None, None
+
+
+def cross_entropy(logits, labels, ignore_index: int = -100):
+    if _use_hip(logits):
+        return _CrossEntropy.apply(logits, labels, ignore_index)
+    return torch_ref.cross_entropy(logits, labels, ignore_index)
+
+
+# --------------------------------------------------------------------------- #
+# Fused multi-tensor AdamW                                                    #
+# --------------------------------------------------------------------------- #
+def adamw_step(params_f32, grads, exp_avgs, exp_avg_sqs, *, lr, beta1, beta2,
+               eps, weight_decay, step, grad_scale: float = 1.0, outs_lowp=None):
+    if params_f32 and _use_hip(params_f32[0]):
+        _load_ext().adamw_multi(
+            params_f32, grads, exp_avgs, exp_avg_sqs,
+            outs_lowp if outs_lowp is not None else [],
+            lr, beta1, beta2, eps, weight_decay, step, grad_scale,
+        )
+        return
+    torch_ref.adamw_step(
+        params_f32, grads, exp_avgs, exp_avg_sqs, lr=lr, beta1=beta1,
+        beta2=beta2, eps=eps, weight_decay=weight_decay, step=step,
+        grad_scale=grad_scale, outs_lowp=outs_lowp,
+    )
+
+
+# --------------------------------------------------------------------------- #
+# Single-token decode attention over a KV cache (generate hot loop)           #
+# --------------------------------------------------------------------------- #
+def decode_attention(q, k_cache, v_cache, cache_len: int, bias=None):
+    """q: (B,H,1,D); caches: (B,H,Tmax,D) with the first ``cache_len`` rows
+    valid; bias broadcastable to (B,H,1,cache_len). SURVEY.md K11."""
+    if _use_hip(q):
+        return _load_ext().decode_attn(q, k_cache, v_cache, cache_len,
+                                       bias.contiguous() if bias is not None else None)
+    k = k_cache[:, :, :cache_len]
+    v = v_cache[:, :, :cache_len]
+    return torch_ref.attention(q, k, v, bias=bias, causal=False)
