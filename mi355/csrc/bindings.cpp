@@ -1,0 +1,34 @@
+// Python bindings for the mi355 CDNA4 kernel library (built by
+// tools/build_ext.py with hipcc --offload-arch=gfx950).
+#include <torch/extension.h>
+
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps);
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
+                                       torch::Tensor inv);
+torch::Tensor geglu_fwd(torch::Tensor gate, torch::Tensor up);
+std::vector<torch::Tensor> geglu_bwd(torch::Tensor dout, torch::Tensor gate, torch::Tensor up);
+std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels, long ignore_index);
+torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels, torch::Tensor lse,
+                     long ignore_index, double scale);
+void adamw_multi(std::vector<torch::Tensor> masters, std::vector<torch::Tensor> grads,
+                 std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
+                 std::vector<torch::Tensor> lowps, double lr, double beta1, double beta2,
+                 double eps, double weight_decay, long step, torch::Tensor grad_scale);
+torch::Tensor decode_attn(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
+                          long cache_len, c10::optional<torch::Tensor> bias);
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                    c10::optional<torch::Tensor> bias,
+                                    c10::optional<torch::Tensor> kpad, bool causal);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("geglu_fwd", &geglu_fwd);
+  m.def("geglu_bwd", &geglu_bwd);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd", &ce_bwd);
+  m.def("adamw_multi", &adamw_multi);
+  m.def("decode_attn", &decode_attn);
+  m.def("attn_fwd", &attn_fwd);
+  m.attr("__gfx__") = "gfx950";
+}

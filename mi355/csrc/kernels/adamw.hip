@@ -1,0 +1,84 @@
+// Fused AdamW with fp32 master weights + low-precision weight write-back
+// (SURVEY.md K10). With the FlatReplica layout the whole model is 1-2 flat
+// tensors, so a train step's optimizer is 1-2 launches of this kernel.
+// grad_scale (gradient clipping) can come from device memory (0-dim tensor)
+// so clipping never forces a host sync — keeps the step hipGraph-capturable.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "../common.h"
+
+namespace {
+
+template <typename G, typename P>
+__global__ void adamw_kernel(float* __restrict__ master, const G* __restrict__ grad,
+                             float* __restrict__ m, float* __restrict__ v,
+                             P* __restrict__ lowp, long n, float lr, float beta1,
+                             float beta2, float eps, float wd, float bc1, float bc2,
+                             float grad_scale, const float* __restrict__ grad_scale_dev) {
+  const float gs = grad_scale_dev ? fminf(*grad_scale_dev, 1.0f) : grad_scale;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float g = DT<G>::to_f32(grad[i]) * gs;
+    float mi = beta1 * m[i] + (1.f - beta1) * g;
+    float vi = beta2 * v[i] + (1.f - beta2) * g * g;
+    m[i] = mi;
+    v[i] = vi;
+    float denom = sqrtf(vi / bc2) + eps;
+    float p = master[i] * (1.f - lr * wd);
+    p -= lr * (mi / bc1) / denom;
+    master[i] = p;
+    if (lowp) lowp[i] = DT<P>::from_f32(p);
+  }
+}
+
+template <typename G, typename P>
+void launch(torch::Tensor& master, const torch::Tensor& grad, torch::Tensor& m,
+            torch::Tensor& v, torch::Tensor* lowp, float lr, float b1, float b2,
+            float eps, float wd, int step, float gscale,
+            const torch::Tensor* gscale_dev, hipStream_t stream) {
+  long n = master.numel();
+  float bc1 = 1.f - powf(b1, (float)step);
+  float bc2 = 1.f - powf(b2, (float)step);
+  long blocks = std::min<long>((n + 255) / 256, 4096);
+  hipLaunchKernelGGL((adamw_kernel<G, P>), dim3(blocks), dim3(256), 0, stream,
+                     master.data_ptr<float>(), (const G*)grad.data_ptr(),
+                     m.data_ptr<float>(), v.data_ptr<float>(),
+                     lowp ? (P*)lowp->data_ptr() : nullptr, n, lr, b1, b2, eps, wd,
+                     bc1, bc2, gscale,
+                     gscale_dev ? gscale_dev->data_ptr<float>() : nullptr);
+  HIP_CHECK_KERNEL();
+}
+
+}  // namespace
+
+void adamw_multi(std::vector<torch::Tensor> masters, std::vector<torch::Tensor> grads,
+                 std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
+                 std::vector<torch::Tensor> lowps, double lr, double beta1, double beta2,
+                 double eps, double weight_decay, long step, torch::Tensor grad_scale) {
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  // grad_scale: 0-dim fp32 tensor on device (no sync) or on CPU (constant)
+  float gs_const = 1.0f;
+  const torch::Tensor* gs_dev = nullptr;
+  if (grad_scale.defined() && grad_scale.numel() == 1) {
+    if (grad_scale.is_cuda()) gs_dev = &grad_scale;
+    else gs_const = grad_scale.item<float>();
+  }
+  for (size_t t = 0; t < masters.size(); ++t) {
+    TORCH_CHECK(masters[t].scalar_type() == torch::kFloat, "master must be fp32");
+    torch::Tensor* lowp = (t < lowps.size() && lowps[t].defined()) ? &lowps[t] : nullptr;
+    auto gt = grads[t].scalar_type();
+    auto pt = lowp ? lowp->scalar_type() : torch::kFloat;
+#define DISPATCH(GT, PT)                                                            \
+  launch<GT, PT>(masters[t], grads[t], ms[t], vs[t], lowp, (float)lr, (float)beta1, \
+                 (float)beta2, (float)eps, (float)weight_decay, (int)step,          \
+                 gs_const, gs_dev, stream)
+    if (gt == torch::kBFloat16 && pt == torch::kBFloat16) DISPATCH(__hip_bfloat16, __hip_bfloat16);
+    else if (gt == torch::kFloat && pt == torch::kFloat) DISPATCH(float, float);
+    else if (gt == torch::kFloat && pt == torch::kBFloat16) DISPATCH(float, __hip_bfloat16);
+    else if (gt == torch::kHalf && pt == torch::kHalf) DISPATCH(__half, __half);
+    else TORCH_CHECK(false, "adamw: unsupported grad/param dtype combo");
+#undef DISPATCH
+  }
+}

@@ -1,0 +1,161 @@
+// T5LayerNorm (RMSNorm, no mean-subtract, no bias) forward + backward.
+// Reference op: mi355/ops/torch_ref.py::rms_norm.
+// Memory-bound: vectorized 2-byte loads (8 elems = 16 B per lane).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "../common.h"
+
+namespace {
+
+// one block per row; D up to 8192 with 256 threads x 8-vec grid-stride
+template <typename T, int VEC>
+__global__ void rmsnorm_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                                   T* __restrict__ y, float* __restrict__ inv_rms,
+                                   int rows, int D, float eps) {
+  __shared__ float red[16];
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* xr = x + (long)row * D;
+    T* yr = y + (long)row * D;
+    float ss = 0.f;
+    for (int i = threadIdx.x * VEC; i < D; i += blockDim.x * VEC) {
+      VecT<T, VEC> v = vload<T, VEC>(xr + i);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float f = DT<T>::to_f32(v.data[j]);
+        ss += f * f;
+      }
+    }
+    ss = block_reduce_sum(ss, red);
+    float inv = rsqrtf(ss / D + eps);
+    if (threadIdx.x == 0) inv_rms[row] = inv;
+    for (int i = threadIdx.x * VEC; i < D; i += blockDim.x * VEC) {
+      VecT<T, VEC> v = vload<T, VEC>(xr + i);
+      VecT<T, VEC> wv = vload<T, VEC>(w + i);
+      VecT<T, VEC> o;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        o.data[j] = DT<T>::from_f32(DT<T>::to_f32(v.data[j]) * inv *
+                                    DT<T>::to_f32(wv.data[j]));
+      vstore<T, VEC>(yr + i, o);
+    }
+    __syncthreads();
+  }
+}
+
+// dx = inv * w * dy - x * inv^3/D * sum(dy*w*x) ; dw += sum_rows dy * x * inv
+// Each block owns a strided set of rows; per-thread dw partials for its fixed
+// columns, one atomicAdd per column per block at the end.
+template <typename T, int VEC>
+__global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
+                                   const T* __restrict__ w, const float* __restrict__ inv_rms,
+                                   T* __restrict__ dx, float* __restrict__ dw,
+                                   int rows, int D) {
+  __shared__ float red[16];
+  // per-thread dw accumulator for its grid-stride columns (D/(256*VEC) <= 4
+  // vectors for D <= 8192)
+  float dwacc[4 * VEC];
+#pragma unroll
+  for (int j = 0; j < 4 * VEC; ++j) dwacc[j] = 0.f;
+
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* xr = x + (long)row * D;
+    const T* dyr = dy + (long)row * D;
+    T* dxr = dx + (long)row * D;
+    const float inv = inv_rms[row];
+    float dot = 0.f;
+    for (int i = threadIdx.x * VEC; i < D; i += blockDim.x * VEC) {
+      VecT<T, VEC> xv = vload<T, VEC>(xr + i);
+      VecT<T, VEC> dyv = vload<T, VEC>(dyr + i);
+      VecT<T, VEC> wv = vload<T, VEC>(w + i);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        dot += DT<T>::to_f32(dyv.data[j]) * DT<T>::to_f32(wv.data[j]) *
+               DT<T>::to_f32(xv.data[j]);
+    }
+    dot = block_reduce_sum(dot, red);
+    const float c = inv * inv * inv / D * dot;
+    int slot = 0;
+    for (int i = threadIdx.x * VEC; i < D; i += blockDim.x * VEC, ++slot) {
+      VecT<T, VEC> xv = vload<T, VEC>(xr + i);
+      VecT<T, VEC> dyv = vload<T, VEC>(dyr + i);
+      VecT<T, VEC> wv = vload<T, VEC>(w + i);
+      VecT<T, VEC> o;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float xf = DT<T>::to_f32(xv.data[j]);
+        float dyf = DT<T>::to_f32(dyv.data[j]);
+        float wf = DT<T>::to_f32(wv.data[j]);
+        o.data[j] = DT<T>::from_f32(inv * wf * dyf - xf * c);
+        dwacc[slot * VEC + j] += dyf * xf * inv;
+      }
+      vstore<T, VEC>(dxr + i, o);
+    }
+    __syncthreads();
+  }
+  int slot = 0;
+  for (int i = threadIdx.x * VEC; i < D; i += blockDim.x * VEC, ++slot)
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) atomicAdd(&dw[i + j], dwacc[slot * VEC + j]);
+}
+
+template <typename T>
+void launch_fwd(const torch::Tensor& x, const torch::Tensor& w, torch::Tensor& y,
+                torch::Tensor& inv, float eps, hipStream_t stream) {
+  const int rows = x.size(0), D = x.size(1);
+  constexpr int VEC = sizeof(T) == 4 ? 4 : 8;
+  TORCH_CHECK(D % VEC == 0, "rmsnorm: D must be a multiple of ", VEC);
+  int grid = std::min(rows, 2048);
+  hipLaunchKernelGGL((rmsnorm_fwd_kernel<T, VEC>), dim3(grid), dim3(256), 0, stream,
+                     (const T*)x.data_ptr(), (const T*)w.data_ptr(), (T*)y.data_ptr(),
+                     inv.data_ptr<float>(), rows, D, eps);
+  HIP_CHECK_KERNEL();
+}
+
+template <typename T>
+void launch_bwd(const torch::Tensor& dy, const torch::Tensor& x, const torch::Tensor& w,
+                const torch::Tensor& inv, torch::Tensor& dx, torch::Tensor& dw,
+                hipStream_t stream) {
+  const int rows = x.size(0), D = x.size(1);
+  constexpr int VEC = sizeof(T) == 4 ? 4 : 8;
+  TORCH_CHECK(D <= 256 * VEC * 4, "rmsnorm bwd: D too large for dw accumulator");
+  int grid = std::min(rows, 1024);
+  hipLaunchKernelGGL((rmsnorm_bwd_kernel<T, VEC>), dim3(grid), dim3(256), 0, stream,
+                     (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (const T*)w.data_ptr(),
+                     inv.data_ptr<float>(), (T*)dx.data_ptr(), dw.data_ptr<float>(), rows, D);
+  HIP_CHECK_KERNEL();
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  auto y = torch::empty_like(x);
+  auto inv = torch::empty({x.size(0)}, x.options().dtype(torch::kFloat));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  if (x.scalar_type() == torch::kBFloat16)
+    launch_fwd<__hip_bfloat16>(x, w, y, inv, (float)eps, stream);
+  else if (x.scalar_type() == torch::kHalf)
+    launch_fwd<__half>(x, w, y, inv, (float)eps, stream);
+  else if (x.scalar_type() == torch::kFloat)
+    launch_fwd<float>(x, w, y, inv, (float)eps, stream);
+  else
+    TORCH_CHECK(false, "rmsnorm: unsupported dtype");
+  return {y, inv};
+}
+
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
+                                       torch::Tensor inv) {
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros({x.size(1)}, x.options().dtype(torch::kFloat));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  if (x.scalar_type() == torch::kBFloat16)
+    launch_bwd<__hip_bfloat16>(dy, x, w, inv, dx, dw, stream);
+  else if (x.scalar_type() == torch::kHalf)
+    launch_bwd<__half>(dy, x, w, inv, dx, dw, stream);
+  else if (x.scalar_type() == torch::kFloat)
+    launch_bwd<float>(dy, x, w, inv, dx, dw, stream);
+  else
+    TORCH_CHECK(false, "rmsnorm: unsupported dtype");
+  return {dx, dw};
+}

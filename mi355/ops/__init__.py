@@ -208,10 +208,12 @@ def cross_entropy(logits, labels, ignore_index: int = -100):
 def adamw_step(params_f32, grads, exp_avgs, exp_avg_sqs, *, lr, beta1, beta2,
                eps, weight_decay, step, grad_scale: float = 1.0, outs_lowp=None):
     if params_f32 and _use_hip(params_f32[0]):
+        gs = grad_scale if torch.is_tensor(grad_scale) else \
+            torch.tensor(float(grad_scale), dtype=torch.float32)
         _load_ext().adamw_multi(
             params_f32, grads, exp_avgs, exp_avg_sqs,
             outs_lowp if outs_lowp is not None else [],
-            lr, beta1, beta2, eps, weight_decay, step, grad_scale,
+            lr, beta1, beta2, eps, weight_decay, step, gs.float(),
         )
         return
     torch_ref.adamw_step(

@@ -1,0 +1,127 @@
+"""Multi-process data-parallel correctness on CPU (gloo, world_size=2),
+mirroring the reference's no-GPU distributed mode (SURVEY.md §4.2:
+Scaling_model_training.ipynb runs 2-worker DDP with use_gpu=False)."""
+import multiprocessing as mp
+import os
+
+import pytest
+import torch
+
+
+def _worker(rank, world, port, q):
+    os.environ.update({
+        "RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": str(world),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    import torch.distributed as dist
+
+    from mi355.models.t5.configuration import t5_config
+    from mi355.models.t5.modeling import T5ForConditionalGeneration
+    from mi355.parallel import FlatAdamW, FlatReplica, init_distributed
+
+    init_distributed(backend="gloo")
+    torch.manual_seed(42)  # same init on both ranks before broadcast
+    cfg = t5_config("flan-t5-small", num_layers=2, num_decoder_layers=2, d_model=64,
+                    d_ff=128, num_heads=4, d_kv=16, vocab_size=256, dropout_rate=0.0)
+    model = T5ForConditionalGeneration(cfg)
+    replica = FlatReplica(model, bucket_mb=0.05)  # force several buckets
+    opt = FlatAdamW(replica, lr=1e-3, weight_decay=0.01)
+
+    torch.manual_seed(100 + rank)  # different data per rank
+    ids = torch.randint(0, 256, (2, 16))
+    for _ in range(3):
+        opt.zero_grad()
+        out = replica(input_ids=ids, labels=ids.clone())
+        out.loss.backward()
+        replica.finish_grad_sync()
+        opt.step()
+    # after sync steps the flat params must be identical across ranks
+    flat = torch.cat([g.flat_param for _, g in replica.groups]).double()
+    gathered = [torch.zeros_like(flat) for _ in range(world)]
+    dist.all_gather(gathered, flat)
+    same = torch.allclose(gathered[0], gathered[1], atol=1e-6)
+    # single-process equivalent: average both ranks' grads explicitly
+    q.put((rank, float(out.loss.item()), bool(same),
+           float(flat.sum().item())))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_flat_replica_two_ranks_stay_in_sync():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29531
+    procs = [ctx.Process(target=_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=240) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=60)
+    assert all(r[2] for r in results), f"ranks diverged: {results}"
+    # both ranks ended with the same params
+    assert results[0][3] == pytest.approx(results[1][3], rel=1e-9)
+
+
+def _worker_equiv(rank, world, port, q):
+    """2-rank run must equal a 1-process run on the concatenated batch."""
+    os.environ.update({
+        "RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": str(world),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    from mi355.models.t5.configuration import t5_config
+    from mi355.models.t5.modeling import T5ForConditionalGeneration
+    from mi355.parallel import FlatAdamW, FlatReplica, init_distributed
+
+    init_distributed(backend="gloo")
+    torch.manual_seed(7)
+    cfg = t5_config("flan-t5-small", num_layers=1, num_decoder_layers=1, d_model=32,
+                    d_ff=64, num_heads=2, d_kv=16, vocab_size=128, dropout_rate=0.0)
+    model = T5ForConditionalGeneration(cfg)
+    replica = FlatReplica(model, bucket_mb=0.01)
+    opt = FlatAdamW(replica, lr=1e-3, weight_decay=0.0, max_grad_norm=None)
+    torch.manual_seed(55)
+    ids = torch.randint(0, 128, (4, 8))  # same on both ranks
+    my = ids[rank * 2:(rank + 1) * 2]
+    opt.zero_grad()
+    out = replica(input_ids=my, labels=my.clone())
+    out.loss.backward()
+    replica.finish_grad_sync()
+    grad = torch.cat([g.flat_grad for _, g in replica.groups]).clone()
+    q.put((rank, grad))
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_ddp_grad_equals_full_batch_grad():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_equiv, args=(r, 2, 29532, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    res = {r: g for r, g in [q.get(timeout=240) for _ in range(2)]}
+    for p in procs:
+        p.join(timeout=60)
+
+    # single-process full-batch reference
+    from mi355.models.t5.configuration import t5_config
+    from mi355.models.t5.modeling import T5ForConditionalGeneration
+    from mi355.parallel import FlatReplica
+
+    torch.manual_seed(7)
+    cfg = t5_config("flan-t5-small", num_layers=1, num_decoder_layers=1, d_model=32,
+                    d_ff=64, num_heads=2, d_kv=16, vocab_size=128, dropout_rate=0.0)
+    model = T5ForConditionalGeneration(cfg)
+    replica = FlatReplica(model, bucket_mb=0.01)
+    torch.manual_seed(55)
+    ids = torch.randint(0, 128, (4, 8))
+    # loss is a mean over tokens; each rank's half has equal token count, so
+    # full-batch loss = mean of per-rank losses -> grads average the same way
+    l0 = replica(input_ids=ids[:2], labels=ids[:2].clone()).loss
+    l1 = replica(input_ids=ids[2:], labels=ids[2:].clone()).loss
+    ((l0 + l1) / 2).backward()
+    ref = torch.cat([g.flat_grad for _, g in replica.groups])
+
+    assert torch.allclose(res[0], res[1], atol=1e-6)
+    assert torch.allclose(res[0].float(), ref.float(), atol=1e-4), \
+        (res[0] - ref).abs().max()

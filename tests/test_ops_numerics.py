@@ -1,0 +1,212 @@
+"""Numerics: every HIP/CDNA4 kernel vs the plain-PyTorch fp32 reference
+(mi355/ops/torch_ref.py), per SURVEY.md §4's test strategy."""
+import math
+
+import pytest
+import torch
+
+from mi355 import ops
+from mi355.ops import torch_ref
+
+gpu = pytest.mark.gpu
+
+
+def _relerr(a, b):
+    a, b = a.float(), b.float()
+    return ((a - b).abs().max() / (b.abs().max() + 1e-6)).item()
+
+
+# --------------------------------------------------------------------------- #
+# CPU reference self-checks (run everywhere)                                  #
+# --------------------------------------------------------------------------- #
+def test_rms_norm_ref_matches_manual():
+    x = torch.randn(4, 8, dtype=torch.float64)
+    w = torch.randn(8, dtype=torch.float64)
+    y = torch_ref.rms_norm(x.float(), w.float(), 1e-6)
+    man = (x / torch.sqrt((x * x).mean(-1, keepdim=True) + 1e-6)) * w
+    assert torch.allclose(y, man.float(), atol=1e-5)
+
+
+def test_attention_ref_causal_mask():
+    q = torch.randn(1, 1, 4, 8)
+    k = torch.randn(1, 1, 4, 8)
+    v = torch.randn(1, 1, 4, 8)
+    o = torch_ref.attention(q, k, v, causal=True)
+    # row 0 can only attend to key 0
+    o0 = v[0, 0, 0]
+    assert torch.allclose(o[0, 0, 0], o0, atol=1e-5)
+
+
+def test_cross_entropy_ref_matches_torch():
+    logits = torch.randn(10, 50)
+    labels = torch.randint(0, 50, (10,))
+    labels[3] = -100
+    ref = torch.nn.functional.cross_entropy(logits, labels, ignore_index=-100)
+    assert torch.allclose(torch_ref.cross_entropy(logits, labels), ref, atol=1e-6)
+
+
+# --------------------------------------------------------------------------- #
+# HIP kernel vs fp32 reference (GPU)                                          #
+# --------------------------------------------------------------------------- #
+@gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16, torch.float16])
+@pytest.mark.parametrize("shape", [(128, 768), (1024, 768), (33, 512), (64, 2048)])
+def test_rmsnorm_fwd_bwd(dtype, shape):
+    torch.manual_seed(0)
+    tol = 1e-5 if dtype == torch.float32 else 2e-2
+    x = torch.randn(*shape, device="cuda", dtype=dtype, requires_grad=True)
+    w = torch.randn(shape[-1], device="cuda", dtype=dtype, requires_grad=True)
+    y = ops.rms_norm(x, w, 1e-6)
+    xr = x.detach().float().requires_grad_(True)
+    wr = w.detach().float().requires_grad_(True)
+    yr = torch_ref.rms_norm(xr, wr, 1e-6)
+    assert _relerr(y, yr) < tol
+    dy = torch.randn_like(yr)
+    y.backward(dy.to(dtype))
+    yr.backward(dy)
+    assert _relerr(x.grad, xr.grad) < tol
+    assert _relerr(w.grad, wr.grad) < tol * 4  # atomics + fp32 accum over rows
+
+
+@gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_geglu_fwd_bwd(dtype):
+    torch.manual_seed(1)
+    tol = 1e-5 if dtype == torch.float32 else 2e-2
+    g = torch.randn(512, 2048, device="cuda", dtype=dtype, requires_grad=True)
+    u = torch.randn(512, 2048, device="cuda", dtype=dtype, requires_grad=True)
+    out = ops.geglu(g, u)
+    gr = g.detach().float().requires_grad_(True)
+    ur = u.detach().float().requires_grad_(True)
+    outr = torch_ref.geglu(gr, ur)
+    assert _relerr(out, outr) < tol
+    d = torch.randn_like(outr)
+    out.backward(d.to(dtype))
+    outr.backward(d)
+    assert _relerr(g.grad, gr.grad) < tol
+    assert _relerr(u.grad, ur.grad) < tol
+
+
+@gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("rows,V", [(64, 32128), (1024, 32128), (7, 4096)])
+def test_cross_entropy_fwd_bwd(dtype, rows, V):
+    torch.manual_seed(2)
+    tol = 1e-5 if dtype == torch.float32 else 3e-2
+    logits = (torch.randn(rows, V, device="cuda") * 4).to(dtype).requires_grad_(True)
+    labels = torch.randint(0, V, (rows,), device="cuda")
+    labels[::5] = -100
+    loss = ops.cross_entropy(logits, labels)
+    lr = logits.detach().float().requires_grad_(True)
+    loss_r = torch_ref.cross_entropy(lr, labels)
+    assert abs(loss.item() - loss_r.item()) / abs(loss_r.item()) < tol
+    loss.backward()
+    loss_r.backward()
+    assert _relerr(logits.grad, lr.grad) < tol
+
+
+@gpu
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float16])
+@pytest.mark.parametrize(
+    "B,H,Sq,Sk,D,causal,with_bias,with_kpad",
+    [
+        (2, 12, 512, 512, 64, False, True, False),   # T5 encoder self-attn
+        (2, 12, 512, 512, 64, True, True, False),    # T5 decoder self-attn
+        (2, 12, 512, 512, 64, False, False, True),   # cross-attn w/ padding
+        (1, 4, 100, 333, 64, False, True, True),     # ragged tails
+        (1, 2, 64, 64, 32, True, False, False),      # SegFormer head dim
+        (3, 2, 1, 128, 64, False, True, False),      # single-query path
+    ],
+)
+def test_attention_fwd(dtype, B, H, Sq, Sk, D, causal, with_bias, with_kpad):
+    torch.manual_seed(3)
+    q = torch.randn(B, H, Sq, D, device="cuda", dtype=dtype)
+    k = torch.randn(B, H, Sk, D, device="cuda", dtype=dtype)
+    v = torch.randn(B, H, Sk, D, device="cuda", dtype=dtype)
+    bias = None
+    if with_bias:
+        bias = (torch.randn(1, H, Sq, Sk, device="cuda") * 2).to(dtype)
+    kpad = None
+    if with_kpad:
+        keep = torch.ones(B, Sk, device="cuda")
+        keep[:, Sk - Sk // 4:] = 0
+        kpad = (1 - keep) * -1e9
+    with torch.no_grad():
+        o = ops.attention(q, k, v, bias=bias, causal=causal, kpad=kpad)
+        o_ref = torch_ref.attention(
+            q.float(), k.float(), v.float(),
+            bias=bias.float() if bias is not None else None,
+            causal=causal, kpad=kpad,
+        )
+    err = _relerr(o, o_ref)
+    assert err < 3e-2, f"attn fwd mismatch: {err}"
+
+
+@gpu
+def test_attention_bwd_composed():
+    torch.manual_seed(4)
+    B, H, S, D = 2, 4, 256, 64
+    q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    v = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    bias = (torch.randn(1, H, S, S, device="cuda") * 2).to(torch.bfloat16).requires_grad_(True)
+    o = ops.attention(q, k, v, bias=bias, causal=True)
+    do = torch.randn_like(o)
+    o.backward(do)
+
+    qr, kr, vr = [t.detach().float().requires_grad_(True) for t in (q, k, v)]
+    br = bias.detach().float().requires_grad_(True)
+    o_ref = torch_ref.attention(qr, kr, vr, bias=br, causal=True)
+    o_ref.backward(do.float())
+    for ours, ref, name in [(q.grad, qr.grad, "dq"), (k.grad, kr.grad, "dk"),
+                            (v.grad, vr.grad, "dv"), (bias.grad, br.grad, "dbias")]:
+        err = _relerr(ours, ref)
+        assert err < 5e-2, f"{name} mismatch: {err}"
+
+
+@gpu
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float16])
+@pytest.mark.parametrize("L,D", [(64, 64), (100, 64), (1, 64), (512, 64), (37, 32)])
+def test_decode_attn(dtype, L, D):
+    torch.manual_seed(5)
+    B, H, Tmax = 4, 6, 512 + 16
+    q = torch.randn(B, H, 1, D, device="cuda", dtype=dtype)
+    kc = torch.randn(B, H, Tmax, D, device="cuda", dtype=dtype)
+    vc = torch.randn(B, H, Tmax, D, device="cuda", dtype=dtype)
+    bias = (torch.randn(1, H, 1, L, device="cuda")).to(dtype)
+    with torch.no_grad():
+        o = ops.decode_attention(q, kc, vc, L, bias=bias)
+        o_ref = torch_ref.attention(q.float(), kc[:, :, :L].float(), vc[:, :, :L].float(),
+                                    bias=bias.float())
+    assert _relerr(o, o_ref) < 3e-2
+
+
+@gpu
+@pytest.mark.parametrize("gdtype", [torch.float32, torch.bfloat16])
+def test_adamw_step(gdtype):
+    torch.manual_seed(6)
+    n = 12345
+    master = torch.randn(n, device="cuda")
+    grad = torch.randn(n, device="cuda").to(gdtype)
+    m = torch.randn(n, device="cuda").abs() * 0.01
+    v = torch.randn(n, device="cuda").abs() * 0.001
+    lowp = torch.zeros(n, device="cuda", dtype=torch.bfloat16)
+    ref = [t.clone() for t in (master, m, v)]
+    ops.adamw_step([master], [grad], [m], [v], lr=1e-3, beta1=0.9, beta2=0.999,
+                   eps=1e-8, weight_decay=0.01, step=3, grad_scale=0.5,
+                   outs_lowp=[lowp])
+    torch_ref.adamw_step([ref[0]], [grad], [ref[1]], [ref[2]], lr=1e-3, beta1=0.9,
+                         beta2=0.999, eps=1e-8, weight_decay=0.01, step=3,
+                         grad_scale=0.5)
+    assert _relerr(master, ref[0]) < 1e-5
+    assert _relerr(m, ref[1]) < 1e-5
+    assert _relerr(v, ref[2]) < 1e-5
+    assert _relerr(lowp.float(), ref[0]) < 1e-2  # bf16 write-back
+
+
+@gpu
+def test_native_ext_is_loaded():
+    """Guard against silent eager fallback on GPU (round-end check)."""
+    assert ops.hip_available()
+    import mi355._C as C
+    assert C.__file__.endswith("_C.so") and "/mi355/" in C.__file__
