@@ -9,7 +9,7 @@ torch::Tensor geglu_fwd(torch::Tensor gate, torch::Tensor up);
 std::vector<torch::Tensor> geglu_bwd(torch::Tensor dout, torch::Tensor gate, torch::Tensor up);
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels, long ignore_index);
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels, torch::Tensor lse,
-                     long ignore_index, double scale);
+                     long ignore_index, torch::Tensor dloss, torch::Tensor nvalid);
 void adamw_multi(std::vector<torch::Tensor> masters, std::vector<torch::Tensor> grads,
                  std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
                  std::vector<torch::Tensor> lowps, double lr, double beta1, double beta2,

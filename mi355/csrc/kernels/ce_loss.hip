@@ -51,7 +51,10 @@ __global__ void ce_fwd_kernel(const T* __restrict__ logits, const long* __restri
 template <typename T, int VEC>
 __global__ void ce_bwd_kernel(const T* __restrict__ logits, const long* __restrict__ labels,
                               const float* __restrict__ lse, T* __restrict__ dlogits,
-                              int rows, int V, long ignore_index, float scale) {
+                              int rows, int V, long ignore_index,
+                              const float* __restrict__ dloss,
+                              const int* __restrict__ nvalid) {
+  const float scale = dloss[0] / (float)max(nvalid[0], 1);
   for (int row = blockIdx.x; row < rows; row += gridDim.x) {
     const long lab = labels[row];
     const T* lr = logits + (long)row * V;
@@ -103,7 +106,7 @@ std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels,
 }
 
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels, torch::Tensor lse,
-                     long ignore_index, double scale) {
+                     long ignore_index, torch::Tensor dloss, torch::Tensor nvalid) {
   const int rows = logits.size(0), V = logits.size(1);
   auto dlogits = torch::empty_like(logits);
   auto stream = at::hip::getCurrentHIPStream().stream();
@@ -112,7 +115,7 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels, torch::Tensor l
   hipLaunchKernelGGL((ce_bwd_kernel<T, VEC>), dim3(grid), dim3(256), 0, stream,       \
                      (const T*)logits.data_ptr(), labels.data_ptr<long>(),            \
                      lse.data_ptr<float>(), (T*)dlogits.data_ptr(), rows, V,          \
-                     ignore_index, (float)scale)
+                     ignore_index, dloss.data_ptr<float>(), nvalid.data_ptr<int>())
   if (logits.scalar_type() == torch::kBFloat16) CASE(__hip_bfloat16, 8);
   else if (logits.scalar_type() == torch::kHalf) CASE(__half, 8);
   else if (logits.scalar_type() == torch::kFloat) CASE(float, 4);

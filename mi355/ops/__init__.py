@@ -182,18 +182,20 @@ class _CrossEntropy(torch.autograd.Function):
         logits2d = logits.contiguous().view(-1, logits.shape[-1])
         labels1d = labels.contiguous().view(-1)
         loss_sum, lse, nvalid = _load_ext().ce_fwd(logits2d, labels1d, ignore_index)
-        ctx.save_for_backward(logits2d, labels1d, lse)
+        ctx.save_for_backward(logits2d, labels1d, lse, nvalid)
+        ctx.orig_shape = logits.shape
         ctx.ignore_index = ignore_index
-        ctx.nvalid = max(int(nvalid), 1)
-        return loss_sum / ctx.nvalid
+        # stays on device: no host sync, keeps the step graph-capturable
+        return loss_sum / nvalid.clamp(min=1).float()
 
     @staticmethod
     def backward(ctx, dloss):
-        logits2d, labels1d, lse = ctx.saved_tensors
+        logits2d, labels1d, lse, nvalid = ctx.saved_tensors
         dlogits = _load_ext().ce_bwd(
-            logits2d, labels1d, lse, ctx.ignore_index, float(dloss) / ctx.nvalid
+            logits2d, labels1d, lse, ctx.ignore_index,
+            dloss.detach().float().reshape(1).contiguous(), nvalid.reshape(1).contiguous(),
         )
-        return dlogits.view(logits2d.shape), None, None
+        return dlogits.view(ctx.orig_shape), None, None
 
 
 def cross_entropy(logits, labels, ignore_index: int = -100):
