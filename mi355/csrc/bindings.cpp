@@ -19,6 +19,11 @@ torch::Tensor decode_attn(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                     c10::optional<torch::Tensor> bias,
                                     c10::optional<torch::Tensor> kpad, bool causal);
+torch::Tensor token_embed_fwd(torch::Tensor ids, torch::Tensor weight);
+torch::Tensor token_embed_bwd(torch::Tensor ids, torch::Tensor dy, long vocab,
+                              torch::ScalarType out_dtype);
+torch::Tensor rel_bias_fwd(torch::Tensor buckets, torch::Tensor weight);
+torch::Tensor rel_bias_bwd(torch::Tensor buckets, torch::Tensor dbias, long NB);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
@@ -30,5 +35,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_multi", &adamw_multi);
   m.def("decode_attn", &decode_attn);
   m.def("attn_fwd", &attn_fwd);
+  m.def("token_embed_fwd", &token_embed_fwd);
+  m.def("token_embed_bwd", &token_embed_bwd);
+  m.def("rel_bias_fwd", &rel_bias_fwd);
+  m.def("rel_bias_bwd", &rel_bias_bwd);
   m.attr("__gfx__") = "gfx950";
 }

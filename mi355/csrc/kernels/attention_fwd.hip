@@ -51,12 +51,15 @@ constexpr int BK = 64;   // key cols per iteration
 constexpr float NEG = -1e30f;
 
 // D in {32, 64}. LDS row pads (+8 elems) break 128B-stride conflicts.
+struct Strides3 { long sb, sh, ss; };  // (batch, head, seq) element strides; d is contiguous
+
 template <typename T, int D, bool CAUSAL, bool HAS_BIAS, bool HAS_KPAD>
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const T* __restrict__ q, const T* __restrict__ k, const T* __restrict__ v,
     const T* __restrict__ bias, const float* __restrict__ kpad,
     T* __restrict__ out, float* __restrict__ lse,
     int B, int H, int Sq, int Sk,
+    Strides3 qs, Strides3 ks, Strides3 vs, Strides3 os,
     long bias_sb, long bias_sh /* element strides; 0 = broadcast */) {
   using FR = typename MFMA16<T>::frag;
   constexpr int KLD = D + 8;       // K tile leading dim
@@ -78,8 +81,10 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int lg = lane >> 4;        // lane group 0..3
   const int lr = lane & 15;        // lane row/col within group
 
-  const long qoff = ((long)bh * Sq) * D;
-  const long koff = ((long)bh * Sk) * D;
+  const T* qbase = q + (long)b * qs.sb + (long)h * qs.sh;
+  const T* kbase = k + (long)b * ks.sb + (long)h * ks.sh;
+  const T* vbase = v + (long)b * vs.sb + (long)h * vs.sh;
+  T* obase = out + (long)b * os.sb + (long)h * os.sh;
   const T* bias_bh = HAS_BIAS ? bias + (long)b * bias_sb + (long)h * bias_sh : nullptr;
   const float* kpad_b = HAS_KPAD ? kpad + (long)b * Sk : nullptr;
 
@@ -89,7 +94,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   FR qfrag[D / 32];
 #pragma unroll
   for (int s = 0; s < D / 32; ++s) {
-    const T* src = q + qoff + (long)qrow_c * D + s * 32 + lg * 8;
+    const T* src = qbase + (long)qrow_c * qs.ss + s * 32 + lg * 8;
     qfrag[s] = *reinterpret_cast<const FR*>(src);
   }
 
@@ -116,8 +121,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         const int kr = k0 + jj;
         VecT<T, 8> kv, vv;
         if (kr < Sk) {
-          kv = vload<T, 8>(k + koff + (long)kr * D + d0);
-          vv = vload<T, 8>(v + koff + (long)kr * D + d0);
+          kv = vload<T, 8>(kbase + (long)kr * ks.ss + d0);
+          vv = vload<T, 8>(vbase + (long)kr * vs.ss + d0);
         } else {
 #pragma unroll
           for (int u = 0; u < 8; ++u) { kv.data[u] = T(0.f); vv.data[u] = T(0.f); }
@@ -148,7 +153,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     float rmax[4], rsum[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) rmax[r] = -INFINITY;
-    const int qbase = q0 + wid * 16 + lg * 4;
+    const int qrow0 = q0 + wid * 16 + lg * 4;
 #pragma unroll
     for (int t = 0; t < NKT; ++t) {
       const int kc = k0 + t * 16 + lr;
@@ -156,7 +161,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       const float kp = (HAS_KPAD && k_ok) ? kpad_b[kc] : 0.f;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int qr = qbase + r;
+        const int qr = qrow0 + r;
         float sv = s_acc[t][r];
         if (HAS_BIAS && k_ok && qr < Sq)
           sv += DT<T>::to_f32(bias_bh[(long)qr * Sk + kc]);
@@ -233,11 +238,15 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const float inv = (l_i[r] > 0.f) ? 1.f / l_i[r] : 0.f;
 #pragma unroll
     for (int t = 0; t < NDT; ++t)
-      out[qoff + (long)qr * D + t * 16 + lr] = DT<T>::from_f32(o_acc[t][r] * inv);
+      obase[(long)qr * os.ss + t * 16 + lr] = DT<T>::from_f32(o_acc[t][r] * inv);
     if (lr == 0)
       lse[(long)bh * Sq + qr] =
           (l_i[r] > 0.f) ? m_i[r] + logf(l_i[r]) : -INFINITY;
   }
+}
+
+static Strides3 strides_of(const torch::Tensor& t) {
+  return Strides3{t.stride(0), t.stride(1), t.stride(2)};
 }
 
 template <typename T, int D>
@@ -247,6 +256,8 @@ void launch_attn(const torch::Tensor& q, const torch::Tensor& k, const torch::Te
                  torch::Tensor& out, torch::Tensor& lse, hipStream_t stream) {
   const int B = q.size(0), H = q.size(1), Sq = q.size(2), Sk = k.size(2);
   dim3 grid((Sq + BQ - 1) / BQ, B * H);
+  Strides3 qst = strides_of(q), kst = strides_of(k), vst = strides_of(v),
+           ost = strides_of(out);
   const bool has_bias = bias.has_value() && bias->defined();
   const bool has_kpad = kpad.has_value() && kpad->defined();
   const T* bias_p = has_bias ? (const T*)bias->data_ptr() : nullptr;
@@ -263,7 +274,8 @@ void launch_attn(const torch::Tensor& q, const torch::Tensor& k, const torch::Te
   hipLaunchKernelGGL((attn_fwd_kernel<T, D, CAUSAL, HB, HK>), grid, dim3(256), 0,   \
                      stream, (const T*)q.data_ptr(), (const T*)k.data_ptr(),        \
                      (const T*)v.data_ptr(), bias_p, kpad_p, (T*)out.data_ptr(),    \
-                     lse.data_ptr<float>(), B, H, Sq, Sk, bias_sb, bias_sh)
+                     lse.data_ptr<float>(), B, H, Sq, Sk, qst, kst, vst, ost,       \
+                     bias_sb, bias_sh)
   if (causal) {
     if (has_bias && has_kpad) LAUNCH(true, true, true);
     else if (has_bias) LAUNCH(true, true, false);
@@ -284,10 +296,14 @@ void launch_attn(const torch::Tensor& q, const torch::Tensor& k, const torch::Te
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                     c10::optional<torch::Tensor> bias,
                                     c10::optional<torch::Tensor> kpad, bool causal) {
-  TORCH_CHECK(q.is_cuda() && q.dim() == 4 && q.is_contiguous(), "q must be (B,H,Sq,D) contiguous");
-  TORCH_CHECK(k.is_contiguous() && v.is_contiguous());
+  // (B,H,S,D) semantics with ARBITRARY (b,h,s) strides, d contiguous — so
+  // (B,S,H,D)-physical projection views need no transpose copies.
+  TORCH_CHECK(q.is_cuda() && q.dim() == 4 && q.stride(3) == 1, "q must be (B,H,Sq,D), d contiguous");
+  TORCH_CHECK(k.stride(3) == 1 && v.stride(3) == 1);
+  const int B = q.size(0), H = q.size(1), Sq = q.size(2);
   const int D = q.size(3);
-  auto out = torch::empty_like(q);
+  // output physically (B,Sq,H,D), returned as a (B,H,Sq,D) view
+  auto out = torch::empty({B, Sq, H, D}, q.options()).permute({0, 2, 1, 3});
   auto lse = torch::empty({q.size(0), q.size(1), q.size(2)},
                           q.options().dtype(torch::kFloat));
   auto stream = at::hip::getCurrentHIPStream().stream();

@@ -13,10 +13,13 @@
 
 namespace {
 
+struct DStr { long sb, sh, ss; };
+
 template <typename T, int D>
 __global__ void decode_attn_kernel(const T* __restrict__ q, const T* __restrict__ kc,
                                    const T* __restrict__ vc, const float* __restrict__ bias,
                                    T* __restrict__ out, int B, int H, int L, long t_max,
+                                   DStr qs, DStr kst, DStr vst, DStr ost,
                                    int bias_b_stride /*0 when broadcast*/) {
   static_assert(D == 32 || D == 64, "decode_attn: D in {32,64}");
   const int bh = blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
@@ -24,9 +27,9 @@ __global__ void decode_attn_kernel(const T* __restrict__ q, const T* __restrict_
   const int b = bh / H, h = bh % H;
   const int lane = threadIdx.x & (WAVE - 1);
 
-  const T* qrow = q + (long)bh * D;
-  const T* kbase = kc + (long)bh * t_max * D;
-  const T* vbase = vc + (long)bh * t_max * D;
+  const T* qrow = q + (long)b * qs.sb + (long)h * qs.sh;
+  const T* kbase = kc + (long)b * kst.sb + (long)h * kst.sh;
+  const T* vbase = vc + (long)b * vst.sb + (long)h * vst.sh;
   const float* brow = bias ? bias + (long)(bias_b_stride ? b : 0) * H * L + (long)h * L : nullptr;
 
   // q into registers: every lane holds the full row (D<=64 floats, L1 broadcast)
@@ -46,7 +49,7 @@ __global__ void decode_attn_kernel(const T* __restrict__ q, const T* __restrict_
     const int j = j0 + lane;
     float s = -INFINITY;
     if (j < L) {
-      const T* krow = kbase + (long)j * D;
+      const T* krow = kbase + (long)j * kst.ss;
       float acc = 0.f;
       constexpr int KV = sizeof(T) == 4 ? 4 : 8;
 #pragma unroll
@@ -72,7 +75,7 @@ __global__ void decode_attn_kernel(const T* __restrict__ q, const T* __restrict_
     const int jmax = min(WAVE, L - j0);
     for (int jj = 0; jj < jmax; ++jj) {
       const float pj = p_lds[wid][jj];
-      const T* vrow = vbase + (long)(j0 + jj) * D;
+      const T* vrow = vbase + (long)(j0 + jj) * vst.ss;
 #pragma unroll
       for (int du = 0; du < (D + WAVE - 1) / WAVE; ++du) {
         int d = du * WAVE + lane;
@@ -85,7 +88,7 @@ __global__ void decode_attn_kernel(const T* __restrict__ q, const T* __restrict_
 #pragma unroll
   for (int du = 0; du < (D + WAVE - 1) / WAVE; ++du) {
     int d = du * WAVE + lane;
-    if (d < D) out[(long)bh * D + d] = DT<T>::from_f32(o[du] * invl);
+    if (d < D) out[(long)b * ost.sb + (long)h * ost.sh + d] = DT<T>::from_f32(o[du] * invl);
   }
 }
 
@@ -93,10 +96,15 @@ __global__ void decode_attn_kernel(const T* __restrict__ q, const T* __restrict_
 
 torch::Tensor decode_attn(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
                           long cache_len, c10::optional<torch::Tensor> bias) {
-  TORCH_CHECK(q.is_cuda() && q.dim() == 4 && q.size(2) == 1, "q must be (B,H,1,D)");
+  TORCH_CHECK(q.is_cuda() && q.dim() == 4 && q.size(2) == 1 && q.stride(3) == 1,
+              "q must be (B,H,1,D), d contiguous");
   const int B = q.size(0), H = q.size(1), D = q.size(3);
   const long t_max = kc.size(2);
-  auto out = torch::empty_like(q);
+  auto out = torch::empty({B, 1, H, D}, q.options()).permute({0, 2, 1, 3});
+  DStr qs{q.stride(0), q.stride(1), q.stride(2)};
+  DStr kst{kc.stride(0), kc.stride(1), kc.stride(2)};
+  DStr vst{vc.stride(0), vc.stride(1), vc.stride(2)};
+  DStr ost{out.stride(0), out.stride(1), out.stride(2)};
   auto stream = at::hip::getCurrentHIPStream().stream();
   const float* bias_ptr = nullptr;
   int bias_b_stride = 0;
@@ -113,7 +121,7 @@ torch::Tensor decode_attn(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   hipLaunchKernelGGL((decode_attn_kernel<T, DD>), dim3(blocks), dim3(waves_per_block * WAVE), \
                      0, stream, (const T*)q.data_ptr(), (const T*)kc.data_ptr(),       \
                      (const T*)vc.data_ptr(), bias_ptr, (T*)out.data_ptr(), B, H,      \
-                     (int)cache_len, t_max, bias_b_stride)
+                     (int)cache_len, t_max, qs, kst, vst, ost, bias_b_stride)
   if (q.scalar_type() == torch::kBFloat16 && D == 64) CASE(__hip_bfloat16, 64);
   else if (q.scalar_type() == torch::kHalf && D == 64) CASE(__half, 64);
   else if (q.scalar_type() == torch::kFloat && D == 64) CASE(float, 64);

@@ -78,14 +78,23 @@ class T5Attention(nn.Module):
 
     def compute_bias(self, sq: int, sk: int, device) -> torch.Tensor:
         """(1, H, sq, sk) additive relative-position bias."""
+        buckets = self._buckets(sq, sk, device)
+        return ops.rel_position_bias(buckets, self.relative_attention_bias.weight)
+
+    def _buckets(self, sq: int, sk: int, device) -> torch.Tensor:
+        # bucket table depends only on shape — cache it across steps
+        key = (sq, sk, str(device))
+        cached = getattr(self, "_bucket_cache", None)
+        if cached is not None and cached[0] == key:
+            return cached[1]
         cpos = torch.arange(sq, dtype=torch.long, device=device)[:, None]
         mpos = torch.arange(sk, dtype=torch.long, device=device)[None, :]
         buckets = relative_position_bucket(
             mpos - cpos, bidirectional=not self.is_decoder,
             num_buckets=self.num_buckets, max_distance=self.max_distance,
-        )
-        vals = self.relative_attention_bias(buckets)  # (sq, sk, H)
-        return vals.permute(2, 0, 1).unsqueeze(0)
+        ).to(torch.int32).contiguous()
+        self._bucket_cache = (key, buckets)
+        return buckets
 
     def _shape(self, x, b):
         return x.view(b, -1, self.n_heads, self.d_kv).transpose(1, 2)
@@ -206,7 +215,9 @@ class T5Stack(nn.Module):
 
     def forward(self, input_ids=None, inputs_embeds=None, kpad=None, enc_hidden=None,
                 enc_kpad=None, caches=None, past_len: int = 0):
-        x = self.embed_tokens(input_ids) if inputs_embeds is None else inputs_embeds
+        if inputs_embeds is None:
+            inputs_embeds = ops.token_embedding(input_ids, self.embed_tokens.weight)
+        x = inputs_embeds
         x = self.dropout(x)
         sq = x.shape[1]
         sk = past_len + sq
@@ -357,7 +368,8 @@ class T5ForConditionalGeneration(nn.Module):
         return torch.stack(out_tokens, dim=1)
 
     def _decode_step(self, tokens, enc, kpad, caches, t, bias_t):
-        x = self.decoder.dropout(self.decoder.embed_tokens(tokens))
+        x = self.decoder.dropout(
+            ops.token_embedding(tokens, self.decoder.embed_tokens.weight))
         for i, blk in enumerate(self.decoder.block):
             x = blk(x, position_bias=bias_t, enc_hidden=enc, enc_kpad=kpad, causal=False,
                     self_cache=caches[i]["self"], cross_cache=caches[i]["cross"])

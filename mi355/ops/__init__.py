@@ -110,7 +110,11 @@ def geglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
 class _Attention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, bias, causal, kpad):
-        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        # arbitrary (b,h,s) strides supported by the kernel; only the head
+        # dim must be contiguous — projection views need no transpose copy
+        if q.stride(-1) != 1: q = q.contiguous()
+        if k.stride(-1) != 1: k = k.contiguous()
+        if v.stride(-1) != 1: v = v.contiguous()
         bias_c = bias.contiguous() if bias is not None else None
         kpad_c = kpad.contiguous().float() if kpad is not None else None
         o, lse = _load_ext().attn_fwd(q, k, v, bias_c, kpad_c, causal)
@@ -237,3 +241,56 @@ def decode_attention(q, k_cache, v_cache, cache_len: int, bias=None):
     k = k_cache[:, :, :cache_len]
     v = v_cache[:, :, :cache_len]
     return torch_ref.attention(q, k, v, bias=bias, causal=False)
+
+
+# --------------------------------------------------------------------------- #
+# Token embedding gather/scatter                               #
+# --------------------------------------------------------------------------- #
+class _TokenEmbed(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, ids, weight):
+        ctx.save_for_backward(ids)
+        ctx.vocab = weight.shape[0]
+        ctx.wdtype = weight.dtype
+        return _load_ext().token_embed_fwd(ids, weight)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (ids,) = ctx.saved_tensors
+        dw = _load_ext().token_embed_bwd(ids, dy, ctx.vocab, ctx.wdtype)
+        return None, dw
+
+
+def token_embedding(ids, weight):
+    if _use_hip(weight):
+        return _TokenEmbed.apply(ids, weight)
+    import torch.nn.functional as F
+
+    return F.embedding(ids, weight)
+
+
+# --------------------------------------------------------------------------- #
+# T5 relative-position bias: gather fwd / LDS-histogram bwd    #
+# --------------------------------------------------------------------------- #
+class _RelBias(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, buckets, weight):
+        ctx.save_for_backward(buckets)
+        ctx.nb = weight.shape[0]
+        ctx.wdtype = weight.dtype
+        return _load_ext().rel_bias_fwd(buckets, weight)
+
+    @staticmethod
+    def backward(ctx, dbias):
+        (buckets,) = ctx.saved_tensors
+        dw = _load_ext().rel_bias_bwd(buckets, dbias, ctx.nb)
+        return None, dw.to(ctx.wdtype)
+
+
+def rel_position_bias(buckets, weight):
+    """buckets: (Sq, Sk) int32; weight: (num_buckets, H).
+    Returns (1, H, Sq, Sk) additive bias."""
+    if _use_hip(weight):
+        return _RelBias.apply(buckets, weight)
+    vals = weight[buckets.long()]  # (Sq, Sk, H)
+    return vals.permute(2, 0, 1).unsqueeze(0)
