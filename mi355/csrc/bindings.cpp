@@ -24,6 +24,11 @@ torch::Tensor token_embed_bwd(torch::Tensor ids, torch::Tensor dy, long vocab,
                               torch::ScalarType out_dtype);
 torch::Tensor rel_bias_fwd(torch::Tensor buckets, torch::Tensor weight);
 torch::Tensor rel_bias_bwd(torch::Tensor buckets, torch::Tensor dbias, long NB);
+std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, torch::Tensor o, torch::Tensor lse,
+                                    c10::optional<torch::Tensor> bias,
+                                    c10::optional<torch::Tensor> kpad,
+                                    bool causal, bool need_dbias);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
@@ -39,5 +44,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("token_embed_bwd", &token_embed_bwd);
   m.def("rel_bias_fwd", &rel_bias_fwd);
   m.def("rel_bias_bwd", &rel_bias_bwd);
+  m.def("attn_bwd", &attn_bwd);
   m.attr("__gfx__") = "gfx950";
 }

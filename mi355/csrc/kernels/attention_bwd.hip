@@ -1,0 +1,425 @@
+// Flash-style attention BACKWARD for CDNA4/gfx950 (SURVEY.md K4 bwd,
+// "hard part #1": bwd through T5's additive relative-position bias).
+//
+// Given dO, Q, K, V, O, LSE: recompute P tile-by-tile and produce
+//   dV = P^T dO
+//   dP = dO V^T ;  dS = P * (dP - rowsum(dO*O))
+//   dK = dS^T Q ;  dQ = dS K  (fp32 atomic accumulation over k-tiles)
+//   dBias = sum_b dS  (fp32 atomic scatter, bias broadcast over batch)
+//
+// Geometry: block = 4 waves owns one (b, h, k-tile of 64 keys); waves own 16
+// key rows each; loop over q-tiles of 64. All MFMA v_mfma_f32_16x16x32.
+// Everything is staged in DYNAMIC LDS (~83 KB/block):
+//   kt[64][D+8], ktT[D][64+8], v[64][D+8], q[64][D+8], qT[D][64+8],
+//   do[64][D+8], doT[D][64+8], ds_img[64][64+8], slab[4][16][64+8]
+// Transposed images exist so every MFMA B-fragment is a contiguous 16-B
+// ds_read_b128 (row pads break the 128-B-stride bank conflict, guide G4).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "../common.h"
+
+namespace {
+
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef _Float16 f16x8 __attribute__((ext_vector_type(8)));
+
+template <typename T> struct MFMA16B;
+template <> struct MFMA16B<__hip_bfloat16> {
+  using frag = bf16x8;
+  static DEVINL f32x4 mma(frag a, frag b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+  }
+};
+template <> struct MFMA16B<__half> {
+  using frag = f16x8;
+  static DEVINL f32x4 mma(frag a, frag b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
+  }
+};
+
+constexpr int BKT = 64;  // keys per block
+constexpr int BQT = 64;  // queries per iteration
+constexpr float NEGB = -1e30f;
+
+struct BStr { long sb, sh, ss; };
+
+// delta[b,h,q] = sum_d dO * O (both strided BSHD views)
+template <typename T>
+__global__ void attn_delta_kernel(const T* __restrict__ dout, const T* __restrict__ o,
+                                  float* __restrict__ delta, int H, int Sq, int D,
+                                  BStr dos, BStr os) {
+  const long row = (long)blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+  const long total = (long)gridDim.y * 0 + 0;  // unused
+  (void)total;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const long nrows = (long)Sq * H * gridDim.y;  // gridDim.y == B
+  (void)nrows;
+  const int b = blockIdx.y;
+  if (row >= (long)H * Sq) return;
+  const int h = row / Sq, qr = row % Sq;
+  const T* dor = dout + b * dos.sb + h * dos.sh + (long)qr * dos.ss;
+  const T* orr = o + b * os.sb + h * os.sh + (long)qr * os.ss;
+  float acc = 0.f;
+  for (int d = lane; d < D; d += WAVE)
+    acc += DT<T>::to_f32(dor[d]) * DT<T>::to_f32(orr[d]);
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) delta[((long)b * H + h) * Sq + qr] = acc;
+}
+
+template <typename T, int D, bool CAUSAL, bool HAS_BIAS, bool HAS_KPAD, bool NEED_DBIAS>
+__global__ __launch_bounds__(256) void attn_bwd_kernel(
+    const T* __restrict__ dout, const T* __restrict__ q, const T* __restrict__ k,
+    const T* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
+    const T* __restrict__ bias, const float* __restrict__ kpad,
+    float* __restrict__ dq32, T* __restrict__ dk, T* __restrict__ dv,
+    float* __restrict__ dbias32,
+    int B, int H, int Sq, int Sk,
+    BStr qs, BStr ks, BStr vs, BStr dos, BStr dks, BStr dvs,
+    long bias_sb, long bias_sh) {
+  using FR = typename MFMA16B<T>::frag;
+  constexpr int LD = D + 8;        // [64][D] images
+  constexpr int LDT = BKT + 8;     // [D][64] transposed images + slabs
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  T* kt_img = (T*)smem;                         // [BKT][LD]
+  T* ktT_img = kt_img + BKT * LD;               // [D][LDT]
+  T* v_img = ktT_img + D * LDT;                 // [BKT][LD]
+  T* q_img = v_img + BKT * LD;                  // [BQT][LD]
+  T* qT_img = q_img + BQT * LD;                 // [D][LDT]
+  T* do_img = qT_img + D * LDT;                 // [BQT][LD]
+  T* doT_img = do_img + BQT * LD;               // [D][LDT]
+  T* ds_img = doT_img + D * LDT;                // [BQT][LDT] (q-major dS)
+  T* slab = ds_img + BQT * LDT;                 // [4][16][LDT] per-wave P^T / dS^T
+
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int k0 = blockIdx.x * BKT;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int lg = lane >> 4;
+  const int lr = lane & 15;
+
+  const T* qbase = q + (long)b * qs.sb + (long)h * qs.sh;
+  const T* kbase = k + (long)b * ks.sb + (long)h * ks.sh;
+  const T* vbase = v + (long)b * vs.sb + (long)h * vs.sh;
+  const T* dobase = dout + (long)b * dos.sb + (long)h * dos.sh;
+  T* dkbase = dk + (long)b * dks.sb + (long)h * dks.sh;
+  T* dvbase = dv + (long)b * dvs.sb + (long)h * dvs.sh;
+  const float* lse_bh = lse + ((long)b * H + h) * Sq;
+  const float* delta_bh = delta + ((long)b * H + h) * Sq;
+  const T* bias_bh = HAS_BIAS ? bias + (long)b * bias_sb + (long)h * bias_sh : nullptr;
+  const float* kpad_b = HAS_KPAD ? kpad + (long)b * Sk : nullptr;
+  float* dq_bh = dq32 + ((long)b * H + h) * (long)Sq * D;
+  float* dbias_h = NEED_DBIAS ? dbias32 + (long)h * Sq * Sk : nullptr;
+
+  // ---- stage this block's K and V (+ K^T image), zero-padded OOB ----
+  {
+    const int j = threadIdx.x / (D / 8);
+    const int d0 = (threadIdx.x % (D / 8)) * 8;
+    constexpr int RPP = 256 / (D / 8);
+    for (int jj = j; jj < BKT; jj += RPP) {
+      const int kr = k0 + jj;
+      VecT<T, 8> kv, vv;
+      if (kr < Sk) {
+        kv = vload<T, 8>(kbase + (long)kr * ks.ss + d0);
+        vv = vload<T, 8>(vbase + (long)kr * vs.ss + d0);
+      } else {
+#pragma unroll
+        for (int u = 0; u < 8; ++u) { kv.data[u] = T(0.f); vv.data[u] = T(0.f); }
+      }
+      vstore<T, 8>(&kt_img[jj * LD + d0], kv);
+      vstore<T, 8>(&v_img[jj * LD + d0], vv);
+#pragma unroll
+      for (int u = 0; u < 8; ++u) ktT_img[(d0 + u) * LDT + jj] = kv.data[u];
+    }
+  }
+
+  // accumulators: wave owns k rows [16*wid, 16*wid+16)
+  f32x4 dv_acc[D / 16], dk_acc[D / 16];
+#pragma unroll
+  for (int t = 0; t < D / 16; ++t) {
+    dv_acc[t] = {0.f, 0.f, 0.f, 0.f};
+    dk_acc[t] = {0.f, 0.f, 0.f, 0.f};
+  }
+
+  const int diag_off = Sk - Sq;
+  // causal: this k-tile sees queries with qr >= k0 - diag_off
+  int qt_start = 0;
+  if (CAUSAL) qt_start = max(0, (k0 - diag_off) / BQT);
+
+  T* my_slab = slab + wid * 16 * LDT;
+
+  for (int qt = qt_start; qt * BQT < Sq; ++qt) {
+    const int q0 = qt * BQT;
+    __syncthreads();  // previous iteration's reads done before restage
+    // ---- stage Q, dO (+ transposed images) ----
+    {
+      const int j = threadIdx.x / (D / 8);
+      const int d0 = (threadIdx.x % (D / 8)) * 8;
+      constexpr int RPP = 256 / (D / 8);
+      for (int jj = j; jj < BQT; jj += RPP) {
+        const int qr = q0 + jj;
+        VecT<T, 8> qv, dov;
+        if (qr < Sq) {
+          qv = vload<T, 8>(qbase + (long)qr * qs.ss + d0);
+          dov = vload<T, 8>(dobase + (long)qr * dos.ss + d0);
+        } else {
+#pragma unroll
+          for (int u = 0; u < 8; ++u) { qv.data[u] = T(0.f); dov.data[u] = T(0.f); }
+        }
+        vstore<T, 8>(&q_img[jj * LD + d0], qv);
+        vstore<T, 8>(&do_img[jj * LD + d0], dov);
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          qT_img[(d0 + u) * LDT + jj] = qv.data[u];
+          doT_img[(d0 + u) * LDT + jj] = dov.data[u];
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- S^T = K Q^T : M = wave's 16 k-rows, N = 64 q-cols, acc over d ----
+    f32x4 st_acc[4], dpt_acc[4];
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      st_acc[t] = {0.f, 0.f, 0.f, 0.f};
+      dpt_acc[t] = {0.f, 0.f, 0.f, 0.f};
+    }
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+#pragma unroll
+      for (int s = 0; s < D / 32; ++s) {
+        FR a = *reinterpret_cast<const FR*>(&kt_img[(wid * 16 + lr) * LD + s * 32 + lg * 8]);
+        FR bfr = *reinterpret_cast<const FR*>(&q_img[(t * 16 + lr) * LD + s * 32 + lg * 8]);
+        st_acc[t] = MFMA16B<T>::mma(a, bfr, st_acc[t]);
+        // dP^T = V dO^T (same geometry, different operands)
+        FR av = *reinterpret_cast<const FR*>(&v_img[(wid * 16 + lr) * LD + s * 32 + lg * 8]);
+        FR bdo = *reinterpret_cast<const FR*>(&do_img[(t * 16 + lr) * LD + s * 32 + lg * 8]);
+        dpt_acc[t] = MFMA16B<T>::mma(av, bdo, dpt_acc[t]);
+      }
+    }
+
+    // ---- recompute P^T, compute dS^T ----
+    // C layout: row (k-local) = lg*4 + r, col (q-local) = t*16 + lr
+    float pt[4][4], dst[4][4];
+    const int krow0 = k0 + wid * 16 + lg * 4;
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      const int qc = q0 + t * 16 + lr;
+      const bool q_ok = qc < Sq;
+      const float l = q_ok ? lse_bh[qc] : -INFINITY;
+      const float dl = q_ok ? delta_bh[qc] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int kr = krow0 + r;
+        float sv = st_acc[t][r];
+        const bool k_ok = kr < Sk;
+        if (HAS_BIAS && k_ok && q_ok) sv += DT<T>::to_f32(bias_bh[(long)qc * Sk + kr]);
+        if (HAS_KPAD && k_ok) sv += kpad_b[kr];
+        bool masked = !k_ok || !q_ok || (CAUSAL && kr > qc + diag_off);
+        float p = (masked || l == -INFINITY) ? 0.f : __expf(sv - l);
+        pt[t][r] = p;
+        dst[t][r] = p * (dpt_acc[t][r] - dl);
+      }
+    }
+
+    // ---- slab <- P^T ; dV += P^T dO ----
+#pragma unroll
+    for (int t = 0; t < 4; ++t)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        my_slab[(lg * 4 + r) * LDT + t * 16 + lr] = DT<T>::from_f32(pt[t][r]);
+    __syncthreads();  // slab visibility (uniform control flow)
+#pragma unroll
+    for (int t = 0; t < D / 16; ++t) {
+#pragma unroll
+      for (int s = 0; s < BQT / 32; ++s) {
+        FR a = *reinterpret_cast<const FR*>(&my_slab[lr * LDT + s * 32 + lg * 8]);
+        FR bfr = *reinterpret_cast<const FR*>(&doT_img[(t * 16 + lr) * LDT + s * 32 + lg * 8]);
+        dv_acc[t] = MFMA16B<T>::mma(a, bfr, dv_acc[t]);
+      }
+    }
+
+    // ---- slab <- dS^T ; dK += dS^T Q ; ds_img <- dS (q-major) ----
+#pragma unroll
+    for (int t = 0; t < 4; ++t)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const T val = DT<T>::from_f32(dst[t][r]);
+        my_slab[(lg * 4 + r) * LDT + t * 16 + lr] = val;
+        ds_img[(t * 16 + lr) * LDT + wid * 16 + lg * 4 + r] = val;
+      }
+    __syncthreads();
+#pragma unroll
+    for (int t = 0; t < D / 16; ++t) {
+#pragma unroll
+      for (int s = 0; s < BQT / 32; ++s) {
+        FR a = *reinterpret_cast<const FR*>(&my_slab[lr * LDT + s * 32 + lg * 8]);
+        FR bfr = *reinterpret_cast<const FR*>(&qT_img[(t * 16 + lr) * LDT + s * 32 + lg * 8]);
+        dk_acc[t] = MFMA16B<T>::mma(a, bfr, dk_acc[t]);
+      }
+    }
+
+    // ---- dBias scatter (bias broadcast over batch -> atomic sum) ----
+    if (NEED_DBIAS) {
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        const int qc = q0 + t * 16 + lr;
+        if (qc >= Sq) continue;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int kr = krow0 + r;
+          if (kr < Sk && dst[t][r] != 0.f)
+            atomicAdd(&dbias_h[(long)qc * Sk + kr], dst[t][r]);
+        }
+      }
+    }
+
+    __syncthreads();  // ds_img complete across waves
+
+    // ---- dQ(q-tile) = dS K : wave owns q rows [16*wid,+16), acc over k ----
+    f32x4 dq_acc[D / 16];
+#pragma unroll
+    for (int t = 0; t < D / 16; ++t) dq_acc[t] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int t = 0; t < D / 16; ++t) {
+#pragma unroll
+      for (int s = 0; s < BKT / 32; ++s) {
+        FR a = *reinterpret_cast<const FR*>(&ds_img[(wid * 16 + lr) * LDT + s * 32 + lg * 8]);
+        FR bfr = *reinterpret_cast<const FR*>(&ktT_img[(t * 16 + lr) * LDT + s * 32 + lg * 8]);
+        dq_acc[t] = MFMA16B<T>::mma(a, bfr, dq_acc[t]);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qr = q0 + wid * 16 + lg * 4 + r;
+      if (qr >= Sq) continue;
+#pragma unroll
+      for (int t = 0; t < D / 16; ++t)
+        atomicAdd(&dq_bh[(long)qr * D + t * 16 + lr], dq_acc[t][r]);
+    }
+  }
+
+  // ---- write dK, dV (block exclusively owns these k rows) ----
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int kr = k0 + wid * 16 + lg * 4 + r;
+    if (kr >= Sk) continue;
+#pragma unroll
+    for (int t = 0; t < D / 16; ++t) {
+      dkbase[(long)kr * dks.ss + t * 16 + lr] = DT<T>::from_f32(dk_acc[t][r]);
+      dvbase[(long)kr * dvs.ss + t * 16 + lr] = DT<T>::from_f32(dv_acc[t][r]);
+    }
+  }
+}
+
+template <typename T, int D>
+void launch_bwd(const torch::Tensor& dout, const torch::Tensor& q, const torch::Tensor& k,
+                const torch::Tensor& v, const torch::Tensor& o, const torch::Tensor& lse,
+                const c10::optional<torch::Tensor>& bias,
+                const c10::optional<torch::Tensor>& kpad, bool causal, bool need_dbias,
+                torch::Tensor& dq32, torch::Tensor& dk, torch::Tensor& dv,
+                torch::Tensor& dbias32, torch::Tensor& delta, hipStream_t stream) {
+  const int B = q.size(0), H = q.size(1), Sq = q.size(2), Sk = k.size(2);
+  BStr qs{q.stride(0), q.stride(1), q.stride(2)};
+  BStr ks{k.stride(0), k.stride(1), k.stride(2)};
+  BStr vs{v.stride(0), v.stride(1), v.stride(2)};
+  BStr dos{dout.stride(0), dout.stride(1), dout.stride(2)};
+  BStr oss{o.stride(0), o.stride(1), o.stride(2)};
+  BStr dks{dk.stride(0), dk.stride(1), dk.stride(2)};
+  BStr dvs{dv.stride(0), dv.stride(1), dv.stride(2)};
+
+  {  // delta = rowsum(dO * O)
+    const long rows = (long)H * Sq;
+    const int wpb = 4;
+    dim3 grid((rows + wpb - 1) / wpb, B);
+    hipLaunchKernelGGL((attn_delta_kernel<T>), grid, dim3(wpb * WAVE), 0, stream,
+                       (const T*)dout.data_ptr(), (const T*)o.data_ptr(),
+                       delta.data_ptr<float>(), H, Sq, D, dos, oss);
+    HIP_CHECK_KERNEL();
+  }
+
+  const bool has_bias = bias.has_value() && bias->defined();
+  const bool has_kpad = kpad.has_value() && kpad->defined();
+  const T* bias_p = has_bias ? (const T*)bias->data_ptr() : nullptr;
+  const float* kpad_p = has_kpad ? kpad->data_ptr<float>() : nullptr;
+  long bias_sb = 0, bias_sh = 0;
+  if (has_bias) {
+    bias_sb = bias->size(0) > 1 ? bias->stride(0) : 0;
+    bias_sh = bias->size(1) > 1 ? bias->stride(1) : 0;
+  }
+  constexpr int LD = D + 8, LDT = BKT + 8;
+  const int smem = sizeof(T) * (BKT * LD * 2 + BQT * LD * 2 + D * LDT * 3 +
+                                BQT * LDT + 4 * 16 * LDT);
+  dim3 grid((Sk + BKT - 1) / BKT, B * H);
+#define LB(C, HB, HK, DB)                                                             \
+  hipLaunchKernelGGL((attn_bwd_kernel<T, D, C, HB, HK, DB>), grid, dim3(256), smem,   \
+                     stream, (const T*)dout.data_ptr(), (const T*)q.data_ptr(),       \
+                     (const T*)k.data_ptr(), (const T*)v.data_ptr(),                  \
+                     lse.data_ptr<float>(), delta.data_ptr<float>(), bias_p, kpad_p,  \
+                     dq32.data_ptr<float>(), (T*)dk.data_ptr(), (T*)dv.data_ptr(),    \
+                     need_dbias ? dbias32.data_ptr<float>() : nullptr, B, H, Sq, Sk,  \
+                     qs, ks, vs, dos, dks, dvs, bias_sb, bias_sh)
+  // 16 template combos
+  if (causal) {
+    if (has_bias && need_dbias) { if (has_kpad) LB(true, true, true, true); else LB(true, true, false, true); }
+    else if (has_bias) { if (has_kpad) LB(true, true, true, false); else LB(true, true, false, false); }
+    else { if (has_kpad) LB(true, false, true, false); else LB(true, false, false, false); }
+  } else {
+    if (has_bias && need_dbias) { if (has_kpad) LB(false, true, true, true); else LB(false, true, false, true); }
+    else if (has_bias) { if (has_kpad) LB(false, true, true, false); else LB(false, true, false, false); }
+    else { if (has_kpad) LB(false, false, true, false); else LB(false, false, false, false); }
+  }
+#undef LB
+  HIP_CHECK_KERNEL();
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, torch::Tensor o, torch::Tensor lse,
+                                    c10::optional<torch::Tensor> bias,
+                                    c10::optional<torch::Tensor> kpad,
+                                    bool causal, bool need_dbias) {
+  TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1 &&
+              dout.stride(3) == 1 && o.stride(3) == 1, "head dim must be contiguous");
+  const int B = q.size(0), H = q.size(1), Sq = q.size(2), Sk = k.size(2), D = q.size(3);
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  auto opts = q.options();
+  auto dq32 = torch::zeros({B, H, Sq, D}, opts.dtype(torch::kFloat));
+  auto dk = torch::empty({B, Sk, H, D}, opts).permute({0, 2, 1, 3});
+  auto dv = torch::empty({B, Sk, H, D}, opts).permute({0, 2, 1, 3});
+  auto delta = torch::empty({B, H, Sq}, opts.dtype(torch::kFloat));
+  torch::Tensor dbias32;
+  if (need_dbias) {
+    TORCH_CHECK(bias.has_value() && bias->size(0) == 1,
+                "dbias path expects batch-broadcast bias (1,H,Sq,Sk)");
+    dbias32 = torch::zeros({1, H, Sq, Sk}, opts.dtype(torch::kFloat));
+  }
+  c10::optional<torch::Tensor> bias_c;
+  if (bias.has_value() && bias->defined()) bias_c = bias->to(q.scalar_type()).contiguous();
+
+  if (q.scalar_type() == torch::kBFloat16 && D == 64)
+    launch_bwd<__hip_bfloat16, 64>(dout, q, k, v, o, lse, bias_c, kpad, causal,
+                                   need_dbias, dq32, dk, dv, dbias32, delta, stream);
+  else if (q.scalar_type() == torch::kHalf && D == 64)
+    launch_bwd<__half, 64>(dout, q, k, v, o, lse, bias_c, kpad, causal, need_dbias,
+                           dq32, dk, dv, dbias32, delta, stream);
+  else if (q.scalar_type() == torch::kBFloat16 && D == 32)
+    launch_bwd<__hip_bfloat16, 32>(dout, q, k, v, o, lse, bias_c, kpad, causal,
+                                   need_dbias, dq32, dk, dv, dbias32, delta, stream);
+  else if (q.scalar_type() == torch::kHalf && D == 32)
+    launch_bwd<__half, 32>(dout, q, k, v, o, lse, bias_c, kpad, causal, need_dbias,
+                           dq32, dk, dv, dbias32, delta, stream);
+  else
+    TORCH_CHECK(false, "attn_bwd: unsupported dtype/D=", D);
+
+  // dq: fp32 BHSD accumulator -> low-precision BSHD-physical view
+  auto dq = torch::empty({B, Sq, H, D}, opts).permute({0, 2, 1, 3});
+  dq.copy_(dq32);
+  std::vector<torch::Tensor> outs = {dq, dk, dv};
+  if (need_dbias) outs.push_back(dbias32.to(q.scalar_type()));
+  return outs;
+}

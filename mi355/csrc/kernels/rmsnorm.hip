@@ -102,13 +102,17 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy, const T* __restrict
   }
 }
 
+// 2D: x over columns, y over row chunks; one atomic per (col, row-chunk)
 __global__ void colsum_kernel(const float* __restrict__ part, float* __restrict__ out,
                               int nparts, int D) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= D) return;
+  const int rpb = (nparts + gridDim.y - 1) / gridDim.y;
+  const int r0 = blockIdx.y * rpb, r1 = min(nparts, r0 + rpb);
   float s = 0.f;
-  for (int r = 0; r < nparts; ++r) s += part[(long)r * D + c];
-  out[c] = s;
+  for (int r = r0; r < r1; ++r) s += part[(long)r * D + c];
+  if (gridDim.y == 1) out[c] = s;
+  else atomicAdd(&out[c], s);
 }
 
 template <typename T>
@@ -140,7 +144,7 @@ void launch_bwd(const torch::Tensor& dy, const torch::Tensor& x, const torch::Te
                      inv.data_ptr<float>(), (T*)dx.data_ptr(), part.data_ptr<float>(),
                      rows, D);
   HIP_CHECK_KERNEL();
-  hipLaunchKernelGGL(colsum_kernel, dim3((D + 255) / 256), dim3(256), 0, stream,
+  hipLaunchKernelGGL(colsum_kernel, dim3((D + 255) / 256, 16), dim3(256), 0, stream,
                      part.data_ptr<float>(), dw.data_ptr<float>(), grid, D);
   HIP_CHECK_KERNEL();
 }
@@ -166,7 +170,7 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double 
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
                                        torch::Tensor inv) {
   auto dx = torch::empty_like(x);
-  auto dw = torch::empty({x.size(1)}, x.options().dtype(torch::kFloat));
+  auto dw = torch::zeros({x.size(1)}, x.options().dtype(torch::kFloat));
   auto stream = at::hip::getCurrentHIPStream().stream();
   if (x.scalar_type() == torch::kBFloat16)
     launch_bwd<__hip_bfloat16>(dy, x, w, inv, dx, dw, stream);

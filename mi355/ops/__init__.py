@@ -128,10 +128,16 @@ class _Attention(torch.autograd.Function):
         q, k, v, bias, kpad, o, lse = ctx.saved_tensors
         do = do.contiguous()
         ext = _load_ext()
-        if hasattr(ext, "attn_bwd"):
-            dq, dk, dv, dbias = ext.attn_bwd(
-                do, q, k, v, o, lse, bias, kpad, ctx.causal, ctx.bias_requires_grad
-            )
+        use_hip_bwd = hasattr(ext, "attn_bwd") and q.shape[-1] in (32, 64) and \
+            (not ctx.bias_requires_grad or (bias is not None and bias.shape[0] == 1)) \
+            and os.environ.get("MI355_ATTN_BWD_REF") != "1"
+        if use_hip_bwd:
+            if do.stride(-1) != 1:
+                do = do.contiguous()
+            res = ext.attn_bwd(do, q, k, v, o, lse, bias, kpad, ctx.causal,
+                               ctx.bias_requires_grad)
+            dq, dk, dv = res[0], res[1], res[2]
+            dbias = res[3] if len(res) > 3 else None
         else:  # composed bwd via recompute (hipBLASLt batched GEMMs)
             dq, dk, dv, dbias = _attn_bwd_composed(
                 do, q, k, v, lse, bias, kpad, ctx.causal, ctx.bias_requires_grad

@@ -143,23 +143,41 @@ def test_attention_fwd(dtype, B, H, Sq, Sk, D, causal, with_bias, with_kpad):
 
 
 @gpu
-def test_attention_bwd_composed():
+@pytest.mark.parametrize(
+    "B,H,Sq,Sk,D,causal,with_bias,with_kpad",
+    [
+        (2, 4, 256, 256, 64, True, True, False),     # decoder self-attn shape
+        (2, 4, 256, 256, 64, False, True, False),    # encoder self-attn shape
+        (2, 4, 128, 320, 64, False, False, True),    # cross-attn + padding
+        (1, 2, 100, 100, 64, True, True, False),     # ragged
+        (1, 2, 64, 64, 32, False, True, False),      # D=32
+    ],
+)
+def test_attention_bwd(B, H, Sq, Sk, D, causal, with_bias, with_kpad):
     torch.manual_seed(4)
-    B, H, S, D = 2, 4, 256, 64
-    q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
-    k = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
-    v = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
-    bias = (torch.randn(1, H, S, S, device="cuda") * 2).to(torch.bfloat16).requires_grad_(True)
-    o = ops.attention(q, k, v, bias=bias, causal=True)
+    q = torch.randn(B, H, Sq, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn(B, H, Sk, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    v = torch.randn(B, H, Sk, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    bias = None
+    if with_bias:
+        bias = (torch.randn(1, H, Sq, Sk, device="cuda") * 2).to(torch.bfloat16).requires_grad_(True)
+    kpad = None
+    if with_kpad:
+        keep = torch.ones(B, Sk, device="cuda")
+        keep[:, Sk - Sk // 4:] = 0
+        kpad = (1 - keep) * -1e9
+    o = ops.attention(q, k, v, bias=bias, causal=causal, kpad=kpad)
     do = torch.randn_like(o)
     o.backward(do)
 
     qr, kr, vr = [t.detach().float().requires_grad_(True) for t in (q, k, v)]
-    br = bias.detach().float().requires_grad_(True)
-    o_ref = torch_ref.attention(qr, kr, vr, bias=br, causal=True)
+    br = bias.detach().float().requires_grad_(True) if with_bias else None
+    o_ref = torch_ref.attention(qr, kr, vr, bias=br, causal=causal, kpad=kpad)
     o_ref.backward(do.float())
-    for ours, ref, name in [(q.grad, qr.grad, "dq"), (k.grad, kr.grad, "dk"),
-                            (v.grad, vr.grad, "dv"), (bias.grad, br.grad, "dbias")]:
+    pairs = [(q.grad, qr.grad, "dq"), (k.grad, kr.grad, "dk"), (v.grad, vr.grad, "dv")]
+    if with_bias:
+        pairs.append((bias.grad, br.grad, "dbias"))
+    for ours, ref, name in pairs:
         err = _relerr(ours, ref)
         assert err < 5e-2, f"{name} mismatch: {err}"
 
