@@ -1,0 +1,7 @@
+from .core import (ObjectRef, init, is_initialized, shutdown, put, get, wait,
+                   remote, Runtime)
+from .actor import ActorHandle, ActorClass
+from .actor_pool import ActorPool
+
+__all__ = ["ObjectRef", "init", "is_initialized", "shutdown", "put", "get",
+           "wait", "remote", "Runtime", "ActorHandle", "ActorClass", "ActorPool"]

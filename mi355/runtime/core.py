@@ -1,0 +1,423 @@
+"""MI355X-native single-node task/actor runtime.
+
+Replaces the Ray core the reference exercises (`ray.init`, `@ray.remote`,
+`ray.put/get/wait`, actors — reference call sites Overview_of_Ray.ipynb:
+cell-31/34/41/44, Scaling_batch_inference.ipynb:cell-88/91/105/115) with a
+from-scratch design sized for ONE 8-GPU MI355X node:
+
+ * worker pool of spawned processes, one task queue per worker;
+ * object store = POSIX shared memory segments (`/dev/shm/mi355_<ref>`),
+   written once then read-only (sealed-object discipline = race freedom,
+   SURVEY.md §5.2); payloads serialized with cloudpickle protocol 5 so
+   numpy/arrow buffers go out-of-band into the segment;
+ * ObjectRef resolution is decentralized: any process attaches the segment
+   by its deterministic name, so workers can `get()` without a round-trip
+   through the driver;
+ * GPU workers: worker i is pinned to GPU (i mod n_gpus) via
+   HIP_VISIBLE_DEVICES before torch initializes.
+"""
+from __future__ import annotations
+
+import atexit
+import io
+import os
+import pickle
+import queue
+import struct
+import threading
+import time
+import uuid
+import multiprocessing as mp
+from dataclasses import dataclass, field
+from multiprocessing import shared_memory
+
+import cloudpickle
+
+def _session_prefix() -> str:
+    sid = os.environ.setdefault("MI355_SESSION", uuid.uuid4().hex[:8])
+    return f"mi355_{sid}"
+
+
+def _untrack(name: str):
+    # Python 3.10's resource_tracker unlinks attached segments when ANY
+    # process exits; we manage lifetime ourselves (driver sweep at shutdown).
+    try:
+        from multiprocessing import resource_tracker
+
+        resource_tracker.unregister("/" + name, "shared_memory")
+    except Exception:
+        pass
+
+
+# --------------------------------------------------------------------------- #
+# Object store                                                                #
+# --------------------------------------------------------------------------- #
+@dataclass(frozen=True)
+class ObjectRef:
+    id: str
+
+    def hex(self):
+        return self.id
+
+    def __repr__(self):
+        return f"ObjectRef({self.id[:12]})"
+
+
+def _shm_name(ref: ObjectRef) -> str:
+    return f"{_session_prefix()}_{ref.id}"
+
+
+def _serialize(obj) -> bytes:
+    buffers = []
+    payload = cloudpickle.dumps(obj, protocol=5, buffer_callback=buffers.append)
+    out = io.BytesIO()
+    out.write(struct.pack("<I", len(buffers)))
+    out.write(struct.pack("<I", len(payload)))
+    out.write(payload)
+    for b in buffers:
+        raw = b.raw()
+        out.write(struct.pack("<Q", raw.nbytes))
+        out.write(raw)
+    return out.getvalue()
+
+
+def _deserialize(buf: memoryview):
+    off = 0
+    (nbuf,) = struct.unpack_from("<I", buf, off); off += 4
+    (plen,) = struct.unpack_from("<I", buf, off); off += 4
+    payload = bytes(buf[off:off + plen]); off += plen
+    buffers = []
+    for _ in range(nbuf):
+        (blen,) = struct.unpack_from("<Q", buf, off); off += 8
+        buffers.append(buf[off:off + blen]); off += blen
+    return pickle.loads(payload, buffers=buffers)
+
+
+class ObjectStore:
+    """Sealed shared-memory objects addressable by ref id from any process.
+
+    Attached segments are cached for the store's lifetime: deserialized
+    numpy/arrow buffers alias the mapped memory (zero-copy reads)."""
+
+    def __init__(self):
+        self._owned: list[str] = []
+        self._attached: dict[str, shared_memory.SharedMemory] = {}
+
+    def put(self, obj, ref: ObjectRef | None = None) -> ObjectRef:
+        ref = ref or ObjectRef(uuid.uuid4().hex)
+        data = _serialize(obj)
+        name = _shm_name(ref)
+        seg = shared_memory.SharedMemory(name=name, create=True, size=len(data) + 8)
+        _untrack(name)
+        # header: 8-byte payload length; the final length write "seals" it
+        seg.buf[8:8 + len(data)] = data
+        struct.pack_into("<Q", seg.buf, 0, len(data))
+        self._owned.append(name)
+        self._attached[name] = seg
+        return ref
+
+    def _attach(self, name):
+        seg = self._attached.get(name)
+        if seg is None:
+            seg = shared_memory.SharedMemory(name=name)
+            _untrack(name)
+            self._attached[name] = seg
+        return seg
+
+    def get(self, ref: ObjectRef, timeout: float | None = None):
+        name = _shm_name(ref)
+        deadline = None if timeout is None else time.monotonic() + timeout
+        delay = 0.0005
+        while True:
+            try:
+                seg = self._attach(name)
+                (n,) = struct.unpack_from("<Q", seg.buf, 0)
+                if n > 0:
+                    return _deserialize(memoryview(seg.buf)[8:8 + n])
+            except FileNotFoundError:
+                pass
+            if deadline is not None and time.monotonic() > deadline:
+                raise TimeoutError(f"object {ref} not available after {timeout}s")
+            time.sleep(delay)
+            delay = min(delay * 1.5, 0.02)
+
+    def ready(self, ref: ObjectRef) -> bool:
+        name = _shm_name(ref)
+        try:
+            seg = self._attach(name)
+            (n,) = struct.unpack_from("<Q", seg.buf, 0)
+            return n > 0
+        except FileNotFoundError:
+            return False
+
+    def cleanup(self):
+        import glob
+
+        for name, seg in list(self._attached.items()):
+            try:
+                seg.close()
+            except BufferError:
+                pass  # a live array still aliases it; unlink below suffices
+            self._attached.pop(name, None)
+        for name in self._owned:
+            try:
+                shared_memory.SharedMemory(name=name).unlink()
+            except FileNotFoundError:
+                pass
+            except Exception:
+                pass
+        self._owned.clear()
+
+    def sweep_session(self):
+        """Driver-side: unlink every segment of this session (incl. worker-
+        created results) at shutdown."""
+        import glob
+
+        for path in glob.glob(f"/dev/shm/{_session_prefix()}_*"):
+            try:
+                os.unlink(path)
+            except OSError:
+                pass
+
+
+# --------------------------------------------------------------------------- #
+# Worker process                                                              #
+# --------------------------------------------------------------------------- #
+def _resolve_args(store, args, kwargs):
+    def r(x):
+        return store.get(x) if isinstance(x, ObjectRef) else x
+
+    return [r(a) for a in args], {k: r(v) for k, v in kwargs.items()}
+
+
+def _worker_main(wid, gpu_id, task_q, done_q):
+    if gpu_id is not None:
+        os.environ["HIP_VISIBLE_DEVICES"] = str(gpu_id)
+        os.environ["CUDA_VISIBLE_DEVICES"] = str(gpu_id)
+    store = ObjectStore()
+    while True:
+        msg = task_q.get()
+        if msg is None:
+            break
+        kind = msg[0]
+        if kind == "task":
+            _, ref_id, fn_bytes, args, kwargs = msg
+            ref = ObjectRef(ref_id)
+            try:
+                fn = cloudpickle.loads(fn_bytes)
+                a, kw = _resolve_args(store, args, kwargs)
+                result = fn(*a, **kw)
+                store.put(result, ref)
+            except Exception as e:  # store the exception; raised at get()
+                import traceback
+
+                store.put(_RemoteError(e, traceback.format_exc()), ref)
+            done_q.put((wid, ref_id))
+    store.cleanup()
+
+
+@dataclass
+class _RemoteError:
+    exc: Exception
+    tb: str
+
+
+# --------------------------------------------------------------------------- #
+# Runtime (driver side)                                                       #
+# --------------------------------------------------------------------------- #
+class Runtime:
+    def __init__(self, num_workers: int | None = None, num_gpus: int | None = None):
+        import torch
+
+        _session_prefix()  # pin the session id BEFORE spawn so workers inherit it
+
+        self.num_gpus = torch.cuda.device_count() if num_gpus is None else num_gpus
+        if num_workers is None:
+            num_workers = max(2, min(os.cpu_count() or 4, 16))
+            if self.num_gpus:
+                num_workers = max(num_workers, self.num_gpus)
+        ctx = mp.get_context("spawn")
+        self.store = ObjectStore()
+        self.workers = []
+        self.done_q = ctx.Queue()
+        self._idle: "queue.Queue[int]" = queue.Queue()
+        self._inflight: dict[str, int] = {}
+        self._lock = threading.Lock()
+        for wid in range(num_workers):
+            gpu_id = wid % self.num_gpus if self.num_gpus else None
+            tq = ctx.Queue()
+            p = ctx.Process(target=_worker_main, args=(wid, gpu_id, tq, self.done_q),
+                            daemon=True)
+            p.start()
+            self.workers.append((p, tq, gpu_id))
+            self._idle.put(wid)
+        self._reaper = threading.Thread(target=self._reap, daemon=True)
+        self._alive = True
+        self._reaper.start()
+        self.actors: list = []
+        atexit.register(self.shutdown)
+
+    def _reap(self):
+        while self._alive:
+            try:
+                wid, ref_id = self.done_q.get(timeout=0.2)
+            except queue.Empty:
+                continue
+            except (EOFError, OSError):
+                return
+            with self._lock:
+                self._inflight.pop(ref_id, None)
+            self._idle.put(wid)
+
+    # -- tasks ---------------------------------------------------------- #
+    def submit(self, fn, args, kwargs, num_gpus: float = 0) -> ObjectRef:
+        ref = ObjectRef(uuid.uuid4().hex)
+        wid = self._pick_worker(num_gpus)
+        fn_bytes = fn if isinstance(fn, bytes) else cloudpickle.dumps(fn)
+        with self._lock:
+            self._inflight[ref.id] = wid
+        self.workers[wid][1].put(("task", ref.id, fn_bytes, list(args), dict(kwargs)))
+        return ref
+
+    def _pick_worker(self, num_gpus: float = 0) -> int:
+        # workers are GPU-pinned round-robin; any worker satisfies a CPU task
+        while True:
+            wid = self._idle.get()
+            if num_gpus <= 0 or self.workers[wid][2] is not None or self.num_gpus == 0:
+                return wid
+            self._idle.put(wid)  # CPU-only worker, GPU requested: rotate
+            time.sleep(0.0005)
+
+    # -- store ---------------------------------------------------------- #
+    def put(self, obj) -> ObjectRef:
+        return self.store.put(obj)
+
+    def get(self, refs, timeout: float | None = None):
+        single = isinstance(refs, ObjectRef)
+        lst = [refs] if single else list(refs)
+        out = []
+        for r in lst:
+            v = self.store.get(r, timeout=timeout)
+            if isinstance(v, _RemoteError):
+                raise RuntimeError(f"remote task failed:\n{v.tb}") from v.exc
+            out.append(v)
+        return out[0] if single else out
+
+    def wait(self, refs, num_returns=1, timeout=None):
+        refs = list(refs)
+        deadline = None if timeout is None else time.monotonic() + timeout
+        ready: list[ObjectRef] = []
+        while True:
+            ready = [r for r in refs if self.store.ready(r)]
+            if len(ready) >= num_returns:
+                break
+            if deadline is not None and time.monotonic() > deadline:
+                break
+            time.sleep(0.001)
+        ready = ready[:max(num_returns, len(ready))]
+        not_ready = [r for r in refs if r not in ready]
+        return ready, not_ready
+
+    def shutdown(self):
+        if not self._alive:
+            return
+        self._alive = False
+        for a in list(self.actors):
+            try:
+                a._kill()
+            except Exception:
+                pass
+        for p, tq, _ in self.workers:
+            try:
+                tq.put(None)
+            except Exception:
+                pass
+        for p, _, _ in self.workers:
+            p.join(timeout=2)
+            if p.is_alive():
+                p.terminate()
+        self.store.cleanup()
+        self.store.sweep_session()
+
+
+# --------------------------------------------------------------------------- #
+# Module-level API (ray-shaped)                                               #
+# --------------------------------------------------------------------------- #
+_runtime: Runtime | None = None
+
+
+def init(num_workers: int | None = None, num_gpus: int | None = None,
+         ignore_reinit_error: bool = True, **_ignored) -> Runtime:
+    global _runtime
+    if _runtime is not None:
+        if ignore_reinit_error:
+            return _runtime
+        raise RuntimeError("mi355 runtime already initialized")
+    _runtime = Runtime(num_workers=num_workers, num_gpus=num_gpus)
+    return _runtime
+
+
+def is_initialized() -> bool:
+    return _runtime is not None
+
+
+def shutdown():
+    global _runtime
+    if _runtime is not None:
+        _runtime.shutdown()
+        _runtime = None
+
+
+def _rt() -> Runtime:
+    if _runtime is None:
+        init()
+    return _runtime
+
+
+def put(obj) -> ObjectRef:
+    return _rt().put(obj)
+
+
+def get(refs, timeout: float | None = None):
+    return _rt().get(refs, timeout=timeout)
+
+
+def wait(refs, num_returns: int = 1, timeout: float | None = None):
+    return _rt().wait(refs, num_returns=num_returns, timeout=timeout)
+
+
+# -- @remote ---------------------------------------------------------------- #
+class RemoteFunction:
+    def __init__(self, fn, num_gpus=0, num_cpus=1, **_):
+        self._fn = fn
+        self._fn_bytes = None
+        self._num_gpus = num_gpus
+
+    def remote(self, *args, **kwargs) -> ObjectRef:
+        if self._fn_bytes is None:
+            self._fn_bytes = cloudpickle.dumps(self._fn)
+        return _rt().submit(self._fn_bytes, args, kwargs, num_gpus=self._num_gpus)
+
+    def options(self, **opts):
+        return RemoteFunction(self._fn, num_gpus=opts.get("num_gpus", self._num_gpus))
+
+    def __call__(self, *a, **k):
+        raise TypeError("remote functions must be called with .remote()")
+
+
+def remote(*args, **kwargs):
+    """@remote decorator for functions and classes."""
+    from .actor import ActorClass
+
+    if len(args) == 1 and callable(args[0]) and not kwargs:
+        target = args[0]
+        if isinstance(target, type):
+            return ActorClass(target)
+        return RemoteFunction(target)
+
+    def deco(target):
+        if isinstance(target, type):
+            return ActorClass(target, **kwargs)
+        return RemoteFunction(target, **kwargs)
+
+    return deco

@@ -1,0 +1,97 @@
+"""Runtime layer tests (C1-C5): tasks, object store, actors, ActorPool —
+mirrors the reference's Ray-core usage patterns (Overview_of_Ray.ipynb,
+Scaling_batch_inference.ipynb Part 5)."""
+import time
+
+import pytest
+
+import mi355
+
+
+@pytest.fixture(scope="module", autouse=True)
+def runtime():
+    mi355.init(num_workers=3, num_gpus=0)
+    yield
+    mi355.shutdown()
+
+
+def test_put_get_roundtrip():
+    import numpy as np
+    ref = mi355.put({"a": np.arange(10), "b": "text"})
+    out = mi355.get(ref)
+    assert out["b"] == "text" and out["a"].sum() == 45
+
+
+def test_remote_task_fanout():
+    @mi355.remote
+    def square(x):
+        return x * x
+
+    refs = [square.remote(i) for i in range(8)]
+    assert mi355.get(refs) == [i * i for i in range(8)]
+
+
+def test_task_with_object_ref_arg():
+    @mi355.remote
+    def total(data, scale):
+        return sum(data) * scale
+
+    data_ref = mi355.put([1, 2, 3])
+    assert mi355.get(total.remote(data_ref, 2)) == 12
+
+
+def test_task_error_propagates():
+    @mi355.remote
+    def boom():
+        raise ValueError("kapow")
+
+    with pytest.raises(RuntimeError, match="kapow"):
+        mi355.get(boom.remote())
+
+
+def test_wait_returns_ready_subset():
+    @mi355.remote
+    def slow(t):
+        time.sleep(t)
+        return t
+
+    refs = [slow.remote(0.01), slow.remote(1.5)]
+    ready, not_ready = mi355.wait(refs, num_returns=1, timeout=1.0)
+    assert len(ready) == 1 and mi355.get(ready[0]) == 0.01
+    assert len(not_ready) == 1
+
+
+def test_actor_holds_state():
+    @mi355.remote
+    class Counter:
+        def __init__(self, start):
+            self.n = start
+
+        def incr(self, by=1):
+            self.n += by
+            return self.n
+
+    c = Counter.remote(10)
+    assert mi355.get(c.incr.remote()) == 11
+    assert mi355.get(c.incr.remote(5)) == 16
+
+
+def test_actor_pool_map_unordered():
+    @mi355.remote
+    class Worker:
+        def process(self, x):
+            return x + 100
+
+    pool = mi355.ActorPool([Worker.remote() for _ in range(2)])
+    out = pool.map_unordered(lambda a, v: a.process.remote(v), list(range(6)))
+    assert sorted(out) == [100, 101, 102, 103, 104, 105]
+
+
+def test_actor_pool_map_ordered():
+    @mi355.remote
+    class Worker:
+        def process(self, x):
+            return x * 2
+
+    pool = mi355.ActorPool([Worker.remote() for _ in range(2)])
+    assert pool.map(lambda a, v: a.process.remote(v), [3, 1, 2]) == [6, 2, 4]
