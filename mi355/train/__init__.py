@@ -1,0 +1,7 @@
+from . import session
+from .hf import Trainer, TrainingArguments
+from .hf_trainer import HuggingFaceTrainer
+from .trainer import DataParallelTrainer, TorchTrainer
+
+__all__ = ["session", "Trainer", "TrainingArguments", "HuggingFaceTrainer",
+           "DataParallelTrainer", "TorchTrainer"]

@@ -1,0 +1,127 @@
+"""Dataset + Preprocessor tests (C6/C7) mirroring the reference's data ops
+(Introduction_to_Ray_AI_Runtime.ipynb, Scaling_model_training.ipynb)."""
+import numpy as np
+import pandas as pd
+import pytest
+
+import mi355.data as mdata
+from mi355.data import BatchMapper, Chain, Dataset, MinMaxScaler, PowerTransformer
+
+
+@pytest.fixture
+def taxi_like():
+    rng = np.random.default_rng(0)
+    return mdata.from_pandas(pd.DataFrame({
+        "trip_distance": rng.uniform(0, 30, 200),
+        "fare_amount": rng.uniform(2, 100, 200),
+        "passenger_count": rng.integers(1, 5, 200),
+    }))
+
+
+def test_from_items_count_take():
+    ds = mdata.from_items([{"a": i, "b": i * 2} for i in range(10)])
+    assert ds.count() == 10
+    assert ds.take(3) == [{"a": 0, "b": 0}, {"a": 1, "b": 2}, {"a": 2, "b": 4}]
+
+
+def test_repartition_and_split(taxi_like):
+    ds = taxi_like.repartition(5)
+    assert ds.num_blocks() == 5
+    assert ds.count() == 200
+    train, test = ds.train_test_split(test_size=0.3)
+    assert train.count() == 140 and test.count() == 60
+
+
+def test_limit_schema_columns(taxi_like):
+    ds = taxi_like.limit(17)
+    assert ds.count() == 17
+    assert set(ds.columns()) == {"trip_distance", "fare_amount", "passenger_count"}
+    assert ds.schema() is not None
+
+
+def test_map_batches_pandas(taxi_like):
+    def add_tip(df):
+        df = df.copy()
+        df["tip"] = df["fare_amount"] * 0.2
+        return df
+
+    out = taxi_like.map_batches(add_tip, batch_format="pandas", batch_size=32)
+    assert "tip" in out.columns()
+    assert out.count() == 200
+
+
+def test_map_batches_numpy():
+    ds = mdata.from_items([{"x": float(i)} for i in range(100)])
+    out = ds.map_batches(lambda b: {"x": b["x"] * 2}, batch_format="numpy",
+                         batch_size=30)
+    assert out.to_pandas()["x"].tolist() == [2.0 * i for i in range(100)]
+
+
+def test_groupby_mean(taxi_like):
+    g = taxi_like.groupby("passenger_count").mean("fare_amount")
+    df = g.to_pandas()
+    assert "mean(fare_amount)" in df.columns
+    ref = taxi_like.to_pandas().groupby("passenger_count")["fare_amount"].mean()
+    got = dict(zip(df["passenger_count"], df["mean(fare_amount)"]))
+    for k, v in ref.items():
+        assert got[k] == pytest.approx(v)
+
+
+def test_drop_columns_iter_batches(taxi_like):
+    ds = taxi_like.drop_columns(["passenger_count"])
+    assert set(ds.columns()) == {"trip_distance", "fare_amount"}
+    sizes = [len(b) for b in ds.iter_batches(batch_size=64, batch_format="pandas")]
+    assert sum(sizes) == 200 and max(sizes) == 64
+
+
+def test_min_max_scaler(taxi_like):
+    sc = MinMaxScaler(columns=["trip_distance"])
+    out = sc.fit_transform(taxi_like).to_pandas()
+    assert out["trip_distance"].min() == pytest.approx(0.0)
+    assert out["trip_distance"].max() == pytest.approx(1.0)
+    # transform_batch on a raw batch (predictor path)
+    b = sc.transform_batch(pd.DataFrame({"trip_distance": [0.0], "fare_amount": [1.0],
+                                         "passenger_count": [1]}))
+    assert b["trip_distance"].iloc[0] <= 0.0
+
+
+def test_power_transformer(taxi_like):
+    pt = PowerTransformer(columns=["fare_amount"], power=0.5)
+    out = pt.fit_transform(taxi_like).to_pandas()
+    raw = taxi_like.to_pandas()["fare_amount"].to_numpy()
+    expect = ((raw + 1) ** 0.5 - 1) / 0.5
+    np.testing.assert_allclose(out["fare_amount"].to_numpy(), expect, rtol=1e-6)
+
+
+def test_batch_mapper_tokenize_style():
+    ds = mdata.from_items([{"instruction": f"do {i}", "input": str(i)} for i in range(8)])
+
+    def preprocess(df):
+        return pd.DataFrame({
+            "input_ids": [[len(a), len(b)] for a, b in zip(df.instruction, df.input)],
+        })
+
+    bm = BatchMapper(preprocess, batch_format="pandas", batch_size=4096)
+    out = bm.transform(ds)
+    assert out.count() == 8 and out.columns() == ["input_ids"]
+
+
+def test_chain(taxi_like):
+    chain = Chain(MinMaxScaler(columns=["trip_distance"]),
+                  BatchMapper(lambda df: df.assign(z=df.trip_distance + 1)))
+    out = chain.fit(taxi_like).transform(taxi_like).to_pandas()
+    assert out["z"].max() <= 2.0
+
+
+def test_iter_torch_batches():
+    import torch
+    ds = mdata.from_items([{"x": [1.0, 2.0], "y": 3} for _ in range(10)])
+    batches = list(ds.iter_torch_batches(batch_size=4))
+    assert batches[0]["x"].shape == (4, 2)
+    assert batches[0]["y"].dtype in (torch.int64,)
+
+
+def test_sort_and_unique(taxi_like):
+    s = taxi_like.sort("fare_amount", descending=True).take(2)
+    assert s[0]["fare_amount"] >= s[1]["fare_amount"]
+    assert taxi_like.unique("passenger_count") == [1, 2, 3, 4]
