@@ -36,18 +36,20 @@ class _Session:
     stopped: bool = False
 
     def report(self, metrics: dict, checkpoint_dir: str | None = None) -> bool:
-        """Send metrics (+ checkpoint path, rank0) to the driver. Returns
-        False if the driver asked this trial to STOP (ASHA early stop)."""
+        """Send metrics (+ checkpoint path, rank0) to the driver and BLOCK
+        for its continue/stop decision (keeps DDP ranks in lockstep so an
+        ASHA early stop lands on every rank at the same epoch). Returns
+        False when the trial should stop."""
         if self.report_q is not None:
             self.report_q.put(("report", self.context.world_rank, dict(metrics),
                                checkpoint_dir))
-        if self.control_q is not None:
-            try:
-                msg = self.control_q.get_nowait()
-                if msg == "stop":
+            if self.control_q is not None:
+                try:
+                    msg = self.control_q.get(timeout=600)
+                    if msg == "stop":
+                        self.stopped = True
+                except _queue.Empty:
                     self.stopped = True
-            except _queue.Empty:
-                pass
         return not self.stopped
 
 

@@ -166,7 +166,11 @@ class DataParallelTrainer:
         error = None
         done = set()
         n_ckpt = 0
-        self._stop_sent = False
+        # synchronous report protocol: the i-th report of every rank gets the
+        # decision computed at rank 0's i-th report (lockstep early stop)
+        report_seq = {r: 0 for r in range(world)}
+        decisions = []
+        waiting = {}
         while len(done) < world:
             try:
                 kind, rank, payload, ckpt_dir = report_q.get(timeout=600)
@@ -181,6 +185,8 @@ class DataParallelTrainer:
                 for q in control_qs:
                     q.put("stop")
             elif kind == "report":
+                seq = report_seq[rank]
+                report_seq[rank] += 1
                 if rank == 0:
                     last_metrics = payload
                     metrics_rows.append(payload)
@@ -202,7 +208,13 @@ class DataParallelTrainer:
                             for _, old, _ in tracked[keep:]:
                                 shutil.rmtree(old.path, ignore_errors=True)
                             tracked = tracked[:keep]
-                    self._on_report(payload, control_qs)
+                    decisions.append(self._decide(payload))
+                    for w in waiting.pop(len(decisions) - 1, []):
+                        control_qs[w].put(decisions[-1])
+                if seq < len(decisions):
+                    control_qs[rank].put(decisions[seq])
+                else:
+                    waiting.setdefault(seq, []).append(rank)
 
         for p in procs:
             p.join(timeout=30)
@@ -222,14 +234,16 @@ class DataParallelTrainer:
             config=dict(self._train_loop_config or {}),
         )
 
-    def _on_report(self, metrics: dict, control_qs):
-        """Tuner/ASHA hook: subclass/driver may send 'stop'."""
+    def _decide(self, metrics: dict) -> str:
+        """Tuner/ASHA hook: continue/stop decision for this report epoch."""
         cb = getattr(self, "_report_callback", None)
-        if cb is not None and not self._stop_sent:
-            if cb(metrics) == "stop":
-                self._stop_sent = True
-                for q in control_qs:
-                    q.put("stop")
+        if cb is not None:
+            try:
+                if cb(metrics) == "stop":
+                    return "stop"
+            except Exception:
+                pass
+        return "continue"
 
 
 class TorchTrainer(DataParallelTrainer):
