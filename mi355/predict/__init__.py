@@ -1,0 +1,5 @@
+from .batch_predictor import BatchPredictor
+from .hf_predictor import HuggingFacePredictor, TorchPredictor
+from .predictor import Predictor
+
+__all__ = ["BatchPredictor", "Predictor", "HuggingFacePredictor", "TorchPredictor"]

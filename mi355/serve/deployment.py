@@ -1,0 +1,189 @@
+"""serve internals: deployments, replicas, HTTP proxy."""
+from __future__ import annotations
+
+import itertools
+import json
+import threading
+from dataclasses import dataclass, field, replace
+from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+
+import pandas as pd
+
+
+# -- HTTP adapters (reference: ray.serve.http_adapters.pandas_read_json) ---- #
+def pandas_read_json(body: bytes) -> pd.DataFrame:
+    return pd.DataFrame(json.loads(body))
+
+
+def json_request(body: bytes):
+    return json.loads(body)
+
+
+# -- deployments ------------------------------------------------------------ #
+@dataclass
+class Deployment:
+    func_or_class: object = None
+    name: str = "default"
+    num_replicas: int = 1
+    route_prefix: str = "/"
+    ray_actor_options: dict = field(default_factory=dict)
+
+    def options(self, name=None, num_replicas=None, route_prefix=None,
+                ray_actor_options=None, **_):
+        return replace(
+            self,
+            name=name or self.name,
+            num_replicas=num_replicas or self.num_replicas,
+            route_prefix=route_prefix or self.route_prefix,
+            ray_actor_options=ray_actor_options or self.ray_actor_options,
+        )
+
+    def bind(self, *args, **kwargs) -> "Application":
+        return Application(self, args, kwargs)
+
+
+def deployment(cls_or_fn=None, **opts):
+    if cls_or_fn is not None:
+        return Deployment(func_or_class=cls_or_fn)
+
+    def deco(target):
+        return Deployment(func_or_class=target, **opts)
+
+    return deco
+
+
+@dataclass
+class Application:
+    deployment: Deployment
+    args: tuple
+    kwargs: dict
+
+
+class _PredictorReplicaGroup:
+    """Round-robin pool of predictor replicas for a PredictorDeployment."""
+
+    def __init__(self, app: Application):
+        d = app.deployment
+        predictor_cls, checkpoint = app.args[0], app.args[1]
+        self.http_adapter = app.kwargs.get("http_adapter", pandas_read_json)
+        pred_kwargs = {k: v for k, v in app.kwargs.items() if k != "http_adapter"}
+        self.replicas = [predictor_cls.from_checkpoint(checkpoint, **pred_kwargs)
+                         for _ in range(d.num_replicas)]
+        self._rr = itertools.cycle(range(len(self.replicas)))
+        self._locks = [threading.Lock() for _ in self.replicas]
+
+    def handle(self, body: bytes, query_kwargs: dict):
+        data = self.http_adapter(body)
+        i = next(self._rr)
+        with self._locks[i]:
+            out = self.replicas[i].predict(data, **query_kwargs)
+        if isinstance(out, pd.DataFrame):
+            return out.to_dict(orient="records")
+        import numpy as np
+
+        if isinstance(out, (list, np.ndarray)):
+            return [{"prediction": _py(v)} for v in out]
+        return out
+
+
+def _py(v):
+    import numpy as np
+
+    if isinstance(v, np.generic):
+        return v.item()
+    if isinstance(v, np.ndarray):
+        return v.tolist()
+    return v
+
+
+PredictorDeployment = Deployment(func_or_class=_PredictorReplicaGroup,
+                                 name="PredictorDeployment")
+
+
+# -- proxy / controller ------------------------------------------------------ #
+class _ServeState:
+    def __init__(self):
+        self.routes: dict[str, _PredictorReplicaGroup] = {}
+        self.apps: dict[str, Application] = {}
+        self.server: ThreadingHTTPServer | None = None
+        self.thread: threading.Thread | None = None
+
+
+_state = _ServeState()
+
+
+class _Handler(BaseHTTPRequestHandler):
+    def log_message(self, *a):  # quiet
+        pass
+
+    def do_POST(self):
+        length = int(self.headers.get("Content-Length", 0))
+        body = self.rfile.read(length)
+        route = self.path.split("?")[0].rstrip("/") or "/"
+        group = _state.routes.get(route)
+        if group is None:
+            self.send_response(404)
+            self.end_headers()
+            self.wfile.write(b'{"error": "no such route"}')
+            return
+        try:
+            result = group.handle(body, {})
+            payload = json.dumps(result).encode()
+            self.send_response(200)
+            self.send_header("Content-Type", "application/json")
+            self.end_headers()
+            self.wfile.write(payload)
+        except Exception as e:
+            self.send_response(500)
+            self.end_headers()
+            self.wfile.write(json.dumps({"error": str(e)}).encode())
+
+    def do_GET(self):
+        if self.path in ("/", "/-/routes"):
+            self.send_response(200)
+            self.end_headers()
+            self.wfile.write(json.dumps(
+                {r: d.deployment.name for r, d in _state.apps.items()}).encode())
+        else:
+            self.do_POST()
+
+
+def run(app: Application, host: str = "127.0.0.1", port: int = 8000,
+        _blocking: bool = False, **_):
+    """Deploy the application and (idempotently) start the HTTP proxy."""
+    d = app.deployment
+    group = _PredictorReplicaGroup(app)
+    route = d.route_prefix.rstrip("/") or "/"
+    _state.routes[route] = group
+    _state.apps[route] = app
+    if _state.server is None:
+        _state.server = ThreadingHTTPServer((host, port), _Handler)
+        _state.thread = threading.Thread(target=_state.server.serve_forever,
+                                         daemon=True)
+        _state.thread.start()
+    return RayServeHandle(group)
+
+
+@dataclass
+class RayServeHandle:
+    group: _PredictorReplicaGroup
+
+    def predict(self, df):
+        i = next(self.group._rr)
+        return self.group.replicas[i].predict(df)
+
+
+def get_deployment(name: str):
+    for route, app in _state.apps.items():
+        if app.deployment.name == name:
+            return app.deployment
+    raise KeyError(name)
+
+
+def shutdown():
+    if _state.server is not None:
+        _state.server.shutdown()
+        _state.server = None
+        _state.thread = None
+    _state.routes.clear()
+    _state.apps.clear()

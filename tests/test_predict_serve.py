@@ -1,0 +1,101 @@
+"""BatchPredictor / Predictor / serve tests (C14-C17) — mirror the reference
+inference architectures."""
+import json
+import urllib.request
+
+import numpy as np
+import pandas as pd
+import pytest
+import torch
+
+from mi355.air import Checkpoint
+from mi355.data import Dataset, from_pandas
+from mi355.predict import BatchPredictor, HuggingFacePredictor, Predictor
+
+
+class DoublingPredictor(Predictor):
+    """Custom Predictor subclass via _predict_pandas (reference
+    Scaling_batch_inference.ipynb:cell-73 pattern)."""
+
+    def __init__(self, factor, preprocessor=None):
+        super().__init__(preprocessor)
+        self.factor = factor
+
+    @classmethod
+    def from_checkpoint(cls, ckpt, **kw):
+        d = ckpt.to_dict()
+        return cls(d["factor"], preprocessor=ckpt.get_preprocessor())
+
+    def _predict_pandas(self, df, **kw):
+        return pd.DataFrame({"out": df["x"] * self.factor})
+
+
+def test_checkpoint_dict_predictor_roundtrip():
+    ckpt = Checkpoint.from_dict({"factor": 3})
+    p = DoublingPredictor.from_checkpoint(ckpt)
+    out = p.predict(pd.DataFrame({"x": [1, 2, 3]}))
+    assert out["out"].tolist() == [3, 6, 9]
+
+
+def test_batch_predictor_over_dataset():
+    ds = from_pandas(pd.DataFrame({"x": list(range(100))})).repartition(4)
+    bp = BatchPredictor.from_checkpoint(Checkpoint.from_dict({"factor": 2}),
+                                        DoublingPredictor)
+    out = bp.predict(ds, batch_size=16)
+    assert out.count() == 100
+    assert out.to_pandas()["out"].tolist() == [2 * i for i in range(100)]
+
+
+def test_batch_predictor_keep_columns():
+    ds = from_pandas(pd.DataFrame({"x": [1, 2], "id": ["a", "b"]}))
+    bp = BatchPredictor.from_checkpoint(Checkpoint.from_dict({"factor": 10}),
+                                        DoublingPredictor)
+    out = bp.predict(ds, feature_columns=["x"], keep_columns=["id"]).to_pandas()
+    assert out["id"].tolist() == ["a", "b"] and out["out"].tolist() == [10, 20]
+
+
+def test_hf_predictor_generate_from_checkpoint(tmp_path):
+    from mi355.models.t5.configuration import t5_config
+    from mi355.models.t5.modeling import T5ForConditionalGeneration
+
+    torch.manual_seed(0)
+    cfg = t5_config("flan-t5-small", num_layers=1, num_decoder_layers=1,
+                    d_model=32, d_ff=64, num_heads=2, d_kv=16, vocab_size=128,
+                    dropout_rate=0.0)
+    model = T5ForConditionalGeneration(cfg)
+    model.save_pretrained(str(tmp_path))
+    ckpt = Checkpoint.from_directory(str(tmp_path))
+    pred = HuggingFacePredictor.from_checkpoint(ckpt, task="text2text-generation")
+    df = pd.DataFrame({"input_ids": [np.arange(8) + 2] * 3,
+                       "attention_mask": [np.ones(8, dtype=np.int64)] * 3})
+    out = pred.predict(df, max_new_tokens=4, min_new_tokens=4)
+    assert len(out) == 3
+    assert all(len(t) == 4 for t in out["generated_tokens"])
+
+
+@pytest.mark.timeout(120)
+def test_serve_http_roundtrip():
+    from mi355 import serve
+
+    app = serve.PredictorDeployment.options(
+        name="DoubleService", num_replicas=2, route_prefix="/double",
+    ).bind(DoublingPredictor, Checkpoint.from_dict({"factor": 5}),
+           http_adapter=serve.pandas_read_json)
+    serve.run(app, port=18755)
+    try:
+        req = urllib.request.Request(
+            "http://127.0.0.1:18755/double",
+            data=json.dumps([{"x": 1}, {"x": 4}]).encode(),
+            headers={"Content-Type": "application/json"})
+        with urllib.request.urlopen(req, timeout=30) as resp:
+            body = json.loads(resp.read())
+        assert body == [{"out": 5}, {"out": 20}]
+        # unknown route -> 404
+        req2 = urllib.request.Request("http://127.0.0.1:18755/nope", data=b"[]")
+        try:
+            urllib.request.urlopen(req2, timeout=30)
+            assert False, "expected 404"
+        except urllib.error.HTTPError as e:
+            assert e.code == 404
+    finally:
+        serve.shutdown()
