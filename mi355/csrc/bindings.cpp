@@ -29,6 +29,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
                                     c10::optional<torch::Tensor> bias,
                                     c10::optional<torch::Tensor> kpad,
                                     bool causal, bool need_dbias);
+torch::Tensor seg_postprocess(torch::Tensor logits, long Ho, long Wo);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
@@ -45,5 +46,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rel_bias_fwd", &rel_bias_fwd);
   m.def("rel_bias_bwd", &rel_bias_bwd);
   m.def("attn_bwd", &attn_bwd);
+  m.def("seg_postprocess", &seg_postprocess);
   m.attr("__gfx__") = "gfx950";
 }

@@ -1,0 +1,299 @@
+"""MI355X-native SegFormer (reference
+Scaling_batch_inference.ipynb:cell-20 segformer-b0-finetuned-ade-512-512,
+Scaling_model_training.ipynb:cell-16 nvidia/mit-b0).
+
+Hierarchical transformer encoder (4 stages, overlapping patch embeds,
+efficient self-attention with spatial reduction, Mix-FFN with depthwise
+conv) + all-MLP decode head. The attention core runs the same hand-written
+CDNA4 flash kernel as T5 (head dim 32; SegFormer's 1/sqrt(d) scaling is
+folded into Q before the call since the kernel is unscaled-T5-style);
+convolutions go through torch's MIOpen path.
+"""
+from __future__ import annotations
+
+import json
+import math
+import os
+from dataclasses import asdict, dataclass, field
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ... import ops
+
+
+@dataclass
+class SegformerConfig:
+    num_channels: int = 3
+    num_encoder_blocks: int = 4
+    depths: list = field(default_factory=lambda: [2, 2, 2, 2])
+    sr_ratios: list = field(default_factory=lambda: [8, 4, 2, 1])
+    hidden_sizes: list = field(default_factory=lambda: [32, 64, 160, 256])
+    patch_sizes: list = field(default_factory=lambda: [7, 3, 3, 3])
+    strides: list = field(default_factory=lambda: [4, 2, 2, 2])
+    num_attention_heads: list = field(default_factory=lambda: [1, 2, 5, 8])
+    mlp_ratios: list = field(default_factory=lambda: [4, 4, 4, 4])
+    hidden_dropout_prob: float = 0.0
+    attention_probs_dropout_prob: float = 0.0
+    classifier_dropout_prob: float = 0.1
+    layer_norm_eps: float = 1e-6
+    decoder_hidden_size: int = 256
+    num_labels: int = 150
+    semantic_loss_ignore_index: int = 255
+    model_type: str = "segformer"
+    id2label: dict | None = None
+
+    def to_dict(self):
+        return asdict(self)
+
+    @classmethod
+    def from_dict(cls, d):
+        known = set(cls.__dataclass_fields__)
+        return cls(**{k: v for k, v in d.items() if k in known})
+
+    def save_pretrained(self, path):
+        os.makedirs(path, exist_ok=True)
+        with open(os.path.join(path, "config.json"), "w") as f:
+            json.dump(self.to_dict(), f, indent=2, sort_keys=True)
+
+    @classmethod
+    def from_pretrained(cls, path):
+        with open(os.path.join(path, "config.json")) as f:
+            return cls.from_dict(json.load(f))
+
+
+class OverlapPatchEmbed(nn.Module):
+    def __init__(self, in_ch, out_ch, patch, stride, eps):
+        super().__init__()
+        self.proj = nn.Conv2d(in_ch, out_ch, patch, stride, patch // 2)
+        self.layer_norm = nn.LayerNorm(out_ch, eps=eps)
+
+    def forward(self, x):
+        x = self.proj(x)
+        _, _, h, w = x.shape
+        x = x.flatten(2).transpose(1, 2)  # (B, N, C)
+        return self.layer_norm(x), h, w
+
+
+class EfficientSelfAttention(nn.Module):
+    def __init__(self, dim, heads, sr_ratio, eps):
+        super().__init__()
+        self.dim, self.heads = dim, heads
+        self.head_dim = dim // heads
+        self.scale = self.head_dim**-0.5
+        self.query = nn.Linear(dim, dim)
+        self.key = nn.Linear(dim, dim)
+        self.value = nn.Linear(dim, dim)
+        self.sr_ratio = sr_ratio
+        if sr_ratio > 1:
+            self.sr = nn.Conv2d(dim, dim, sr_ratio, sr_ratio)
+            self.layer_norm = nn.LayerNorm(dim, eps=eps)
+
+    def forward(self, x, h, w):
+        b, n, c = x.shape
+        # scale folded into Q: kernel computes softmax(QK^T) unscaled
+        q = (self.query(x) * self.scale).view(b, n, self.heads, self.head_dim).transpose(1, 2)
+        if self.sr_ratio > 1:
+            xr = x.transpose(1, 2).reshape(b, c, h, w)
+            xr = self.sr(xr).flatten(2).transpose(1, 2)
+            xr = self.layer_norm(xr)
+        else:
+            xr = x
+        m = xr.shape[1]
+        k = self.key(xr).view(b, m, self.heads, self.head_dim).transpose(1, 2)
+        v = self.value(xr).view(b, m, self.heads, self.head_dim).transpose(1, 2)
+        o = ops.attention(q, k, v, causal=False)
+        return o.transpose(1, 2).reshape(b, n, c)
+
+
+class SegformerAttention(nn.Module):
+    def __init__(self, dim, heads, sr_ratio, eps):
+        super().__init__()
+        self.self = EfficientSelfAttention(dim, heads, sr_ratio, eps)
+        self.output = nn.Linear(dim, dim)
+
+    def forward(self, x, h, w):
+        return self.output(self.self(x, h, w))
+
+
+class DWConv(nn.Module):
+    def __init__(self, dim):
+        super().__init__()
+        self.dwconv = nn.Conv2d(dim, dim, 3, 1, 1, groups=dim)
+
+    def forward(self, x, h, w):
+        b, n, c = x.shape
+        x = x.transpose(1, 2).view(b, c, h, w)
+        x = self.dwconv(x)
+        return x.flatten(2).transpose(1, 2)
+
+
+class MixFFN(nn.Module):
+    def __init__(self, dim, hidden, eps):
+        super().__init__()
+        self.dense1 = nn.Linear(dim, hidden)
+        self.dwconv = DWConv(hidden)
+        self.dense2 = nn.Linear(hidden, dim)
+
+    def forward(self, x, h, w):
+        x = self.dense1(x)
+        x = self.dwconv(x, h, w)
+        x = F.gelu(x)
+        return self.dense2(x)
+
+
+class SegformerLayer(nn.Module):
+    def __init__(self, dim, heads, sr_ratio, mlp_ratio, eps):
+        super().__init__()
+        self.layer_norm_1 = nn.LayerNorm(dim, eps=eps)
+        self.attention = SegformerAttention(dim, heads, sr_ratio, eps)
+        self.layer_norm_2 = nn.LayerNorm(dim, eps=eps)
+        self.mlp = MixFFN(dim, dim * mlp_ratio, eps)
+
+    def forward(self, x, h, w):
+        x = x + self.attention(self.layer_norm_1(x), h, w)
+        x = x + self.mlp(self.layer_norm_2(x), h, w)
+        return x
+
+
+class SegformerEncoder(nn.Module):
+    def __init__(self, cfg: SegformerConfig):
+        super().__init__()
+        self.cfg = cfg
+        embeds, blocks, norms = [], [], []
+        in_ch = cfg.num_channels
+        for i in range(cfg.num_encoder_blocks):
+            embeds.append(OverlapPatchEmbed(in_ch, cfg.hidden_sizes[i],
+                                            cfg.patch_sizes[i], cfg.strides[i],
+                                            cfg.layer_norm_eps))
+            blocks.append(nn.ModuleList([
+                SegformerLayer(cfg.hidden_sizes[i], cfg.num_attention_heads[i],
+                               cfg.sr_ratios[i], cfg.mlp_ratios[i],
+                               cfg.layer_norm_eps)
+                for _ in range(cfg.depths[i])
+            ]))
+            norms.append(nn.LayerNorm(cfg.hidden_sizes[i], eps=cfg.layer_norm_eps))
+            in_ch = cfg.hidden_sizes[i]
+        self.patch_embeddings = nn.ModuleList(embeds)
+        self.block = nn.ModuleList(blocks)
+        self.layer_norm = nn.ModuleList(norms)
+
+    def forward(self, pixel_values):
+        b = pixel_values.shape[0]
+        hidden = pixel_values
+        features = []
+        for embed, layers, norm in zip(self.patch_embeddings, self.block,
+                                       self.layer_norm):
+            hidden, h, w = embed(hidden)
+            for layer in layers:
+                hidden = layer(hidden, h, w)
+            hidden = norm(hidden)
+            feat = hidden.transpose(1, 2).reshape(b, -1, h, w)
+            features.append(feat)
+            hidden = feat
+        return features
+
+
+class SegformerDecodeHead(nn.Module):
+    def __init__(self, cfg: SegformerConfig):
+        super().__init__()
+        self.linear_c = nn.ModuleList(
+            [nn.Linear(d, cfg.decoder_hidden_size) for d in cfg.hidden_sizes])
+        self.linear_fuse = nn.Conv2d(cfg.decoder_hidden_size * 4,
+                                     cfg.decoder_hidden_size, 1, bias=False)
+        self.batch_norm = nn.BatchNorm2d(cfg.decoder_hidden_size)
+        self.dropout = nn.Dropout(cfg.classifier_dropout_prob)
+        self.classifier = nn.Conv2d(cfg.decoder_hidden_size, cfg.num_labels, 1)
+
+    def forward(self, features):
+        b = features[0].shape[0]
+        size = features[0].shape[2:]
+        ups = []
+        for feat, proj in zip(features, self.linear_c):
+            h, w = feat.shape[2], feat.shape[3]
+            x = proj(feat.flatten(2).transpose(1, 2))
+            x = x.transpose(1, 2).reshape(b, -1, h, w)
+            if x.shape[2:] != size:
+                x = F.interpolate(x.float(), size=size, mode="bilinear",
+                                  align_corners=False).to(x.dtype)
+            ups.append(x)
+        x = self.linear_fuse(torch.cat(ups[::-1], dim=1))
+        x = self.batch_norm(x.float()).to(x.dtype)
+        x = F.relu(x)
+        return self.classifier(self.dropout(x))
+
+
+@dataclass
+class SegformerOutput:
+    loss: torch.Tensor | None = None
+    logits: torch.Tensor | None = None
+
+    def __getitem__(self, k):
+        return getattr(self, k)
+
+
+class SegformerForSemanticSegmentation(nn.Module):
+    config_class = SegformerConfig
+
+    def __init__(self, config: SegformerConfig):
+        super().__init__()
+        self.config = config
+        self.segformer = SegformerEncoder(config)
+        self.decode_head = SegformerDecodeHead(config)
+        self.apply(self._init)
+
+    @staticmethod
+    def _init(m):
+        if isinstance(m, nn.Linear):
+            nn.init.trunc_normal_(m.weight, std=0.02)
+            if m.bias is not None:
+                nn.init.zeros_(m.bias)
+        elif isinstance(m, nn.Conv2d):
+            fan_out = m.kernel_size[0] * m.kernel_size[1] * m.out_channels // m.groups
+            m.weight.data.normal_(0, math.sqrt(2.0 / fan_out))
+            if m.bias is not None:
+                nn.init.zeros_(m.bias)
+
+    @property
+    def device(self):
+        return next(self.parameters()).device
+
+    def forward(self, pixel_values=None, labels=None, **_):
+        features = self.segformer(pixel_values)
+        logits = self.decode_head(features)  # (B, L, H/4, W/4)
+        loss = None
+        if labels is not None:
+            up = F.interpolate(logits.float(), size=labels.shape[-2:],
+                               mode="bilinear", align_corners=False)
+            loss = F.cross_entropy(up, labels.long(),
+                                   ignore_index=self.config.semantic_loss_ignore_index)
+        return SegformerOutput(loss=loss, logits=logits)
+
+    # HF-layout checkpoints
+    def save_pretrained(self, path, safe_serialization=True):
+        os.makedirs(path, exist_ok=True)
+        self.config.save_pretrained(path)
+        from safetensors.torch import save_file
+
+        save_file({k: v.contiguous().cpu() for k, v in self.state_dict().items()},
+                  os.path.join(path, "model.safetensors"), metadata={"format": "pt"})
+
+    @classmethod
+    def from_pretrained(cls, path, torch_dtype=None, config=None, **_):
+        if config is None and os.path.isdir(path):
+            config = SegformerConfig.from_pretrained(path)
+        elif config is None:
+            config = SegformerConfig()  # mit-b0 defaults
+        model = cls(config)
+        st = os.path.join(path, "model.safetensors") if os.path.isdir(path) else None
+        if st and os.path.exists(st):
+            from safetensors.torch import load_file
+
+            model.load_state_dict(load_file(st))
+        if torch_dtype is not None:
+            model = model.to(torch_dtype)
+        return model
+
+
+SegformerForImageSegmentation = SegformerForSemanticSegmentation

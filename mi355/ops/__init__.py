@@ -300,3 +300,10 @@ def rel_position_bias(buckets, weight):
         return _RelBias.apply(buckets, weight)
     vals = weight[buckets.long()]  # (Sq, Sk, H)
     return vals.permute(2, 0, 1).unsqueeze(0)
+
+
+def seg_postprocess(logits, size):
+    """Fused bilinear-upsample + channel argmax."""
+    if _use_hip(logits):
+        return _load_ext().seg_postprocess(logits, int(size[0]), int(size[1]))
+    return torch_ref.seg_postprocess(logits, size)

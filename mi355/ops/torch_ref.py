@@ -118,3 +118,11 @@ def adamw_step(
 def softmax_bias(scores: torch.Tensor, bias: torch.Tensor | None) -> torch.Tensor:
     out = scores if bias is None else scores + bias
     return F.softmax(out.float(), dim=-1).to(scores.dtype)
+
+
+def seg_postprocess(logits: torch.Tensor, size) -> torch.Tensor:
+    """Bilinear-upsample class logits to ``size`` and argmax over channels
+    (SURVEY.md K17 — reference post_process_semantic_segmentation)."""
+    up = F.interpolate(logits.float(), size=size, mode="bilinear",
+                       align_corners=False)
+    return up.argmax(dim=1).to(torch.int32)
