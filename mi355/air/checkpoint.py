@@ -6,9 +6,10 @@ from __future__ import annotations
 
 import json
 import os
-import pickle
 import shutil
 import tempfile
+
+import cloudpickle as pickle
 
 
 class Checkpoint:

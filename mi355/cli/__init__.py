@@ -1,0 +1,89 @@
+"""mi355 job CLI (reference anyscale job submit +
+flan-t5-batch-inference-job-setup.yml:1-7 — name / compute_config /
+cluster_env / entrypoint). `python -m mi355.cli job submit job.yml` runs the
+entrypoint under the local runtime with env + logging set up; `status`
+tails run metrics."""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import subprocess
+import sys
+import time
+
+
+def _load_yaml(path: str) -> dict:
+    import yaml
+
+    with open(path) as f:
+        return yaml.safe_load(f)
+
+
+def cmd_submit(args):
+    spec = _load_yaml(args.spec)
+    name = spec.get("name", os.path.basename(args.spec))
+    entrypoint = spec["entrypoint"]
+    env = dict(os.environ)
+    env.update({str(k): str(v) for k, v in (spec.get("env") or {}).items()})
+    env.setdefault("MI355_JOB_NAME", name)
+    workdir = spec.get("working_dir", os.path.dirname(os.path.abspath(args.spec)) or ".")
+    print(f"[mi355 job] name={name!r} entrypoint={entrypoint!r} cwd={workdir}")
+    t0 = time.time()
+    proc = subprocess.run(entrypoint, shell=True, cwd=workdir, env=env)
+    dt = time.time() - t0
+    status = "SUCCEEDED" if proc.returncode == 0 else f"FAILED(rc={proc.returncode})"
+    print(f"[mi355 job] {name}: {status} in {dt:.1f}s")
+    return proc.returncode
+
+
+def cmd_status(args):
+    base = os.path.expanduser(args.dir or "~/mi355_results")
+    if not os.path.isdir(base):
+        print(f"no results under {base}")
+        return 0
+    for run in sorted(os.listdir(base)):
+        run_dir = os.path.join(base, run)
+        metrics = os.path.join(run_dir, "metrics.jsonl")
+        line = ""
+        if os.path.exists(metrics):
+            with open(metrics) as f:
+                rows = f.readlines()
+            if rows:
+                line = rows[-1].strip()
+        ckpts = [d for d in os.listdir(run_dir) if d.startswith("checkpoint")]
+        print(f"{run}: {len(ckpts)} checkpoint(s) {line}")
+    return 0
+
+
+def cmd_gpus(_args):
+    try:
+        out = subprocess.run(["rocm-smi", "--showuse", "--showmemuse"],
+                             capture_output=True, text=True, timeout=20)
+        print(out.stdout or out.stderr)
+    except FileNotFoundError:
+        import torch
+
+        print(f"torch.cuda.device_count() = {torch.cuda.device_count()}")
+    return 0
+
+
+def main(argv=None):
+    ap = argparse.ArgumentParser(prog="mi355")
+    sub = ap.add_subparsers(dest="cmd", required=True)
+    job = sub.add_parser("job")
+    jobsub = job.add_subparsers(dest="jobcmd", required=True)
+    s = jobsub.add_parser("submit")
+    s.add_argument("spec")
+    s.set_defaults(fn=cmd_submit)
+    st = sub.add_parser("status")
+    st.add_argument("--dir", default=None)
+    st.set_defaults(fn=cmd_status)
+    g = sub.add_parser("gpus")
+    g.set_defaults(fn=cmd_gpus)
+    args = ap.parse_args(argv)
+    return args.fn(args)
+
+
+if __name__ == "__main__":
+    sys.exit(main())

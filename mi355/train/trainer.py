@@ -190,6 +190,11 @@ class DataParallelTrainer:
                 if rank == 0:
                     last_metrics = payload
                     metrics_rows.append(payload)
+                    # observability: JSONL metrics per run dir
+                    import json as _json
+
+                    with open(os.path.join(run_dir, "metrics.jsonl"), "a") as mf:
+                        mf.write(_json.dumps(payload, default=str) + "\n")
                     if ckpt_dir:
                         n_ckpt += 1
                         kept = os.path.join(run_dir, f"checkpoint_{n_ckpt:06d}")
