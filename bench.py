@@ -37,7 +37,7 @@ def parse_args():
     ap.add_argument("--batch", type=int, default=2, help="per-device batch (reference config)")
     ap.add_argument("--seq", type=int, default=512)
     ap.add_argument("--device", default=None, help="cpu override for local testing")
-    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp16", "fp32"])
     ap.add_argument("--graphs", default="auto", choices=["auto", "on", "off"],
                     help="hipGraph-capture the train step")
     ap.add_argument("--mode", default="train", choices=["train", "infer"])
@@ -174,7 +174,8 @@ def main():
         torch.cuda.set_device(device)
     else:
         device = torch.device("cpu")
-    dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+    dtype = {"bf16": torch.bfloat16, "fp16": torch.float16,
+             "fp32": torch.float32}[args.dtype]
 
     if args.mode == "train":
         elapsed, extra = run_train(args, device, dtype)

@@ -87,7 +87,9 @@ def _worker_equiv(rank, world, port, q):
     out.loss.backward()
     replica.finish_grad_sync()
     grad = torch.cat([g.flat_grad for _, g in replica.groups]).clone()
-    q.put((rank, grad))
+    # plain numpy copy: torch tensors over mp.Queue use fd-based shm reduction
+    # that races with child exit
+    q.put((rank, grad.numpy().copy()))
     import torch.distributed as dist
     dist.destroy_process_group()
 
@@ -99,7 +101,7 @@ def test_ddp_grad_equals_full_batch_grad():
     procs = [ctx.Process(target=_worker_equiv, args=(r, 2, 29532, q)) for r in range(2)]
     for p in procs:
         p.start()
-    res = {r: g for r, g in [q.get(timeout=240) for _ in range(2)]}
+    res = {r: torch.from_numpy(g) for r, g in [q.get(timeout=240) for _ in range(2)]}
     for p in procs:
         p.join(timeout=60)
 
