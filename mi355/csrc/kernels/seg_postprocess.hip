@@ -21,15 +21,16 @@ __global__ void seg_post_kernel(const T* __restrict__ logits, int* __restrict__ 
   const int y = (idx / Wo) % Ho;
   const int b = idx / ((long)Ho * Wo);
 
-  // align_corners=False mapping
-  const float sy = ((y + 0.5f) * Hi) / Ho - 0.5f;
-  const float sx = ((x + 0.5f) * Wi) / Wo - 0.5f;
-  const int y0 = max(0, min(Hi - 1, (int)floorf(sy)));
-  const int x0 = max(0, min(Wi - 1, (int)floorf(sx)));
+  // align_corners=False mapping, negative coords clamped to 0 exactly like
+  // torch's area_pixel_compute_source_index
+  const float sy = fmaxf(((y + 0.5f) * Hi) / Ho - 0.5f, 0.f);
+  const float sx = fmaxf(((x + 0.5f) * Wi) / Wo - 0.5f, 0.f);
+  const int y0 = min(Hi - 1, (int)sy);
+  const int x0 = min(Wi - 1, (int)sx);
   const int y1 = min(Hi - 1, y0 + 1);
   const int x1 = min(Wi - 1, x0 + 1);
-  const float fy = fminf(fmaxf(sy - floorf(sy), 0.f), 1.f);
-  const float fx = fminf(fmaxf(sx - floorf(sx), 0.f), 1.f);
+  const float fy = sy - y0;
+  const float fx = sx - x0;
   const float w00 = (1 - fy) * (1 - fx), w01 = (1 - fy) * fx;
   const float w10 = fy * (1 - fx), w11 = fy * fx;
 

@@ -178,7 +178,9 @@ def attention(q, k, v, bias=None, causal=False, scale: float = 1.0, kpad=None):
     """kpad: optional (B, Sk) ADDITIVE fp32 key-padding mask (0 keep / -1e9
     drop), kept separate from ``bias`` so a (1,H,Sq,Sk) relative-position bias
     never has to be materialized per batch element."""
-    if _use_hip(q) and scale == 1.0:
+    hip_ok = (q.dtype in (torch.bfloat16, torch.float16)
+              and q.shape[-1] in (32, 64) and scale == 1.0)
+    if hip_ok and _use_hip(q):
         return _Attention.apply(q, k, v, bias, causal, kpad)
     return torch_ref.attention(q, k, v, bias=bias, causal=causal, scale=scale, kpad=kpad)
 
@@ -241,7 +243,7 @@ def adamw_step(params_f32, grads, exp_avgs, exp_avg_sqs, *, lr, beta1, beta2,
 def decode_attention(q, k_cache, v_cache, cache_len: int, bias=None):
     """q: (B,H,1,D); caches: (B,H,Tmax,D) with the first ``cache_len`` rows
     valid; bias broadcastable to (B,H,1,cache_len). SURVEY.md K11."""
-    if _use_hip(q):
+    if q.shape[-1] in (32, 64) and _use_hip(q):
         return _load_ext().decode_attn(q, k_cache, v_cache, cache_len,
                                        bias.contiguous() if bias is not None else None)
     k = k_cache[:, :, :cache_len]
