@@ -57,6 +57,53 @@ def relative_position_bucket(rel_pos: torch.Tensor, bidirectional: bool,
     return ret + torch.where(is_small, n, val_large)
 
 
+class _FusedLinears(nn.Module):
+    """N same-shaped bias-free projections stored as ONE fused weight so the
+    forward is a single hipBLASLt GEMM (q/k/v -> 1 GEMM, wi_0/wi_1 -> 1 GEMM:
+    the small-M projection GEMMs at B=2 are launch/M-bound).
+    state_dict stays HF-compatible: each sub-projection saves/loads under its
+    own key as a slice view."""
+
+    def __init__(self, names, d_in, d_out):
+        super().__init__()
+        self.names = list(names)
+        self.d_out = d_out
+        self.weight = nn.Parameter(torch.empty(len(self.names) * d_out, d_in))
+
+    def forward(self, x):
+        y = F.linear(x, self.weight)
+        return y.split(self.d_out, dim=-1)
+
+    def part(self, i):
+        return self.weight[i * self.d_out:(i + 1) * self.d_out]
+
+
+def _register_fused_statedict_compat(parent: nn.Module, attr: str):
+    """Keep HF key names for a _FusedLinears child: on save, split
+    ``<prefix><attr>.weight`` into ``<prefix><name>.weight`` slices; on load,
+    merge HF per-projection keys back into the fused weight."""
+    fused: _FusedLinears = getattr(parent, attr)
+    names = list(fused.names)
+
+    def save_hook(module, state_dict, prefix, local_metadata):
+        key = f"{prefix}{attr}.weight"
+        if key in state_dict:
+            w = state_dict.pop(key)
+            for i, n in enumerate(names):
+                state_dict[f"{prefix}{n}.weight"] =                     w[i * fused.d_out:(i + 1) * fused.d_out]
+        return state_dict
+
+    def load_pre_hook(state_dict, prefix, local_metadata, strict, missing_keys,
+                      unexpected_keys, error_msgs):
+        part_keys = [f"{prefix}{n}.weight" for n in names]
+        if all(k in state_dict for k in part_keys):
+            state_dict[f"{prefix}{attr}.weight"] = torch.cat(
+                [state_dict.pop(k) for k in part_keys], dim=0)
+
+    parent._register_state_dict_hook(save_hook)
+    parent._register_load_state_dict_pre_hook(load_pre_hook)
+
+
 class T5Attention(nn.Module):
     def __init__(self, config: T5Config, has_relative_attention_bias: bool, is_decoder: bool):
         super().__init__()
@@ -68,13 +115,22 @@ class T5Attention(nn.Module):
         self.d_kv = config.d_kv
         self.n_heads = config.num_heads
         self.inner_dim = self.n_heads * self.d_kv
-        self.q = nn.Linear(self.d_model, self.inner_dim, bias=False)
-        self.k = nn.Linear(self.d_model, self.inner_dim, bias=False)
-        self.v = nn.Linear(self.d_model, self.inner_dim, bias=False)
+        self.qkv = _FusedLinears(["q", "k", "v"], self.d_model, self.inner_dim)
+        _register_fused_statedict_compat(self, "qkv")
         self.o = nn.Linear(self.inner_dim, self.d_model, bias=False)
         if has_relative_attention_bias:
             self.relative_attention_bias = nn.Embedding(self.num_buckets, self.n_heads)
         self.dropout = config.dropout_rate
+
+    # q/k/v as single projections (cross-attention + generate cache paths)
+    def q(self, x):
+        return F.linear(x, self.qkv.part(0))
+
+    def k(self, x):
+        return F.linear(x, self.qkv.part(1))
+
+    def v(self, x):
+        return F.linear(x, self.qkv.part(2))
 
     def compute_bias(self, sq: int, sk: int, device) -> torch.Tensor:
         """(1, H, sq, sk) additive relative-position bias."""
@@ -105,8 +161,17 @@ class T5Attention(nn.Module):
         {"k","v": (B,H,Tmax,D) preallocated, "len": int} for self-attn append,
         or {"k","v" precomputed static} for cross-attn."""
         b, sq, _ = hidden.shape
-        qs = self._shape(self.q(hidden), b)
-        if kv_cache is not None and kv_cache.get("static", False):
+        fused_self = kv_hidden is None and kv_cache is None
+        if fused_self:
+            q3, k3, v3 = self.qkv(hidden)
+            qs = self._shape(q3, b)
+            ks = self._shape(k3, b)
+            vs = self._shape(v3, b)
+        else:
+            qs = self._shape(self.q(hidden), b)
+        if fused_self:
+            pass
+        elif kv_cache is not None and kv_cache.get("static", False):
             ks, vs = kv_cache["k"], kv_cache["v"]
         else:
             src = hidden if kv_hidden is None else kv_hidden
@@ -158,13 +223,14 @@ class T5LayerCrossAttention(nn.Module):
 class T5DenseGatedActDense(nn.Module):
     def __init__(self, config):
         super().__init__()
-        self.wi_0 = nn.Linear(config.d_model, config.d_ff, bias=False)
-        self.wi_1 = nn.Linear(config.d_model, config.d_ff, bias=False)
+        self.wi = _FusedLinears(["wi_0", "wi_1"], config.d_model, config.d_ff)
+        _register_fused_statedict_compat(self, "wi")
         self.wo = nn.Linear(config.d_ff, config.d_model, bias=False)
         self.dropout = nn.Dropout(config.dropout_rate)
 
     def forward(self, x):
-        h = ops.geglu(self.wi_0(x), self.wi_1(x))
+        gate, up = self.wi(x)
+        h = ops.geglu(gate, up)
         return self.wo(self.dropout(h))
 
 
@@ -281,15 +347,14 @@ class T5ForConditionalGeneration(nn.Module):
             if isinstance(m, T5LayerNorm):
                 m.weight.data.fill_(f * 1.0)
             elif isinstance(m, T5Attention):
-                m.q.weight.data.normal_(0.0, f * (d_model * d_kv) ** -0.5)
-                m.k.weight.data.normal_(0.0, f * d_model**-0.5)
-                m.v.weight.data.normal_(0.0, f * d_model**-0.5)
+                m.qkv.part(0).data.normal_(0.0, f * (d_model * d_kv) ** -0.5)
+                m.qkv.part(1).data.normal_(0.0, f * d_model**-0.5)
+                m.qkv.part(2).data.normal_(0.0, f * d_model**-0.5)
                 m.o.weight.data.normal_(0.0, f * (n_heads * d_kv) ** -0.5)
                 if m.has_relative_attention_bias:
                     m.relative_attention_bias.weight.data.normal_(0.0, f * d_model**-0.5)
             elif isinstance(m, T5DenseGatedActDense):
-                m.wi_0.weight.data.normal_(0.0, f * d_model**-0.5)
-                m.wi_1.weight.data.normal_(0.0, f * d_model**-0.5)
+                m.wi.weight.data.normal_(0.0, f * d_model**-0.5)
                 m.wo.weight.data.normal_(0.0, f * d_ff**-0.5)
 
     # -- training / scoring forward ----------------------------------------- #
