@@ -161,22 +161,13 @@ class T5Attention(nn.Module):
         {"k","v": (B,H,Tmax,D) preallocated, "len": int} for self-attn append,
         or {"k","v" precomputed static} for cross-attn."""
         b, sq, _ = hidden.shape
-        fused_self = kv_hidden is None and kv_cache is None
-        if fused_self:
+        static = kv_cache is not None and kv_cache.get("static", False)
+        if kv_hidden is None and not static:
+            # self-attention (train or decode): ONE fused qkv GEMM
             q3, k3, v3 = self.qkv(hidden)
             qs = self._shape(q3, b)
             ks = self._shape(k3, b)
             vs = self._shape(v3, b)
-        else:
-            qs = self._shape(self.q(hidden), b)
-        if fused_self:
-            pass
-        elif kv_cache is not None and kv_cache.get("static", False):
-            ks, vs = kv_cache["k"], kv_cache["v"]
-        else:
-            src = hidden if kv_hidden is None else kv_hidden
-            ks = self._shape(self.k(src), b)
-            vs = self._shape(self.v(src), b)
             if kv_cache is not None:
                 t0 = kv_cache["len"]
                 kv_cache["k"][:, :, t0:t0 + sq] = ks
@@ -184,6 +175,13 @@ class T5Attention(nn.Module):
                 kv_cache["len"] = t0 + sq
                 ks = kv_cache["k"][:, :, : t0 + sq]
                 vs = kv_cache["v"][:, :, : t0 + sq]
+        elif static:
+            qs = self._shape(self.q(hidden), b)
+            ks, vs = kv_cache["k"], kv_cache["v"]
+        else:
+            qs = self._shape(self.q(hidden), b)
+            ks = self._shape(self.k(kv_hidden), b)
+            vs = self._shape(self.v(kv_hidden), b)
         if kv_cache is not None and sq == 1 and not kv_cache.get("static", False):
             out = ops.decode_attention(qs, kv_cache["k"], kv_cache["v"],
                                        kv_cache["len"], bias=position_bias)
