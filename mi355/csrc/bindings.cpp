@@ -15,7 +15,8 @@ void adamw_multi(std::vector<torch::Tensor> masters, std::vector<torch::Tensor> 
                  std::vector<torch::Tensor> lowps, double lr, double beta1, double beta2,
                  double eps, double weight_decay, long step, torch::Tensor grad_scale);
 torch::Tensor decode_attn(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
-                          long cache_len, c10::optional<torch::Tensor> bias);
+                          long cache_len, c10::optional<torch::Tensor> bias,
+                          c10::optional<torch::Tensor> kpad);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                     c10::optional<torch::Tensor> bias,
                                     c10::optional<torch::Tensor> kpad, bool causal);

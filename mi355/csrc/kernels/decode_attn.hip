@@ -18,6 +18,7 @@ struct DStr { long sb, sh, ss; };
 template <typename T, int D>
 __global__ void decode_attn_kernel(const T* __restrict__ q, const T* __restrict__ kc,
                                    const T* __restrict__ vc, const float* __restrict__ bias,
+                                   const float* __restrict__ kpad,
                                    T* __restrict__ out, int B, int H, int L, long t_max,
                                    DStr qs, DStr kst, DStr vst, DStr ost,
                                    int bias_b_stride /*0 when broadcast*/) {
@@ -58,7 +59,7 @@ __global__ void decode_attn_kernel(const T* __restrict__ q, const T* __restrict_
 #pragma unroll
         for (int u = 0; u < KV; ++u) acc += qreg[d + u] * DT<T>::to_f32(kv.data[u]);
       }
-      s = acc + (brow ? brow[j] : 0.f);
+      s = acc + (brow ? brow[j] : 0.f) + (kpad ? kpad[(long)b * L + j] : 0.f);
     }
     const float tile_max = wave_reduce_max(s);
     const float mn = fmaxf(m, tile_max);
@@ -95,7 +96,8 @@ __global__ void decode_attn_kernel(const T* __restrict__ q, const T* __restrict_
 }  // namespace
 
 torch::Tensor decode_attn(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
-                          long cache_len, c10::optional<torch::Tensor> bias) {
+                          long cache_len, c10::optional<torch::Tensor> bias,
+                          c10::optional<torch::Tensor> kpad) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 4 && q.size(2) == 1 && q.stride(3) == 1,
               "q must be (B,H,1,D), d contiguous");
   const int B = q.size(0), H = q.size(1), D = q.size(3);
@@ -107,8 +109,13 @@ torch::Tensor decode_attn(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   DStr ost{out.stride(0), out.stride(1), out.stride(2)};
   auto stream = at::hip::getCurrentHIPStream().stream();
   const float* bias_ptr = nullptr;
+  const float* kpad_ptr = nullptr;
   int bias_b_stride = 0;
   torch::Tensor bias_f;
+  if (kpad.has_value() && kpad->defined()) {
+    TORCH_CHECK(kpad->size(-1) == cache_len && kpad->scalar_type() == torch::kFloat);
+    kpad_ptr = kpad->data_ptr<float>();
+  }
   if (bias.has_value() && bias->defined()) {
     bias_f = bias->to(torch::kFloat).contiguous();
     TORCH_CHECK(bias_f.size(-1) == cache_len, "bias last dim must equal cache_len");
@@ -120,8 +127,8 @@ torch::Tensor decode_attn(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
 #define CASE(T, DD)                                                                   \
   hipLaunchKernelGGL((decode_attn_kernel<T, DD>), dim3(blocks), dim3(waves_per_block * WAVE), \
                      0, stream, (const T*)q.data_ptr(), (const T*)kc.data_ptr(),       \
-                     (const T*)vc.data_ptr(), bias_ptr, (T*)out.data_ptr(), B, H,      \
-                     (int)cache_len, t_max, qs, kst, vst, ost, bias_b_stride)
+                     (const T*)vc.data_ptr(), bias_ptr, kpad_ptr, (T*)out.data_ptr(),  \
+                     B, H, (int)cache_len, t_max, qs, kst, vst, ost, bias_b_stride)
   if (q.scalar_type() == torch::kBFloat16 && D == 64) CASE(__hip_bfloat16, 64);
   else if (q.scalar_type() == torch::kHalf && D == 64) CASE(__half, 64);
   else if (q.scalar_type() == torch::kFloat && D == 64) CASE(float, 64);

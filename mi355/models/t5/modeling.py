@@ -182,9 +182,13 @@ class T5Attention(nn.Module):
             qs = self._shape(self.q(hidden), b)
             ks = self._shape(self.k(kv_hidden), b)
             vs = self._shape(self.v(kv_hidden), b)
-        if kv_cache is not None and sq == 1 and not kv_cache.get("static", False):
+        if kv_cache is not None and sq == 1 and not static:
             out = ops.decode_attention(qs, kv_cache["k"], kv_cache["v"],
                                        kv_cache["len"], bias=position_bias)
+        elif static and sq == 1 and not torch.is_grad_enabled():
+            # single-token cross-attention over the full encoder KV
+            out = ops.decode_attention(qs, ks, vs, ks.shape[2],
+                                       bias=position_bias, kpad=kpad)
         else:
             out = ops.attention(qs, ks, vs, bias=position_bias, causal=causal, kpad=kpad)
         out = out.transpose(1, 2).reshape(b, sq, self.inner_dim)

@@ -240,15 +240,18 @@ def adamw_step(params_f32, grads, exp_avgs, exp_avg_sqs, *, lr, beta1, beta2,
 # --------------------------------------------------------------------------- #
 # Single-token decode attention over a KV cache (generate hot loop)           #
 # --------------------------------------------------------------------------- #
-def decode_attention(q, k_cache, v_cache, cache_len: int, bias=None):
+def decode_attention(q, k_cache, v_cache, cache_len: int, bias=None, kpad=None):
     """q: (B,H,1,D); caches: (B,H,Tmax,D) with the first ``cache_len`` rows
-    valid; bias broadcastable to (B,H,1,cache_len). SURVEY.md K11."""
-    if q.shape[-1] in (32, 64) and _use_hip(q):
-        return _load_ext().decode_attn(q, k_cache, v_cache, cache_len,
-                                       bias.contiguous() if bias is not None else None)
+    valid; bias broadcastable to (B,H,1,cache_len); kpad (B,cache_len)
+    additive fp32. SURVEY.md K11."""
+    if q.shape[-1] in (32, 64) and q.dtype in (torch.bfloat16, torch.float16)             and _use_hip(q):
+        return _load_ext().decode_attn(
+            q, k_cache, v_cache, cache_len,
+            bias.contiguous() if bias is not None else None,
+            kpad.contiguous().float() if kpad is not None else None)
     k = k_cache[:, :, :cache_len]
     v = v_cache[:, :, :cache_len]
-    return torch_ref.attention(q, k, v, bias=bias, causal=False)
+    return torch_ref.attention(q, k, v, bias=bias, causal=False, kpad=kpad)
 
 
 # --------------------------------------------------------------------------- #
