@@ -4,7 +4,9 @@
 
 std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps);
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
-                                       torch::Tensor inv);
+                                       torch::Tensor inv, c10::optional<torch::Tensor> dextra);
+std::vector<torch::Tensor> add_rmsnorm_fwd(torch::Tensor x, torch::Tensor r,
+                                           torch::Tensor w, double eps);
 torch::Tensor geglu_fwd(torch::Tensor gate, torch::Tensor up);
 std::vector<torch::Tensor> geglu_bwd(torch::Tensor dout, torch::Tensor gate, torch::Tensor up);
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels, long ignore_index);
@@ -35,6 +37,7 @@ torch::Tensor seg_postprocess(torch::Tensor logits, long Ho, long Wo);
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("add_rmsnorm_fwd", &add_rmsnorm_fwd);
   m.def("geglu_fwd", &geglu_fwd);
   m.def("geglu_bwd", &geglu_bwd);
   m.def("ce_fwd", &ce_fwd);

@@ -42,9 +42,9 @@ __global__ void decode_attn_kernel(const T* __restrict__ q, const T* __restrict_
   const int wid = threadIdx.x / WAVE;
 
   float m = -INFINITY, l = 0.f;
-  float o[D / WAVE + 1];
+  float oacc[8];
 #pragma unroll
-  for (int j = 0; j < D / WAVE + 1; ++j) o[j] = 0.f;
+  for (int j = 0; j < 8; ++j) oacc[j] = 0.f;
 
   for (int j0 = 0; j0 < L; j0 += WAVE) {
     const int j = j0 + lane;
@@ -67,29 +67,42 @@ __global__ void decode_attn_kernel(const T* __restrict__ q, const T* __restrict_
     const float p = (j < L) ? __expf(s - mn) : 0.f;
     p_lds[wid][lane] = p;
     l = l * alpha + wave_reduce_sum(p);
-    // o *= alpha; o += sum_j p_j * V[j][lane-th d]
+    // o accumulate, 8 keys x 8 dims per lane per pass: lane handles key
+    // j0 + 8*pass + (lane>>3), dims 8*(lane&7)..+8 — each instruction moves
+    // a full 1 KiB per wave instead of 128 B (the old d=lane layout)
+    constexpr int DCH = 8;                 // dims per lane chunk
+    static_assert(D % DCH == 0, "");
+    const int dgrp = lane & 7;             // dim chunk id (D=64: 8 chunks)
+    const int krow = lane >> 3;            // 8 keys in flight
 #pragma unroll
-    for (int du = 0; du < (D + WAVE - 1) / WAVE; ++du) {
-      int d = du * WAVE + lane;
-      if (d < D) o[du] *= alpha;
-    }
+    for (int u = 0; u < DCH; ++u) oacc[u] *= alpha;
     const int jmax = min(WAVE, L - j0);
-    for (int jj = 0; jj < jmax; ++jj) {
+    for (int jj = krow; jj < jmax; jj += 8) {
       const float pj = p_lds[wid][jj];
-      const T* vrow = vbase + (long)(j0 + jj) * vst.ss;
+      if (dgrp * DCH < D) {
+        const T* vrow = vbase + (long)(j0 + jj) * vst.ss + dgrp * DCH;
+        VecT<T, 8> vv = vload<T, 8>(vrow);
 #pragma unroll
-      for (int du = 0; du < (D + WAVE - 1) / WAVE; ++du) {
-        int d = du * WAVE + lane;
-        if (d < D) o[du] += pj * DT<T>::to_f32(vrow[d]);
+        for (int u = 0; u < DCH; ++u) oacc[u] += pj * DT<T>::to_f32(vv.data[u]);
       }
     }
     m = mn;
   }
-  const float invl = (l > 0.f) ? 1.f / l : 0.f;
+  // fold the 8 key-groups: lanes l and l^8,l^16,l^32 share the same dim chunk
 #pragma unroll
-  for (int du = 0; du < (D + WAVE - 1) / WAVE; ++du) {
-    int d = du * WAVE + lane;
-    if (d < D) out[(long)b * ost.sb + (long)h * ost.sh + d] = DT<T>::from_f32(o[du] * invl);
+  for (int u = 0; u < 8; ++u) {
+#pragma unroll
+    for (int off = 8; off < WAVE; off <<= 1) oacc[u] += __shfl_xor(oacc[u], off, WAVE);
+  }
+  const float invl = (l > 0.f) ? 1.f / l : 0.f;
+  if ((lane >> 3) == 0) {  // lanes 0..7 hold the final chunks
+    const int d0 = (lane & 7) * 8;
+    if (d0 < D) {
+      VecT<T, 8> ov;
+#pragma unroll
+      for (int u = 0; u < 8; ++u) ov.data[u] = DT<T>::from_f32(oacc[u] * invl);
+      vstore<T, 8>(out + (long)b * ost.sb + (long)h * ost.sh + d0, ov);
+    }
   }
 }
 

@@ -46,10 +46,54 @@ __global__ void rmsnorm_fwd_kernel(const T* __restrict__ x, const T* __restrict_
   }
 }
 
+// fused residual-add + RMSNorm: s = x + r; y = rmsnorm(s) (pre-norm T5
+// chains "resid += delta; norm(resid)" — one kernel instead of add + norm)
+template <typename T, int VEC>
+__global__ void add_rmsnorm_fwd_kernel(const T* __restrict__ x, const T* __restrict__ r,
+                                       T* __restrict__ s_out, T* __restrict__ y,
+                                       const T* __restrict__ w, float* __restrict__ inv_rms,
+                                       int rows, int D, float eps) {
+  __shared__ float red[16];
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* xr = x + (long)row * D;
+    const T* rr = r + (long)row * D;
+    T* sr = s_out + (long)row * D;
+    T* yr = y + (long)row * D;
+    float ss = 0.f;
+    for (int i = threadIdx.x * VEC; i < D; i += blockDim.x * VEC) {
+      VecT<T, VEC> xv = vload<T, VEC>(xr + i);
+      VecT<T, VEC> rv = vload<T, VEC>(rr + i);
+      VecT<T, VEC> sv;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float f = DT<T>::to_f32(xv.data[j]) + DT<T>::to_f32(rv.data[j]);
+        sv.data[j] = DT<T>::from_f32(f);
+        ss += f * f;
+      }
+      vstore<T, VEC>(sr + i, sv);
+    }
+    ss = block_reduce_sum(ss, red);
+    float inv = rsqrtf(ss / D + eps);
+    if (threadIdx.x == 0) inv_rms[row] = inv;
+    for (int i = threadIdx.x * VEC; i < D; i += blockDim.x * VEC) {
+      VecT<T, VEC> sv = vload<T, VEC>(sr + i);
+      VecT<T, VEC> wv = vload<T, VEC>(w + i);
+      VecT<T, VEC> o;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        o.data[j] = DT<T>::from_f32(DT<T>::to_f32(sv.data[j]) * inv *
+                                    DT<T>::to_f32(wv.data[j]));
+      vstore<T, VEC>(yr + i, o);
+    }
+    __syncthreads();
+  }
+}
+
 // dx = inv * w * dy - x * inv^3/D * sum(dy*w*x) ; dw_part[blk] = sum_rows dy*x*inv
 template <typename T, int VEC, int MAXSLOT>
 __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                                    const T* __restrict__ w, const float* __restrict__ inv_rms,
+                                   const T* __restrict__ dextra /*nullable*/,
                                    T* __restrict__ dx, float* __restrict__ dw_part,
                                    int rows, int D) {
   __shared__ float red[16];
@@ -80,12 +124,15 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy, const T* __restrict
       VecT<T, VEC> dyv = vload<T, VEC>(dyr + i);
       VecT<T, VEC> wv = vload<T, VEC>(w + i);
       VecT<T, VEC> o;
+      VecT<T, VEC> ev;
+      if (dextra) ev = vload<T, VEC>(dextra + (long)row * D + i);
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
         float xf = DT<T>::to_f32(xv.data[j]);
         float dyf = DT<T>::to_f32(dyv.data[j]);
         float wf = DT<T>::to_f32(wv.data[j]);
-        o.data[j] = DT<T>::from_f32(inv * wf * dyf - xf * c);
+        float extra = dextra ? DT<T>::to_f32(ev.data[j]) : 0.f;
+        o.data[j] = DT<T>::from_f32(inv * wf * dyf - xf * c + extra);
         dwacc[slot * VEC + j] += dyf * xf * inv;
       }
       vstore<T, VEC>(dxr + i, o);
@@ -131,8 +178,8 @@ void launch_fwd(const torch::Tensor& x, const torch::Tensor& w, torch::Tensor& y
 
 template <typename T>
 void launch_bwd(const torch::Tensor& dy, const torch::Tensor& x, const torch::Tensor& w,
-                const torch::Tensor& inv, torch::Tensor& dx, torch::Tensor& dw,
-                hipStream_t stream) {
+                const torch::Tensor& inv, const torch::Tensor* dextra,
+                torch::Tensor& dx, torch::Tensor& dw, hipStream_t stream) {
   const int rows = x.size(0), D = x.size(1);
   constexpr int VEC = sizeof(T) == 4 ? 4 : 8;
   const int threads = (D / VEC >= 256) ? 256 : 128;
@@ -141,8 +188,9 @@ void launch_bwd(const torch::Tensor& dy, const torch::Tensor& x, const torch::Te
   auto part = torch::empty({grid, D}, dw.options());
   hipLaunchKernelGGL((rmsnorm_bwd_kernel<T, VEC, 4>), dim3(grid), dim3(threads), 0, stream,
                      (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (const T*)w.data_ptr(),
-                     inv.data_ptr<float>(), (T*)dx.data_ptr(), part.data_ptr<float>(),
-                     rows, D);
+                     inv.data_ptr<float>(),
+                     dextra ? (const T*)dextra->data_ptr() : nullptr,
+                     (T*)dx.data_ptr(), part.data_ptr<float>(), rows, D);
   HIP_CHECK_KERNEL();
   hipLaunchKernelGGL(colsum_kernel, dim3((D + 255) / 256, 16), dim3(256), 0, stream,
                      part.data_ptr<float>(), dw.data_ptr<float>(), grid, D);
@@ -168,17 +216,42 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double 
 }
 
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
-                                       torch::Tensor inv) {
+                                       torch::Tensor inv,
+                                       c10::optional<torch::Tensor> dextra) {
   auto dx = torch::empty_like(x);
   auto dw = torch::zeros({x.size(1)}, x.options().dtype(torch::kFloat));
   auto stream = at::hip::getCurrentHIPStream().stream();
+  const torch::Tensor* de = (dextra.has_value() && dextra->defined()) ? &*dextra : nullptr;
   if (x.scalar_type() == torch::kBFloat16)
-    launch_bwd<__hip_bfloat16>(dy, x, w, inv, dx, dw, stream);
+    launch_bwd<__hip_bfloat16>(dy, x, w, inv, de, dx, dw, stream);
   else if (x.scalar_type() == torch::kHalf)
-    launch_bwd<__half>(dy, x, w, inv, dx, dw, stream);
+    launch_bwd<__half>(dy, x, w, inv, de, dx, dw, stream);
   else if (x.scalar_type() == torch::kFloat)
-    launch_bwd<float>(dy, x, w, inv, dx, dw, stream);
+    launch_bwd<float>(dy, x, w, inv, de, dx, dw, stream);
   else
     TORCH_CHECK(false, "rmsnorm: unsupported dtype");
   return {dx, dw};
+}
+
+std::vector<torch::Tensor> add_rmsnorm_fwd(torch::Tensor x, torch::Tensor r,
+                                           torch::Tensor w, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous() && r.is_contiguous());
+  auto s_out = torch::empty_like(x);
+  auto y = torch::empty_like(x);
+  auto inv = torch::empty({x.size(0)}, x.options().dtype(torch::kFloat));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  const int rows = x.size(0), D = x.size(1);
+#define ACASE(T, VEC)                                                                  \
+  hipLaunchKernelGGL((add_rmsnorm_fwd_kernel<T, VEC>),                                 \
+                     dim3(std::min(rows, 2048)), dim3((D / VEC >= 256) ? 256 : 128),   \
+                     0, stream, (const T*)x.data_ptr(), (const T*)r.data_ptr(),        \
+                     (T*)s_out.data_ptr(), (T*)y.data_ptr(), (const T*)w.data_ptr(),   \
+                     inv.data_ptr<float>(), rows, D, (float)eps)
+  if (x.scalar_type() == torch::kBFloat16) ACASE(__hip_bfloat16, 8);
+  else if (x.scalar_type() == torch::kHalf) ACASE(__half, 8);
+  else if (x.scalar_type() == torch::kFloat) ACASE(float, 4);
+  else TORCH_CHECK(false, "add_rmsnorm: unsupported dtype");
+#undef ACASE
+  HIP_CHECK_KERNEL();
+  return {s_out, y, inv};
 }

@@ -285,8 +285,7 @@ class T5Stack(nn.Module):
                 enc_kpad=None, caches=None, past_len: int = 0):
         if inputs_embeds is None:
             inputs_embeds = ops.token_embedding(input_ids, self.embed_tokens.weight)
-        x = inputs_embeds
-        x = self.dropout(x)
+        x = self.dropout(inputs_embeds)
         sq = x.shape[1]
         sk = past_len + sq
         # The relative-position bias is computed once by block 0 and shared by
@@ -295,13 +294,34 @@ class T5Stack(nn.Module):
         position_bias = attn0.compute_bias(sk, sk, x.device)[:, :, past_len:].to(x.dtype) \
             if attn0.has_relative_attention_bias else None
         causal = self.is_decoder and sq > 1
+        # pre-norm chain with fused residual-add + RMSNorm (SURVEY.md K2+K8):
+        # resid/delta threading — each sublayer's input norm absorbs the
+        # previous sublayer's residual add in one kernel
+        resid, delta = x, None
+
+        def add_norm(ln, resid, delta):
+            if delta is None:
+                return resid, ln(resid)
+            return ops.add_rms_norm(resid, delta, ln.weight, ln.variance_epsilon)
+
         for i, blk in enumerate(self.block):
             self_cache = caches[i]["self"] if caches is not None else None
             cross_cache = caches[i]["cross"] if (caches is not None and self.is_decoder) else None
-            x = blk(x, position_bias=position_bias, kpad=kpad, enc_hidden=enc_hidden,
-                    enc_kpad=enc_kpad, causal=causal, self_cache=self_cache,
-                    cross_cache=cross_cache)
-        return self.dropout(self.final_layer_norm(x))
+            sa = blk.layer[0]
+            resid, normed = add_norm(sa.layer_norm, resid, delta)
+            delta = sa.dropout(sa.SelfAttention(
+                normed, position_bias=position_bias, kpad=kpad, causal=causal,
+                kv_cache=self_cache))
+            if self.is_decoder:
+                ca = blk.layer[1]
+                resid, normed = add_norm(ca.layer_norm, resid, delta)
+                delta = ca.dropout(ca.EncDecAttention(
+                    normed, kv_hidden=enc_hidden, kpad=enc_kpad, kv_cache=cross_cache))
+            ff = blk.layer[-1]
+            resid, normed = add_norm(ff.layer_norm, resid, delta)
+            delta = ff.dropout(ff.DenseReluDense(normed))
+        _, normed = add_norm(self.final_layer_norm, resid, delta)
+        return self.dropout(normed)
 
 
 @dataclass
@@ -437,10 +457,26 @@ class T5ForConditionalGeneration(nn.Module):
     def _decode_step(self, tokens, enc, kpad, caches, t, bias_t):
         x = self.decoder.dropout(
             ops.token_embedding(tokens, self.decoder.embed_tokens.weight))
-        for i, blk in enumerate(self.decoder.block):
-            x = blk(x, position_bias=bias_t, enc_hidden=enc, enc_kpad=kpad, causal=False,
-                    self_cache=caches[i]["self"], cross_cache=caches[i]["cross"])
-        return self.decoder.final_layer_norm(x)
+        dec = self.decoder
+        resid, delta = x, None
+
+        def add_norm(ln, resid, delta):
+            if delta is None:
+                return resid, ln(resid)
+            return ops.add_rms_norm(resid, delta, ln.weight, ln.variance_epsilon)
+
+        for i, blk in enumerate(dec.block):
+            sa, ca, ff = blk.layer[0], blk.layer[1], blk.layer[2]
+            resid, normed = add_norm(sa.layer_norm, resid, delta)
+            delta = sa.SelfAttention(normed, position_bias=bias_t,
+                                     kv_cache=caches[i]["self"])
+            resid, normed = add_norm(ca.layer_norm, resid, delta)
+            delta = ca.EncDecAttention(normed, kv_hidden=enc, kpad=kpad,
+                                       kv_cache=caches[i]["cross"])
+            resid, normed = add_norm(ff.layer_norm, resid, delta)
+            delta = ff.DenseReluDense(normed)
+        _, normed = add_norm(dec.final_layer_norm, resid, delta)
+        return normed
 
     # -- HF-layout checkpoint I/O --------------------------- #
     def save_pretrained(self, save_dir: str, safe_serialization: bool = True):

@@ -69,7 +69,8 @@ class _RMSNorm(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x2d, weight, inv_rms = ctx.saved_tensors
-        dx, dw = _load_ext().rmsnorm_bwd(dy.contiguous().view_as(x2d), x2d, weight, inv_rms)
+        dx, dw = _load_ext().rmsnorm_bwd(dy.contiguous().view_as(x2d), x2d, weight,
+                                         inv_rms, None)
         return dx.view_as(dy), dw.to(weight.dtype), None
 
 
@@ -77,6 +78,38 @@ def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6) -> torch.
     if _use_hip(x):
         return _RMSNorm.apply(x, weight, eps)
     return torch_ref.rms_norm(x, weight, eps)
+
+
+class _AddRMSNorm(torch.autograd.Function):
+    """(s, y) = (x + r, rmsnorm(x + r) * w) — the pre-norm chain fused
+    (SURVEY.md K2+K8). backward receives grads for BOTH outputs: the
+    residual stream grad (ds) and the normed-branch grad (dy)."""
+
+    @staticmethod
+    def forward(ctx, x, r, weight, eps):
+        x2d = x.contiguous().view(-1, x.shape[-1])
+        r2d = r.contiguous().view(-1, x.shape[-1])
+        s, y, inv = _load_ext().add_rmsnorm_fwd(x2d, r2d, weight, eps)
+        ctx.save_for_backward(s, weight, inv)
+        ctx.shape = x.shape
+        return s.view(x.shape), y.view(x.shape)
+
+    @staticmethod
+    def backward(ctx, ds, dy):
+        s, weight, inv = ctx.saved_tensors
+        ds2d = ds.contiguous().view_as(s) if ds is not None else None
+        dx, dw = _load_ext().rmsnorm_bwd(dy.contiguous().view_as(s), s, weight,
+                                         inv, ds2d)
+        dx = dx.view(ctx.shape)
+        return dx, dx, dw.to(weight.dtype), None
+
+
+def add_rms_norm(x, r, weight, eps: float = 1e-6):
+    """Returns (s, y): s = x + r (the new residual stream), y = rmsnorm(s)."""
+    if _use_hip(x):
+        return _AddRMSNorm.apply(x, r, weight, eps)
+    s = x + r
+    return s, torch_ref.rms_norm(s, weight, eps)
 
 
 # --------------------------------------------------------------------------- #

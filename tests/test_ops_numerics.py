@@ -69,6 +69,47 @@ def test_rmsnorm_fwd_bwd(dtype, shape):
 
 
 @gpu
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_add_rms_norm_fwd_bwd(dtype):
+    torch.manual_seed(7)
+    tol = 1e-5 if dtype == torch.float32 else 2e-2
+    x = torch.randn(256, 768, device="cuda", dtype=dtype, requires_grad=True)
+    r = torch.randn(256, 768, device="cuda", dtype=dtype, requires_grad=True)
+    w = torch.randn(768, device="cuda", dtype=dtype, requires_grad=True)
+    s, y = ops.add_rms_norm(x, r, w, 1e-6)
+    xr = x.detach().float().requires_grad_(True)
+    rr = r.detach().float().requires_grad_(True)
+    wr = w.detach().float().requires_grad_(True)
+    sr = xr + rr
+    yr = torch_ref.rms_norm(sr, wr, 1e-6)
+    assert _relerr(s, sr) < tol and _relerr(y, yr) < tol
+    ds = torch.randn_like(yr)
+    dy = torch.randn_like(yr)
+    (s.float() * ds + y.float() * dy).sum().backward()
+    (sr * ds + yr * dy).sum().backward()
+    assert _relerr(x.grad, xr.grad) < tol
+    assert _relerr(r.grad, rr.grad) < tol
+    assert _relerr(w.grad, wr.grad) < tol * 4
+
+
+@gpu
+def test_decode_attn_kpad():
+    torch.manual_seed(8)
+    B, H, L, D = 3, 4, 96, 64
+    q = torch.randn(B, H, 1, D, device="cuda", dtype=torch.bfloat16)
+    kc = torch.randn(B, H, 128, D, device="cuda", dtype=torch.bfloat16)
+    vc = torch.randn(B, H, 128, D, device="cuda", dtype=torch.bfloat16)
+    keep = torch.ones(B, L, device="cuda")
+    keep[:, L - 20:] = 0
+    kpad = (1 - keep) * -1e9
+    with torch.no_grad():
+        o = ops.decode_attention(q, kc, vc, L, kpad=kpad)
+        o_ref = torch_ref.attention(q.float(), kc[:, :, :L].float(),
+                                    vc[:, :, :L].float(), kpad=kpad)
+    assert _relerr(o, o_ref) < 3e-2
+
+
+@gpu
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
 def test_geglu_fwd_bwd(dtype):
     torch.manual_seed(1)
