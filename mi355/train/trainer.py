@@ -33,9 +33,32 @@ def _free_port() -> int:
     return port
 
 
+class _PrefixedOut:
+    """Worker stdout/stderr tagged with the rank (SURVEY.md C25 — the
+    reference surfaces worker prints at the driver via Ray log redirection)."""
+
+    def __init__(self, inner, tag):
+        self.inner, self.tag, self._nl = inner, tag, True
+
+    def write(self, text):
+        for chunk in text.splitlines(keepends=True):
+            if self._nl:
+                self.inner.write(self.tag)
+            self.inner.write(chunk)
+            self._nl = chunk.endswith("\n")
+
+    def flush(self):
+        self.inner.flush()
+
+
 def _worker_entry(rank, world, port, use_gpu, loop_bytes, loop_config,
                   shard_tables, report_q, control_q, resume_dir, trial_name):
     try:
+        import sys
+
+        tag = f"({trial_name + ' ' if trial_name else ''}rank={rank}) "
+        sys.stdout = _PrefixedOut(sys.stdout, tag)
+        sys.stderr = _PrefixedOut(sys.stderr, tag)
         os.environ.update({
             "RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": str(world),
             "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
@@ -171,9 +194,27 @@ class DataParallelTrainer:
         report_seq = {r: 0 for r in range(world)}
         decisions = []
         waiting = {}
+        import queue as _q
+
+        last_activity = time.monotonic()
         while len(done) < world:
             try:
-                kind, rank, payload, ckpt_dir = report_q.get(timeout=600)
+                kind, rank, payload, ckpt_dir = report_q.get(timeout=1.0)
+                last_activity = time.monotonic()
+            except _q.Empty:
+                # liveness check: a worker that died without reporting (GPU
+                # fault, OOM-kill, os._exit) is surfaced as an error
+                for r, p in enumerate(procs):
+                    if r not in done and not p.is_alive():
+                        error = error or RuntimeError(
+                            f"worker {r} died unexpectedly (exitcode {p.exitcode})")
+                        done.add(r)
+                        for q in control_qs:
+                            q.put("stop")
+                if time.monotonic() - last_activity > 600:
+                    error = error or RuntimeError("training workers timed out")
+                    break
+                continue
             except Exception:
                 error = RuntimeError("training workers timed out")
                 break
