@@ -90,8 +90,11 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   T* qT_img = q_img + BQT * LD;                 // [D][LDT]
   T* do_img = qT_img + D * LDT;                 // [BQT][LD]
   T* doT_img = do_img + BQT * LD;               // [D][LDT]
-  T* ds_img = doT_img + D * LDT;                // [BQT][LDT] (q-major dS)
-  T* slab = ds_img + BQT * LDT;                 // [4][16][LDT] per-wave P^T / dS^T
+  T* slab = doT_img + D * LDT;                  // [4][16][LDT] per-wave P^T / dS^T
+  // ds_img ALIASES q_img when the strides match (D==64): every q_img read
+  // (the S^T B-fragments) completes before the P-slab barrier, and ds_img is
+  // only written after it — saves 9 KB LDS -> 2 blocks/CU instead of 1.
+  T* ds_img = (LD == LDT) ? q_img : slab + 4 * 16 * LDT;  // [BQT][LDT]
 
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
@@ -351,8 +354,9 @@ void launch_bwd(const torch::Tensor& dout, const torch::Tensor& q, const torch::
     bias_sh = bias->size(1) > 1 ? bias->stride(1) : 0;
   }
   constexpr int LD = D + 8, LDT = BKT + 8;
+  const int ds_extra = (LD == LDT) ? 0 : BQT * LDT;  // ds_img aliases q_img at D==64
   const int smem = sizeof(T) * (BKT * LD * 2 + BQT * LD * 2 + D * LDT * 3 +
-                                BQT * LDT + 4 * 16 * LDT);
+                                ds_extra + 4 * 16 * LDT);
   dim3 grid((Sk + BKT - 1) / BKT, B * H);
 #define LB(C, HB, HK, DB)                                                             \
   hipLaunchKernelGGL((attn_bwd_kernel<T, D, C, HB, HK, DB>), grid, dim3(256), smem,   \

@@ -93,20 +93,42 @@ def _deserialize(buf: memoryview):
     return pickle.loads(payload, buffers=buffers)
 
 
+try:  # native C++ store core (csrc/store/shmstore.cpp) — release/acquire
+    from mi355 import _store as _native_store  # type: ignore
+except Exception:  # pragma: no cover - fallback when not built
+    _native_store = None
+
+
 class ObjectStore:
     """Sealed shared-memory objects addressable by ref id from any process.
 
-    Attached segments are cached for the store's lifetime: deserialized
-    numpy/arrow buffers alias the mapped memory (zero-copy reads)."""
+    Backed by the native C++ core (mi355/_store.so: shm_open + mmap +
+    release/acquire seal word) when built; multiprocessing.shared_memory
+    otherwise. Attached segments are cached for the store's lifetime:
+    deserialized numpy/arrow buffers alias the mapped memory (zero-copy)."""
 
     def __init__(self):
         self._owned: list[str] = []
         self._attached: dict[str, shared_memory.SharedMemory] = {}
+        self._native = _native_store
+
+    # -- native-core fast paths ----------------------------------------- #
+    def _nput(self, data: bytes, name: str):
+        mv = self._native.create(name, len(data))
+        mv[:len(data)] = data
+        self._native.seal(name, len(data))
+        self._owned.append(name)
+
+    def _nget(self, name):
+        return self._native.attach(name)  # None until sealed
 
     def put(self, obj, ref: ObjectRef | None = None) -> ObjectRef:
         ref = ref or ObjectRef(uuid.uuid4().hex)
         data = _serialize(obj)
         name = _shm_name(ref)
+        if self._native is not None:
+            self._nput(data, name)
+            return ref
         seg = shared_memory.SharedMemory(name=name, create=True, size=len(data) + 8)
         _untrack(name)
         # header: 8-byte payload length; the final length write "seals" it
@@ -129,13 +151,18 @@ class ObjectStore:
         deadline = None if timeout is None else time.monotonic() + timeout
         delay = 0.0005
         while True:
-            try:
-                seg = self._attach(name)
-                (n,) = struct.unpack_from("<Q", seg.buf, 0)
-                if n > 0:
-                    return _deserialize(memoryview(seg.buf)[8:8 + n])
-            except FileNotFoundError:
-                pass
+            if self._native is not None:
+                buf = self._nget(name)
+                if buf is not None:
+                    return _deserialize(memoryview(buf))
+            else:
+                try:
+                    seg = self._attach(name)
+                    (n,) = struct.unpack_from("<Q", seg.buf, 0)
+                    if n > 0:
+                        return _deserialize(memoryview(seg.buf)[8:8 + n])
+                except FileNotFoundError:
+                    pass
             if deadline is not None and time.monotonic() > deadline:
                 raise TimeoutError(f"object {ref} not available after {timeout}s")
             time.sleep(delay)
@@ -143,6 +170,8 @@ class ObjectStore:
 
     def ready(self, ref: ObjectRef) -> bool:
         name = _shm_name(ref)
+        if self._native is not None:
+            return self._native.ready(name)
         try:
             seg = self._attach(name)
             (n,) = struct.unpack_from("<Q", seg.buf, 0)
@@ -153,6 +182,14 @@ class ObjectStore:
     def cleanup(self):
         import glob
 
+        if self._native is not None:
+            for name in self._owned:
+                try:
+                    self._native.unlink(name)
+                    self._native.close(name)
+                except Exception:
+                    pass
+            self._owned.clear()
         for name, seg in list(self._attached.items()):
             try:
                 seg.close()

@@ -81,6 +81,28 @@ def build(force: bool = False) -> Path:
     return OUT
 
 
+def build_store(force: bool = False) -> Path:
+    """Torch-free native shm store core (g++ + pybind11) -> mi355/_store.so."""
+    import sysconfig
+
+    import pybind11
+
+    src = CSRC / "store" / "shmstore.cpp"
+    out = REPO / "mi355" / "_store.so"
+    if not force and out.exists() and out.stat().st_mtime > src.stat().st_mtime:
+        return out
+    cmd = ["g++", "-O2", "-std=c++17", "-shared", "-fPIC",
+           f"-I{pybind11.get_include()}", f"-I{sysconfig.get_paths()['include']}",
+           str(src), "-o", str(out), "-lrt"]
+    r = subprocess.run(cmd, capture_output=True, text=True)
+    if r.returncode != 0:
+        sys.stderr.write(r.stdout[-2000:] + r.stderr[-6000:])
+        raise SystemExit("store build failed")
+    print(f"  built {out.relative_to(REPO)}")
+    return out
+
+
 if __name__ == "__main__":
     build(force="--force" in sys.argv)
+    build_store(force="--force" in sys.argv)
     print("OK")
