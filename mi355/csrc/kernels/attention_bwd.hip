@@ -114,8 +114,12 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   const float* delta_bh = delta + ((long)b * H + h) * Sq;
   const T* bias_bh = HAS_BIAS ? bias + (long)b * bias_sb + (long)h * bias_sh : nullptr;
   const float* kpad_b = HAS_KPAD ? kpad + (long)b * Sk : nullptr;
-  float* dq_bh = dq32 + ((long)b * H + h) * (long)Sq * D;
-  float* dbias_h = NEED_DBIAS ? dbias32 + (long)h * Sq * Sk : nullptr;
+  // dq32: (NKTILES, B, H, Sq, D) per-k-tile slabs; dbias32: (B, H, Sq, Sk)
+  // per-batch planes — both written with PLAIN stores (atomics made the
+  // dbias variants 2-3x slower than the no-bias one) and summed afterwards.
+  float* dq_bh = dq32 + (((long)blockIdx.x * B * H + (long)b * H + h) * Sq) * D;
+  float* dbias_h = NEED_DBIAS
+      ? dbias32 + ((long)b * H + h) * (long)Sq * Sk : nullptr;
 
   // ---- stage this block's K and V (+ K^T image), zero-padded OOB ----
   {
@@ -265,7 +269,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
       }
     }
 
-    // ---- dBias scatter (bias broadcast over batch -> atomic sum) ----
+    // ---- dBias store (per-batch plane, summed over batch afterwards) ----
     if (NEED_DBIAS) {
 #pragma unroll
       for (int t = 0; t < 4; ++t) {
@@ -274,8 +278,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int kr = krow0 + r;
-          if (kr < Sk && dst[t][r] != 0.f)
-            atomicAdd(&dbias_h[(long)qc * Sk + kr], dst[t][r]);
+          if (kr < Sk) dbias_h[(long)qc * Sk + kr] = dst[t][r];
         }
       }
     }
@@ -301,7 +304,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
       if (qr >= Sq) continue;
 #pragma unroll
       for (int t = 0; t < D / 16; ++t)
-        atomicAdd(&dq_bh[(long)qr * D + t * 16 + lr], dq_acc[t][r]);
+        dq_bh[(long)qr * D + t * 16 + lr] = dq_acc[t][r];
     }
   }
 
@@ -392,7 +395,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
   const int B = q.size(0), H = q.size(1), Sq = q.size(2), Sk = k.size(2), D = q.size(3);
   auto stream = at::hip::getCurrentHIPStream().stream();
   auto opts = q.options();
-  auto dq32 = torch::zeros({B, H, Sq, D}, opts.dtype(torch::kFloat));
+  const long nkt = (Sk + 63) / 64;
+  auto dq32 = torch::zeros({nkt, B, H, Sq, D}, opts.dtype(torch::kFloat));
   auto dk = torch::empty({B, Sk, H, D}, opts).permute({0, 2, 1, 3});
   auto dv = torch::empty({B, Sk, H, D}, opts).permute({0, 2, 1, 3});
   auto delta = torch::empty({B, H, Sq}, opts.dtype(torch::kFloat));
@@ -400,7 +404,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
   if (need_dbias) {
     TORCH_CHECK(bias.has_value() && bias->size(0) == 1,
                 "dbias path expects batch-broadcast bias (1,H,Sq,Sk)");
-    dbias32 = torch::zeros({1, H, Sq, Sk}, opts.dtype(torch::kFloat));
+    dbias32 = torch::zeros({B, H, Sq, Sk}, opts.dtype(torch::kFloat));
   }
   c10::optional<torch::Tensor> bias_c;
   if (bias.has_value() && bias->defined()) bias_c = bias->to(q.scalar_type()).contiguous();
@@ -420,10 +424,11 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
   else
     TORCH_CHECK(false, "attn_bwd: unsupported dtype/D=", D);
 
-  // dq: fp32 BHSD accumulator -> low-precision BSHD-physical view
+  // dq: sum the per-k-tile fp32 slabs -> low-precision BSHD-physical view
   auto dq = torch::empty({B, Sq, H, D}, opts).permute({0, 2, 1, 3});
-  dq.copy_(dq32);
+  dq.copy_(dq32.sum(0));
   std::vector<torch::Tensor> outs = {dq, dk, dv};
-  if (need_dbias) outs.push_back(dbias32.to(q.scalar_type()));
+  if (need_dbias)
+    outs.push_back(dbias32.sum(0, /*keepdim=*/true).to(q.scalar_type()));
   return outs;
 }
