@@ -158,6 +158,27 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
 
   T* my_slab = slab + wid * 16 * LDT;
 
+  // causal-skipped q-tiles: write the zeros the reductions expect (cheaper
+  // than pre-zeroing the whole 25 MB slab on the host)
+  for (int qt = 0; qt < qt_start; ++qt) {
+    const int q0z = qt * BQT;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qr = q0z + wid * 16 + lg * 4 + r;
+      if (qr >= Sq) continue;
+#pragma unroll
+      for (int t = 0; t < D / 16; ++t) dq_bh[(long)qr * D + t * 16 + lr] = 0.f;
+      if (NEED_DBIAS) {
+        const int qc = q0z + wid * 16 + lg * 4 + r;  // zero this block's kr cols
+#pragma unroll
+        for (int t2 = 0; t2 < 4; ++t2) {
+          const int kr = k0 + t2 * 16 + lr;
+          if (kr < Sk) dbias_h[(long)qc * Sk + kr] = 0.f;
+        }
+      }
+    }
+  }
+
   for (int qt = qt_start; qt * BQT < Sq; ++qt) {
     const int q0 = qt * BQT;
     __syncthreads();  // previous iteration's reads done before restage
@@ -396,7 +417,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
   auto stream = at::hip::getCurrentHIPStream().stream();
   auto opts = q.options();
   const long nkt = (Sk + 63) / 64;
-  auto dq32 = torch::zeros({nkt, B, H, Sq, D}, opts.dtype(torch::kFloat));
+  auto dq32 = torch::empty({nkt, B, H, Sq, D}, opts.dtype(torch::kFloat));
   auto dk = torch::empty({B, Sk, H, D}, opts).permute({0, 2, 1, 3});
   auto dv = torch::empty({B, Sk, H, D}, opts).permute({0, 2, 1, 3});
   auto delta = torch::empty({B, H, Sq}, opts.dtype(torch::kFloat));
@@ -404,7 +425,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
   if (need_dbias) {
     TORCH_CHECK(bias.has_value() && bias->size(0) == 1,
                 "dbias path expects batch-broadcast bias (1,H,Sq,Sk)");
-    dbias32 = torch::zeros({B, H, Sq, Sk}, opts.dtype(torch::kFloat));
+    dbias32 = torch::empty({B, H, Sq, Sk}, opts.dtype(torch::kFloat));
   }
   c10::optional<torch::Tensor> bias_c;
   if (bias.has_value() && bias->defined()) bias_c = bias->to(q.scalar_type()).contiguous();

@@ -438,7 +438,9 @@ class T5ForConditionalGeneration(nn.Module):
         attn0 = self.decoder.block[0].layer[0].SelfAttention
         full_bias = attn0.compute_bias(max_new_tokens, max_new_tokens, dev).to(dt)
         for t in range(max_new_tokens):
-            bias_t = full_bias[:, :, t : t + 1, : t + 1]
+            # one fp32-contiguous slice per step, shared by all layers (the
+            # decode kernel wants f32 bias; avoids 12 per-layer re-copies)
+            bias_t = full_bias[:, :, t : t + 1, : t + 1].float().contiguous()
             dec = self._decode_step(tokens, enc, kpad, caches, t, bias_t)
             if cfg.tie_word_embeddings:
                 dec = dec * (self.model_dim**-0.5)
