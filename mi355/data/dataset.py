@@ -37,6 +37,16 @@ def _to_table(batch) -> pa.Table:
     if isinstance(batch, pa.Table):
         return batch
     if isinstance(batch, pd.DataFrame):
+        # multi-dimensional ndarray cells (images, pixel_values, …) become
+        # nested-list Arrow columns
+        fix = {c: [np.asarray(v).tolist() for v in batch[c]]
+               for c in batch.columns
+               if len(batch) and isinstance(batch[c].iloc[0], np.ndarray)
+               and batch[c].iloc[0].ndim >= 2}
+        if fix:
+            batch = batch.copy()
+            for c, v in fix.items():
+                batch[c] = v
         return pa.Table.from_pandas(batch, preserve_index=False)
     if isinstance(batch, dict):
         cols = {}
@@ -358,6 +368,8 @@ class Dataset:
             out = {}
             for k, v in batch.items():
                 t = torch.as_tensor(np.asarray(v))
+                if t.dtype == torch.float64:
+                    t = t.float()  # arrow nested-list columns decode as f64
                 if dtypes and k in dtypes:
                     t = t.to(dtypes[k])
                 if device is not None:

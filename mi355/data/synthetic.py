@@ -46,7 +46,8 @@ def synthetic_ade20k(n_images: int = 16, size: int = 64, num_labels: int = 150,
     for _ in range(n_images):
         img = (rng.random((size, size, 3)) * 255).astype(np.uint8)
         ann = rng.integers(0, num_labels + 1, (size, size)).astype(np.uint8)
-        rows.append({"image": img, "annotation": ann,
+        # nested lists: Arrow stores ndim>1 image data as list-of-list columns
+        rows.append({"image": img.tolist(), "annotation": ann.tolist(),
                      "scene_category": int(rng.integers(0, 16))})
     return Dataset.from_pandas(pd.DataFrame(rows))
 

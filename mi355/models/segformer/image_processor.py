@@ -43,8 +43,17 @@ class SegformerImageProcessor:
                                     else IMAGENET_STD, dtype=np.float32)
         self.do_reduce_labels = do_reduce_labels
 
+    @staticmethod
+    def _to_ndarray(x) -> np.ndarray:
+        # arrow list<list<...>> columns round-trip as nested OBJECT arrays;
+        # stack them level by level into a dense ndarray
+        a = np.asarray(x)
+        if a.dtype != object:
+            return a
+        return np.stack([SegformerImageProcessor._to_ndarray(e) for e in a])
+
     def _one(self, image) -> np.ndarray:
-        img = np.asarray(image)
+        img = self._to_ndarray(image)
         if img.ndim == 2:
             img = np.stack([img] * 3, axis=-1)
         if img.shape[-1] == 4:
@@ -60,7 +69,7 @@ class SegformerImageProcessor:
         return img.transpose(2, 0, 1)  # CHW
 
     def _one_label(self, seg) -> np.ndarray:
-        lab = np.asarray(seg)
+        lab = self._to_ndarray(seg)
         if self.do_resize:
             lab = _resize_image(lab, (self.size["height"], self.size["width"]),
                                 nearest=True)
