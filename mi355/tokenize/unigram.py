@@ -104,6 +104,14 @@ class UnigramModel:
     pieces: list  # [(piece, score, type)]
     unk_id: int = 2
 
+    def __getstate__(self):
+        return {"pieces": self.pieces, "unk_id": self.unk_id}
+
+    def __setstate__(self, state):
+        self.pieces = state["pieces"]
+        self.unk_id = state["unk_id"]
+        self.__post_init__()
+
     def __post_init__(self):
         self.piece_to_id = {p: i for i, (p, _, _) in enumerate(self.pieces)}
         self.scores = [s for _, s, _ in self.pieces]
@@ -111,6 +119,20 @@ class UnigramModel:
         self._max_len = max((len(p) for p, _, t in self.pieces if t == 1), default=1)
         self._vocab = {p: (i, s) for i, (p, s, t) in enumerate(self.pieces) if t != 3}
         self._unk_score = min(self.scores, default=-10.0) - 10.0
+        # native C++ trie+Viterbi core (csrc/tokenize/unigram_core.cpp)
+        self._core = None
+        try:
+            from mi355 import _tokenize
+
+            core = _tokenize.UnigramCore()
+            core.unk_id = self.unk_id
+            core.unk_score = self._unk_score
+            for i, (piece, score, t) in enumerate(self.pieces):
+                if t != 3:  # control pieces never match text
+                    core.add_piece(piece.encode("utf-8"), i, score)
+            self._core = core
+        except ImportError:
+            pass
 
     @classmethod
     def load(cls, path: str) -> "UnigramModel":
@@ -130,6 +152,8 @@ class UnigramModel:
     def encode_text(self, text: str) -> list[int]:
         """Normalize (spaces -> meta-symbol, leading space) then Viterbi."""
         text = SPACE + text.strip().replace(" ", SPACE)
+        if self._core is not None:
+            return list(self._core.encode(text.encode("utf-8")))
         n = len(text)
         best = [(-math.inf, -1, -1)] * (n + 1)  # (score, prev, piece_id)
         best[0] = (0.0, -1, -1)

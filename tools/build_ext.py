@@ -102,7 +102,29 @@ def build_store(force: bool = False) -> Path:
     return out
 
 
+def build_cpp_module(src_rel: str, name: str, force: bool = False) -> Path:
+    import sysconfig
+
+    import pybind11
+
+    src = CSRC / src_rel
+    out = REPO / "mi355" / f"{name}.so"
+    if not force and out.exists() and out.stat().st_mtime > src.stat().st_mtime:
+        return out
+    cmd = ["g++", "-O2", "-std=c++17", "-shared", "-fPIC",
+           f"-I{pybind11.get_include()}", f"-I{sysconfig.get_paths()['include']}",
+           str(src), "-o", str(out)]
+    r = subprocess.run(cmd, capture_output=True, text=True)
+    if r.returncode != 0:
+        sys.stderr.write(r.stdout[-2000:] + r.stderr[-6000:])
+        raise SystemExit(f"{name} build failed")
+    print(f"  built {out.relative_to(REPO)}")
+    return out
+
+
 if __name__ == "__main__":
     build(force="--force" in sys.argv)
     build_store(force="--force" in sys.argv)
+    build_cpp_module("tokenize/unigram_core.cpp", "_tokenize",
+                     force="--force" in sys.argv)
     print("OK")
