@@ -109,30 +109,43 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int diag_off = Sk - Sq;  // causal: allow kc <= qr + diag_off
   const int k_end = CAUSAL ? min(Sk, q0 + BQ + diag_off) : Sk;
 
-  for (int k0 = 0; k0 < k_end; k0 += BK) {
-    // ---- stage K [BK][D] and Vt [D][BK] (transposed) ----
-    {
-      // 256 threads; each stages 16 elems of K and of V (BK*D/256 = 16)
-      const int tid = threadIdx.x;
-      const int j = tid / (D / 8);            // K row within tile (8-elem chunks)
-      const int d0 = (tid % (D / 8)) * 8;
-      constexpr int ROWS_PER_PASS = 256 / (D / 8);
-      for (int jj = j; jj < BK; jj += ROWS_PER_PASS) {
-        const int kr = k0 + jj;
-        VecT<T, 8> kv, vv;
-        if (kr < Sk) {
-          kv = vload<T, 8>(kbase + (long)kr * ks.ss + d0);
-          vv = vload<T, 8>(vbase + (long)kr * vs.ss + d0);
-        } else {
+  // ---- async-STAGE split (guide T14): global loads for tile t+1 issue
+  // BEFORE tile t's compute; the LDS write lands after the barrier, so HBM
+  // latency hides under the MFMA phase. Per thread: 2 K rows + 2 V rows.
+  const int stage_j = threadIdx.x / (D / 8);
+  const int stage_d0 = (threadIdx.x % (D / 8)) * 8;
+  constexpr int RPP = 256 / (D / 8);
+  constexpr int NPASS = BK / RPP;  // row passes per tile (<= 2 for D<=64)
+  VecT<T, 8> kst[NPASS], vst_r[NPASS];
+
+  auto stage_load = [&](int k0) {
 #pragma unroll
-          for (int u = 0; u < 8; ++u) { kv.data[u] = T(0.f); vv.data[u] = T(0.f); }
-        }
-        vstore<T, 8>(&k_lds[jj * KLD + d0], kv);
+    for (int p = 0; p < NPASS; ++p) {
+      const int kr = k0 + stage_j + p * RPP;
+      if (kr < Sk) {
+        kst[p] = vload<T, 8>(kbase + (long)kr * ks.ss + stage_d0);
+        vst_r[p] = vload<T, 8>(vbase + (long)kr * vs.ss + stage_d0);
+      } else {
 #pragma unroll
-        for (int u = 0; u < 8; ++u) vt_lds[(d0 + u) * VLD + jj] = vv.data[u];
+        for (int u = 0; u < 8; ++u) { kst[p].data[u] = T(0.f); vst_r[p].data[u] = T(0.f); }
       }
     }
+  };
+  auto stage_write = [&]() {
+#pragma unroll
+    for (int p = 0; p < NPASS; ++p) {
+      const int jj = stage_j + p * RPP;
+      vstore<T, 8>(&k_lds[jj * KLD + stage_d0], kst[p]);
+#pragma unroll
+      for (int u = 0; u < 8; ++u) vt_lds[(stage_d0 + u) * VLD + jj] = vst_r[p].data[u];
+    }
+  };
+
+  if (k_end > 0) stage_load(0);
+  for (int k0 = 0; k0 < k_end; k0 += BK) {
+    stage_write();
     __syncthreads();
+    if (k0 + BK < k_end) stage_load(k0 + BK);  // in flight under the MFMAs
 
     // ---- S = Q K^T for this wave's 16 rows x BK cols ----
     f32x4 s_acc[NKT];
