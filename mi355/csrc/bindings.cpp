@@ -21,7 +21,8 @@ torch::Tensor decode_attn(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
                           c10::optional<torch::Tensor> kpad);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                     c10::optional<torch::Tensor> bias,
-                                    c10::optional<torch::Tensor> kpad, bool causal);
+                                    c10::optional<torch::Tensor> kpad, bool causal,
+                                    double p_drop, c10::optional<torch::Tensor> seed);
 torch::Tensor token_embed_fwd(torch::Tensor ids, torch::Tensor weight);
 torch::Tensor token_embed_bwd(torch::Tensor ids, torch::Tensor dy, long vocab,
                               torch::ScalarType out_dtype);
@@ -31,7 +32,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
                                     torch::Tensor v, torch::Tensor o, torch::Tensor lse,
                                     c10::optional<torch::Tensor> bias,
                                     c10::optional<torch::Tensor> kpad,
-                                    bool causal, bool need_dbias);
+                                    bool causal, bool need_dbias,
+                                    double p_drop, c10::optional<torch::Tensor> seed);
 torch::Tensor seg_postprocess(torch::Tensor logits, long Ho, long Wo);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

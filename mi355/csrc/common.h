@@ -89,6 +89,18 @@ DEVINL float block_reduce_max(float x, float* smem) {
   return r;
 }
 
+// deterministic per-element uniform in [0,1) for dropout masks: same
+// (seed, idx) in forward and backward reproduces the same mask without
+// storing it (splitmix64 finalizer)
+DEVINL float hash_u01(long long seed, long long idx) {
+  unsigned long long x = (unsigned long long)seed * 0x9E3779B97F4A7C15ull ^
+                         (unsigned long long)idx * 0xBF58476D1CE4E5B9ull;
+  x ^= x >> 30; x *= 0xBF58476D1CE4E5B9ull;
+  x ^= x >> 27; x *= 0x94D049BB133111EBull;
+  x ^= x >> 31;
+  return (float)(x >> 40) * (1.0f / 16777216.0f);
+}
+
 #define HIP_CHECK_KERNEL()                                           \
   do {                                                               \
     hipError_t e_ = hipGetLastError();                               \

@@ -77,7 +77,8 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
     float* __restrict__ dbias32,
     int B, int H, int Sq, int Sk,
     BStr qs, BStr ks, BStr vs, BStr dos, BStr dks, BStr dvs,
-    long bias_sb, long bias_sh) {
+    long bias_sb, long bias_sh,
+    float p_drop, const long* __restrict__ seed_ptr) {
   using FR = typename MFMA16B<T>::frag;
   constexpr int LD = D + 8;        // [64][D] images
   constexpr int LDT = BKT + 8;     // [D][64] transposed images + slabs
@@ -229,8 +230,13 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
       }
     }
 
-    // ---- recompute P^T, compute dS^T ----
+    // ---- recompute P^T (and the dropout mask), compute dS^T ----
     // C layout: row (k-local) = lg*4 + r, col (q-local) = t*16 + lr
+    // with attention-prob dropout: O = sum_k (m_k/keep) Pn_k V_k, so
+    //   dV = (m/keep * Pn)^T dO   (mask applied to the slab P)
+    //   dS = Pn * ((m/keep) * (dO V^T) - delta),  delta = rowsum(dO*O)
+    const long long seed = seed_ptr ? *seed_ptr : 0;
+    const float keep_inv = p_drop > 0.f ? 1.f / (1.f - p_drop) : 1.f;
     float pt[4][4], dst[4][4];
     const int krow0 = k0 + wid * 16 + lg * 4;
 #pragma unroll
@@ -248,8 +254,13 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
         if (HAS_KPAD && k_ok) sv += kpad_b[kr];
         bool masked = !k_ok || !q_ok || (CAUSAL && kr > qc + diag_off);
         float p = (masked || l == -INFINITY) ? 0.f : __expf(sv - l);
-        pt[t][r] = p;
-        dst[t][r] = p * (dpt_acc[t][r] - dl);
+        float mfac = 1.f;
+        if (p_drop > 0.f) {
+          const long long idx = ((long long)bh * Sq + qc) * Sk + kr;
+          mfac = (hash_u01(seed, idx) >= p_drop) ? keep_inv : 0.f;
+        }
+        pt[t][r] = p * mfac;  // masked P -> slab (dV path)
+        dst[t][r] = p * (dpt_acc[t][r] * mfac - dl);
       }
     }
 
@@ -347,6 +358,7 @@ void launch_bwd(const torch::Tensor& dout, const torch::Tensor& q, const torch::
                 const torch::Tensor& v, const torch::Tensor& o, const torch::Tensor& lse,
                 const c10::optional<torch::Tensor>& bias,
                 const c10::optional<torch::Tensor>& kpad, bool causal, bool need_dbias,
+                double p_drop, const c10::optional<torch::Tensor>& seed,
                 torch::Tensor& dq32, torch::Tensor& dk, torch::Tensor& dv,
                 torch::Tensor& dbias32, torch::Tensor& delta, hipStream_t stream) {
   const int B = q.size(0), H = q.size(1), Sq = q.size(2), Sk = k.size(2);
@@ -389,7 +401,9 @@ void launch_bwd(const torch::Tensor& dout, const torch::Tensor& q, const torch::
                      lse.data_ptr<float>(), delta.data_ptr<float>(), bias_p, kpad_p,  \
                      dq32.data_ptr<float>(), (T*)dk.data_ptr(), (T*)dv.data_ptr(),    \
                      need_dbias ? dbias32.data_ptr<float>() : nullptr, B, H, Sq, Sk,  \
-                     qs, ks, vs, dos, dks, dvs, bias_sb, bias_sh)
+                     qs, ks, vs, dos, dks, dvs, bias_sb, bias_sh, (float)p_drop,       \
+                     (seed.has_value() && seed->defined())                             \
+                         ? seed->data_ptr<long>() : nullptr)
   // 16 template combos
   if (causal) {
     if (has_bias && need_dbias) { if (has_kpad) LB(true, true, true, true); else LB(true, true, false, true); }
@@ -410,7 +424,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
                                     torch::Tensor v, torch::Tensor o, torch::Tensor lse,
                                     c10::optional<torch::Tensor> bias,
                                     c10::optional<torch::Tensor> kpad,
-                                    bool causal, bool need_dbias) {
+                                    bool causal, bool need_dbias,
+                                    double p_drop, c10::optional<torch::Tensor> seed) {
   TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1 &&
               dout.stride(3) == 1 && o.stride(3) == 1, "head dim must be contiguous");
   const int B = q.size(0), H = q.size(1), Sq = q.size(2), Sk = k.size(2), D = q.size(3);
@@ -432,16 +447,16 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
 
   if (q.scalar_type() == torch::kBFloat16 && D == 64)
     launch_bwd<__hip_bfloat16, 64>(dout, q, k, v, o, lse, bias_c, kpad, causal,
-                                   need_dbias, dq32, dk, dv, dbias32, delta, stream);
+                                   need_dbias, p_drop, seed, dq32, dk, dv, dbias32, delta, stream);
   else if (q.scalar_type() == torch::kHalf && D == 64)
     launch_bwd<__half, 64>(dout, q, k, v, o, lse, bias_c, kpad, causal, need_dbias,
-                           dq32, dk, dv, dbias32, delta, stream);
+                           p_drop, seed, dq32, dk, dv, dbias32, delta, stream);
   else if (q.scalar_type() == torch::kBFloat16 && D == 32)
     launch_bwd<__hip_bfloat16, 32>(dout, q, k, v, o, lse, bias_c, kpad, causal,
-                                   need_dbias, dq32, dk, dv, dbias32, delta, stream);
+                                   need_dbias, p_drop, seed, dq32, dk, dv, dbias32, delta, stream);
   else if (q.scalar_type() == torch::kHalf && D == 32)
     launch_bwd<__half, 32>(dout, q, k, v, o, lse, bias_c, kpad, causal, need_dbias,
-                           dq32, dk, dv, dbias32, delta, stream);
+                           p_drop, seed, dq32, dk, dv, dbias32, delta, stream);
   else
     TORCH_CHECK(false, "attn_bwd: unsupported dtype/D=", D);
 

@@ -60,7 +60,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     T* __restrict__ out, float* __restrict__ lse,
     int B, int H, int Sq, int Sk,
     Strides3 qs, Strides3 ks, Strides3 vs, Strides3 os,
-    long bias_sb, long bias_sh /* element strides; 0 = broadcast */) {
+    long bias_sb, long bias_sh /* element strides; 0 = broadcast */,
+    float p_drop, const long* __restrict__ seed_ptr) {
   using FR = typename MFMA16<T>::frag;
   constexpr int KLD = D + 8;       // K tile leading dim
   constexpr int VLD = BK + 8;      // Vt tile leading dim
@@ -218,11 +219,23 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       for (int r = 0; r < 4; ++r) o_acc[t][r] *= alpha[r];
 
     // ---- stage P (C layout -> wave-local LDS slab, row-major [16][BK]) ----
+    // T5 attention-prob dropout (training): mask applied to the P used for
+    // the PV accumulation; the softmax normalizer l keeps the full sum.
+    const long long seed = seed_ptr ? *seed_ptr : 0;
+    const float keep_inv = p_drop > 0.f ? 1.f / (1.f - p_drop) : 1.f;
 #pragma unroll
     for (int t = 0; t < NKT; ++t)
 #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        p_lds[wid][(lg * 4 + r) * PLD + t * 16 + lr] = DT<T>::from_f32(p_val[t][r]);
+      for (int r = 0; r < 4; ++r) {
+        float pv = p_val[t][r];
+        if (p_drop > 0.f) {
+          const long long qr = q0 + wid * 16 + lg * 4 + r;
+          const long long kc = k0 + t * 16 + lr;
+          const long long idx = ((long long)bh * Sq + qr) * Sk + kc;
+          pv = (hash_u01(seed, idx) >= p_drop) ? pv * keep_inv : 0.f;
+        }
+        p_lds[wid][(lg * 4 + r) * PLD + t * 16 + lr] = DT<T>::from_f32(pv);
+      }
     // the slab is wave-local, but make the cross-LANE write->read ordering
     // explicit (every wave reaches this point each iteration)
     __syncthreads();
@@ -266,6 +279,7 @@ template <typename T, int D>
 void launch_attn(const torch::Tensor& q, const torch::Tensor& k, const torch::Tensor& v,
                  const c10::optional<torch::Tensor>& bias,
                  const c10::optional<torch::Tensor>& kpad, bool causal,
+                 double p_drop, const c10::optional<torch::Tensor>& seed,
                  torch::Tensor& out, torch::Tensor& lse, hipStream_t stream) {
   const int B = q.size(0), H = q.size(1), Sq = q.size(2), Sk = k.size(2);
   dim3 grid((Sq + BQ - 1) / BQ, B * H);
@@ -288,7 +302,9 @@ void launch_attn(const torch::Tensor& q, const torch::Tensor& k, const torch::Te
                      stream, (const T*)q.data_ptr(), (const T*)k.data_ptr(),        \
                      (const T*)v.data_ptr(), bias_p, kpad_p, (T*)out.data_ptr(),    \
                      lse.data_ptr<float>(), B, H, Sq, Sk, qst, kst, vst, ost,       \
-                     bias_sb, bias_sh)
+                     bias_sb, bias_sh, (float)p_drop,                                  \
+                     (seed.has_value() && seed->defined())                             \
+                         ? seed->data_ptr<long>() : nullptr)
   if (causal) {
     if (has_bias && has_kpad) LAUNCH(true, true, true);
     else if (has_bias) LAUNCH(true, true, false);
@@ -308,7 +324,8 @@ void launch_attn(const torch::Tensor& q, const torch::Tensor& k, const torch::Te
 
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                     c10::optional<torch::Tensor> bias,
-                                    c10::optional<torch::Tensor> kpad, bool causal) {
+                                    c10::optional<torch::Tensor> kpad, bool causal,
+                                    double p_drop, c10::optional<torch::Tensor> seed) {
   // (B,H,S,D) semantics with ARBITRARY (b,h,s) strides, d contiguous — so
   // (B,S,H,D)-physical projection views need no transpose copies.
   TORCH_CHECK(q.is_cuda() && q.dim() == 4 && q.stride(3) == 1, "q must be (B,H,Sq,D), d contiguous");
@@ -323,13 +340,13 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   c10::optional<torch::Tensor> bias_c;
   if (bias.has_value() && bias->defined()) bias_c = bias->to(q.scalar_type()).contiguous();
   if (q.scalar_type() == torch::kBFloat16 && D == 64)
-    launch_attn<__hip_bfloat16, 64>(q, k, v, bias_c, kpad, causal, out, lse, stream);
+    launch_attn<__hip_bfloat16, 64>(q, k, v, bias_c, kpad, causal, p_drop, seed, out, lse, stream);
   else if (q.scalar_type() == torch::kHalf && D == 64)
-    launch_attn<__half, 64>(q, k, v, bias_c, kpad, causal, out, lse, stream);
+    launch_attn<__half, 64>(q, k, v, bias_c, kpad, causal, p_drop, seed, out, lse, stream);
   else if (q.scalar_type() == torch::kBFloat16 && D == 32)
-    launch_attn<__hip_bfloat16, 32>(q, k, v, bias_c, kpad, causal, out, lse, stream);
+    launch_attn<__hip_bfloat16, 32>(q, k, v, bias_c, kpad, causal, p_drop, seed, out, lse, stream);
   else if (q.scalar_type() == torch::kHalf && D == 32)
-    launch_attn<__half, 32>(q, k, v, bias_c, kpad, causal, out, lse, stream);
+    launch_attn<__half, 32>(q, k, v, bias_c, kpad, causal, p_drop, seed, out, lse, stream);
   else
     TORCH_CHECK(false, "attn_fwd: unsupported dtype/head-dim (bf16/f16, D in {32,64}); got D=", D);
   return {out, lse};

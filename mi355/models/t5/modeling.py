@@ -182,6 +182,7 @@ class T5Attention(nn.Module):
             qs = self._shape(self.q(hidden), b)
             ks = self._shape(self.k(kv_hidden), b)
             vs = self._shape(self.v(kv_hidden), b)
+        p_drop = self.dropout if self.training else 0.0
         if kv_cache is not None and sq == 1 and not static:
             out = ops.decode_attention(qs, kv_cache["k"], kv_cache["v"],
                                        kv_cache["len"], bias=position_bias)
@@ -190,7 +191,8 @@ class T5Attention(nn.Module):
             out = ops.decode_attention(qs, ks, vs, ks.shape[2],
                                        bias=position_bias, kpad=kpad)
         else:
-            out = ops.attention(qs, ks, vs, bias=position_bias, causal=causal, kpad=kpad)
+            out = ops.attention(qs, ks, vs, bias=position_bias, causal=causal,
+                                kpad=kpad, p_drop=p_drop)
         out = out.transpose(1, 2).reshape(b, sq, self.inner_dim)
         return self.o(out)
 

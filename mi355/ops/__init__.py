@@ -140,9 +140,32 @@ def geglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
 # --------------------------------------------------------------------------- #
 # Fused attention with additive bias (T5 relative-position bias / masks)      #
 # --------------------------------------------------------------------------- #
+_dropout_seeds: dict = {}
+
+
+def _dropout_seed(device) -> torch.Tensor:
+    """Per-device int64 dropout-seed tensor read by the attention kernels.
+    Lives on device so hipGraph replays see the bumped value (the bump op is
+    captured in the graph)."""
+    key = str(device)
+    t = _dropout_seeds.get(key)
+    if t is None:
+        t = torch.randint(0, 2**62, (1,), dtype=torch.int64, device=device)
+        _dropout_seeds[key] = t
+    return t
+
+
+def bump_dropout_seed(device):
+    """Advance the attention-dropout stream; called once per optimizer step
+    (FlatReplica.zero_grads) BEFORE the forward pass so fwd+bwd of one step
+    see the same masks."""
+    if str(device) in _dropout_seeds or (hasattr(device, "type") and device.type == "cuda"):
+        _dropout_seed(device).add_(1)
+
+
 class _Attention(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, bias, causal, kpad):
+    def forward(ctx, q, k, v, bias, causal, kpad, p_drop):
         # arbitrary (b,h,s) strides supported by the kernel; only the head
         # dim must be contiguous — projection views need no transpose copy
         if q.stride(-1) != 1: q = q.contiguous()
@@ -150,9 +173,11 @@ class _Attention(torch.autograd.Function):
         if v.stride(-1) != 1: v = v.contiguous()
         bias_c = bias.contiguous() if bias is not None else None
         kpad_c = kpad.contiguous().float() if kpad is not None else None
-        o, lse = _load_ext().attn_fwd(q, k, v, bias_c, kpad_c, causal)
+        seed = _dropout_seed(q.device) if p_drop > 0 else None
+        o, lse = _load_ext().attn_fwd(q, k, v, bias_c, kpad_c, causal, p_drop, seed)
         ctx.save_for_backward(q, k, v, bias_c, kpad_c, o, lse)
         ctx.causal = causal
+        ctx.p_drop = p_drop
         ctx.bias_requires_grad = bias is not None and bias.requires_grad
         return o
 
@@ -164,11 +189,14 @@ class _Attention(torch.autograd.Function):
         use_hip_bwd = hasattr(ext, "attn_bwd") and q.shape[-1] in (32, 64) and \
             (not ctx.bias_requires_grad or (bias is not None and bias.shape[0] == 1)) \
             and os.environ.get("MI355_ATTN_BWD_REF") != "1"
+        if ctx.p_drop > 0 and not use_hip_bwd:
+            raise RuntimeError("attention-prob dropout bwd requires the HIP kernel")
         if use_hip_bwd:
             if do.stride(-1) != 1:
                 do = do.contiguous()
+            seed = _dropout_seed(q.device) if ctx.p_drop > 0 else None
             res = ext.attn_bwd(do, q, k, v, o, lse, bias, kpad, ctx.causal,
-                               ctx.bias_requires_grad)
+                               ctx.bias_requires_grad, ctx.p_drop, seed)
             dq, dk, dv = res[0], res[1], res[2]
             dbias = res[3] if len(res) > 3 else None
         else:  # composed bwd via recompute (hipBLASLt batched GEMMs)
@@ -180,7 +208,7 @@ class _Attention(torch.autograd.Function):
             for d in range(dbias.dim()):
                 if bias.shape[d] == 1 and dbias.shape[d] != 1:
                     dbias = dbias.sum(dim=d, keepdim=True)
-        return dq, dk, dv, dbias, None, None
+        return dq, dk, dv, dbias, None, None, None
 
 
 def _attn_bwd_composed(do, q, k, v, lse, bias, kpad, causal, need_dbias):
@@ -207,15 +235,18 @@ def _attn_bwd_composed(do, q, k, v, lse, bias, kpad, causal, need_dbias):
     return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), dbias
 
 
-def attention(q, k, v, bias=None, causal=False, scale: float = 1.0, kpad=None):
+def attention(q, k, v, bias=None, causal=False, scale: float = 1.0, kpad=None,
+              p_drop: float = 0.0):
     """kpad: optional (B, Sk) ADDITIVE fp32 key-padding mask (0 keep / -1e9
     drop), kept separate from ``bias`` so a (1,H,Sq,Sk) relative-position bias
-    never has to be materialized per batch element."""
+    never has to be materialized per batch element. p_drop: T5's
+    attention-probability dropout (training only), fused into the kernels."""
     hip_ok = (q.dtype in (torch.bfloat16, torch.float16)
               and q.shape[-1] in (32, 64) and scale == 1.0)
     if hip_ok and _use_hip(q):
-        return _Attention.apply(q, k, v, bias, causal, kpad)
-    return torch_ref.attention(q, k, v, bias=bias, causal=causal, scale=scale, kpad=kpad)
+        return _Attention.apply(q, k, v, bias, causal, kpad, float(p_drop))
+    return torch_ref.attention(q, k, v, bias=bias, causal=causal, scale=scale,
+                               kpad=kpad, p_drop=p_drop)
 
 
 # --------------------------------------------------------------------------- #

@@ -48,6 +48,7 @@ def attention(
     causal: bool = False,
     scale: float = 1.0,
     kpad: torch.Tensor | None = None,
+    p_drop: float = 0.0,
 ) -> torch.Tensor:
     """Attention core with additive bias (T5 relative-position bias), unscaled
     by default (T5 absorbs 1/sqrt(d) into init — SURVEY.md K4).
@@ -68,6 +69,8 @@ def attention(
         neg = torch.finfo(scores.dtype).min
         scores = scores.masked_fill(idx_k > idx_q, neg)
     probs = F.softmax(scores.float(), dim=-1).to(q.dtype)
+    if p_drop > 0.0:
+        probs = F.dropout(probs, p=p_drop, training=True)
     return torch.matmul(probs, v)
 
 

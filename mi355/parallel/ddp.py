@@ -154,6 +154,11 @@ class FlatReplica:
                 g.flat_grad.div_(world)
 
     def zero_grads(self):
+        dev = self.groups[0][1].flat_grad.device if self.groups else None
+        if dev is not None and dev.type == "cuda":
+            from .. import ops
+
+            ops.bump_dropout_seed(dev)  # new attention-dropout masks per step
         for _, g in self.groups:
             g.flat_grad.zero_()
             g.bucket_ready = [0] * len(g.buckets)
