@@ -4,9 +4,13 @@
 
 std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps);
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
-                                       torch::Tensor inv, c10::optional<torch::Tensor> dextra);
+                                       torch::Tensor inv, c10::optional<torch::Tensor> dextra,
+                                       double p_drop, c10::optional<torch::Tensor> seed,
+                                       long salt);
 std::vector<torch::Tensor> add_rmsnorm_fwd(torch::Tensor x, torch::Tensor r,
-                                           torch::Tensor w, double eps);
+                                           torch::Tensor w, double eps,
+                                           double p_drop, c10::optional<torch::Tensor> seed,
+                                           long salt);
 torch::Tensor geglu_fwd(torch::Tensor gate, torch::Tensor up);
 std::vector<torch::Tensor> geglu_bwd(torch::Tensor dout, torch::Tensor gate, torch::Tensor up);
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels, long ignore_index);

@@ -48,12 +48,18 @@ __global__ void rmsnorm_fwd_kernel(const T* __restrict__ x, const T* __restrict_
 
 // fused residual-add + RMSNorm: s = x + r; y = rmsnorm(s) (pre-norm T5
 // chains "resid += delta; norm(resid)" — one kernel instead of add + norm)
+// optional fused dropout on the delta branch: s = x + drop(r); mask derived
+// from (seed, salt, element index) so backward reproduces it (common.h)
 template <typename T, int VEC>
 __global__ void add_rmsnorm_fwd_kernel(const T* __restrict__ x, const T* __restrict__ r,
                                        T* __restrict__ s_out, T* __restrict__ y,
                                        const T* __restrict__ w, float* __restrict__ inv_rms,
-                                       int rows, int D, float eps) {
+                                       int rows, int D, float eps,
+                                       float p_drop, const long* __restrict__ seed_ptr,
+                                       long salt) {
   __shared__ float red[16];
+  const long long seed = (seed_ptr ? *seed_ptr : 0) ^ salt;
+  const float keep_inv = p_drop > 0.f ? 1.f / (1.f - p_drop) : 1.f;
   for (int row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* xr = x + (long)row * D;
     const T* rr = r + (long)row * D;
@@ -66,7 +72,12 @@ __global__ void add_rmsnorm_fwd_kernel(const T* __restrict__ x, const T* __restr
       VecT<T, VEC> sv;
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
-        float f = DT<T>::to_f32(xv.data[j]) + DT<T>::to_f32(rv.data[j]);
+        float rf = DT<T>::to_f32(rv.data[j]);
+        if (p_drop > 0.f) {
+          const long long idx = (long long)row * D + i + j;
+          rf = (hash_u01(seed, idx) >= p_drop) ? rf * keep_inv : 0.f;
+        }
+        float f = DT<T>::to_f32(xv.data[j]) + rf;
         sv.data[j] = DT<T>::from_f32(f);
         ss += f * f;
       }
@@ -95,8 +106,13 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy, const T* __restrict
                                    const T* __restrict__ w, const float* __restrict__ inv_rms,
                                    const T* __restrict__ dextra /*nullable*/,
                                    T* __restrict__ dx, float* __restrict__ dw_part,
-                                   int rows, int D) {
+                                   int rows, int D,
+                                   T* __restrict__ ddelta /*nullable: dropout branch*/,
+                                   float p_drop, const long* __restrict__ seed_ptr,
+                                   long salt) {
   __shared__ float red[16];
+  const long long seed = (seed_ptr ? *seed_ptr : 0) ^ salt;
+  const float keep_inv = p_drop > 0.f ? 1.f / (1.f - p_drop) : 1.f;
   float dwacc[MAXSLOT * VEC];
 #pragma unroll
   for (int j = 0; j < MAXSLOT * VEC; ++j) dwacc[j] = 0.f;
@@ -126,16 +142,27 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy, const T* __restrict
       VecT<T, VEC> o;
       VecT<T, VEC> ev;
       if (dextra) ev = vload<T, VEC>(dextra + (long)row * D + i);
+      VecT<T, VEC> od;
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
         float xf = DT<T>::to_f32(xv.data[j]);
         float dyf = DT<T>::to_f32(dyv.data[j]);
         float wf = DT<T>::to_f32(wv.data[j]);
         float extra = dextra ? DT<T>::to_f32(ev.data[j]) : 0.f;
-        o.data[j] = DT<T>::from_f32(inv * wf * dyf - xf * c + extra);
+        const float dxf = inv * wf * dyf - xf * c + extra;
+        o.data[j] = DT<T>::from_f32(dxf);
+        if (ddelta) {
+          float mf = 1.f;
+          if (p_drop > 0.f) {
+            const long long idx = (long long)row * D + i + j;
+            mf = (hash_u01(seed, idx) >= p_drop) ? keep_inv : 0.f;
+          }
+          od.data[j] = DT<T>::from_f32(dxf * mf);
+        }
         dwacc[slot * VEC + j] += dyf * xf * inv;
       }
       vstore<T, VEC>(dxr + i, o);
+      if (ddelta) vstore<T, VEC>(ddelta + (long)row * D + i, od);
     }
     __syncthreads();
   }
@@ -179,7 +206,9 @@ void launch_fwd(const torch::Tensor& x, const torch::Tensor& w, torch::Tensor& y
 template <typename T>
 void launch_bwd(const torch::Tensor& dy, const torch::Tensor& x, const torch::Tensor& w,
                 const torch::Tensor& inv, const torch::Tensor* dextra,
-                torch::Tensor& dx, torch::Tensor& dw, hipStream_t stream) {
+                torch::Tensor& dx, torch::Tensor& dw, torch::Tensor* ddelta,
+                float p_drop, const torch::Tensor* seed, long salt,
+                hipStream_t stream) {
   const int rows = x.size(0), D = x.size(1);
   constexpr int VEC = sizeof(T) == 4 ? 4 : 8;
   const int threads = (D / VEC >= 256) ? 256 : 128;
@@ -190,7 +219,9 @@ void launch_bwd(const torch::Tensor& dy, const torch::Tensor& x, const torch::Te
                      (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (const T*)w.data_ptr(),
                      inv.data_ptr<float>(),
                      dextra ? (const T*)dextra->data_ptr() : nullptr,
-                     (T*)dx.data_ptr(), part.data_ptr<float>(), rows, D);
+                     (T*)dx.data_ptr(), part.data_ptr<float>(), rows, D,
+                     ddelta ? (T*)ddelta->data_ptr() : nullptr, p_drop,
+                     seed ? seed->data_ptr<long>() : nullptr, salt);
   HIP_CHECK_KERNEL();
   hipLaunchKernelGGL(colsum_kernel, dim3((D + 255) / 256, 16), dim3(256), 0, stream,
                      part.data_ptr<float>(), dw.data_ptr<float>(), grid, D);
@@ -217,24 +248,36 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double 
 
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
                                        torch::Tensor inv,
-                                       c10::optional<torch::Tensor> dextra) {
+                                       c10::optional<torch::Tensor> dextra,
+                                       double p_drop, c10::optional<torch::Tensor> seed,
+                                       long salt) {
   auto dx = torch::empty_like(x);
   auto dw = torch::zeros({x.size(1)}, x.options().dtype(torch::kFloat));
   auto stream = at::hip::getCurrentHIPStream().stream();
   const torch::Tensor* de = (dextra.has_value() && dextra->defined()) ? &*dextra : nullptr;
+  const torch::Tensor* sp = (seed.has_value() && seed->defined()) ? &*seed : nullptr;
+  torch::Tensor ddelta;
+  torch::Tensor* dd = nullptr;
+  if (p_drop > 0.0) {
+    ddelta = torch::empty_like(x);
+    dd = &ddelta;
+  }
   if (x.scalar_type() == torch::kBFloat16)
-    launch_bwd<__hip_bfloat16>(dy, x, w, inv, de, dx, dw, stream);
+    launch_bwd<__hip_bfloat16>(dy, x, w, inv, de, dx, dw, dd, (float)p_drop, sp, salt, stream);
   else if (x.scalar_type() == torch::kHalf)
-    launch_bwd<__half>(dy, x, w, inv, de, dx, dw, stream);
+    launch_bwd<__half>(dy, x, w, inv, de, dx, dw, dd, (float)p_drop, sp, salt, stream);
   else if (x.scalar_type() == torch::kFloat)
-    launch_bwd<float>(dy, x, w, inv, de, dx, dw, stream);
+    launch_bwd<float>(dy, x, w, inv, de, dx, dw, dd, (float)p_drop, sp, salt, stream);
   else
     TORCH_CHECK(false, "rmsnorm: unsupported dtype");
+  if (dd) return {dx, dw, ddelta};
   return {dx, dw};
 }
 
 std::vector<torch::Tensor> add_rmsnorm_fwd(torch::Tensor x, torch::Tensor r,
-                                           torch::Tensor w, double eps) {
+                                           torch::Tensor w, double eps,
+                                           double p_drop, c10::optional<torch::Tensor> seed,
+                                           long salt) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous() && r.is_contiguous());
   auto s_out = torch::empty_like(x);
   auto y = torch::empty_like(x);
@@ -246,7 +289,9 @@ std::vector<torch::Tensor> add_rmsnorm_fwd(torch::Tensor x, torch::Tensor r,
                      dim3(std::min(rows, 2048)), dim3((D / VEC >= 256) ? 256 : 128),   \
                      0, stream, (const T*)x.data_ptr(), (const T*)r.data_ptr(),        \
                      (T*)s_out.data_ptr(), (T*)y.data_ptr(), (const T*)w.data_ptr(),   \
-                     inv.data_ptr<float>(), rows, D, (float)eps)
+                     inv.data_ptr<float>(), rows, D, (float)eps, (float)p_drop,        \
+                     (seed.has_value() && seed->defined())                             \
+                         ? seed->data_ptr<long>() : nullptr, salt)
   if (x.scalar_type() == torch::kBFloat16) ACASE(__hip_bfloat16, 8);
   else if (x.scalar_type() == torch::kHalf) ACASE(__half, 8);
   else if (x.scalar_type() == torch::kFloat) ACASE(float, 4);

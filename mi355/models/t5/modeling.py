@@ -300,28 +300,30 @@ class T5Stack(nn.Module):
         # resid/delta threading — each sublayer's input norm absorbs the
         # previous sublayer's residual add in one kernel
         resid, delta = x, None
+        pdrop = self.config.dropout_rate if self.training else 0.0
 
         def add_norm(ln, resid, delta):
             if delta is None:
                 return resid, ln(resid)
-            return ops.add_rms_norm(resid, delta, ln.weight, ln.variance_epsilon)
+            # the sublayer-output dropout is fused into the add+norm kernel
+            return ops.add_rms_norm(resid, delta, ln.weight, ln.variance_epsilon,
+                                    p_drop=pdrop)
 
         for i, blk in enumerate(self.block):
             self_cache = caches[i]["self"] if caches is not None else None
             cross_cache = caches[i]["cross"] if (caches is not None and self.is_decoder) else None
             sa = blk.layer[0]
             resid, normed = add_norm(sa.layer_norm, resid, delta)
-            delta = sa.dropout(sa.SelfAttention(
-                normed, position_bias=position_bias, kpad=kpad, causal=causal,
-                kv_cache=self_cache))
+            delta = sa.SelfAttention(normed, position_bias=position_bias, kpad=kpad,
+                                     causal=causal, kv_cache=self_cache)
             if self.is_decoder:
                 ca = blk.layer[1]
                 resid, normed = add_norm(ca.layer_norm, resid, delta)
-                delta = ca.dropout(ca.EncDecAttention(
-                    normed, kv_hidden=enc_hidden, kpad=enc_kpad, kv_cache=cross_cache))
+                delta = ca.EncDecAttention(normed, kv_hidden=enc_hidden,
+                                           kpad=enc_kpad, kv_cache=cross_cache)
             ff = blk.layer[-1]
             resid, normed = add_norm(ff.layer_norm, resid, delta)
-            delta = ff.dropout(ff.DenseReluDense(normed))
+            delta = ff.DenseReluDense(normed)
         _, normed = add_norm(self.final_layer_norm, resid, delta)
         return self.dropout(normed)
 

@@ -69,9 +69,9 @@ class _RMSNorm(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x2d, weight, inv_rms = ctx.saved_tensors
-        dx, dw = _load_ext().rmsnorm_bwd(dy.contiguous().view_as(x2d), x2d, weight,
-                                         inv_rms, None)
-        return dx.view_as(dy), dw.to(weight.dtype), None
+        res = _load_ext().rmsnorm_bwd(dy.contiguous().view_as(x2d), x2d, weight,
+                                      inv_rms, None, 0.0, None, 0)
+        return res[0].view_as(dy), res[1].to(weight.dtype), None
 
 
 def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6) -> torch.Tensor:
@@ -86,28 +86,43 @@ class _AddRMSNorm(torch.autograd.Function):
     residual stream grad (ds) and the normed-branch grad (dy)."""
 
     @staticmethod
-    def forward(ctx, x, r, weight, eps):
+    def forward(ctx, x, r, weight, eps, p_drop):
         x2d = x.contiguous().view(-1, x.shape[-1])
         r2d = r.contiguous().view(-1, x.shape[-1])
-        s, y, inv = _load_ext().add_rmsnorm_fwd(x2d, r2d, weight, eps)
+        # salt: layer identity (weight storage address) so every call site
+        # draws a distinct, step-stable dropout stream
+        salt = weight.data_ptr() if p_drop > 0 else 0
+        seed = _dropout_seed(x.device) if p_drop > 0 else None
+        s, y, inv = _load_ext().add_rmsnorm_fwd(x2d, r2d, weight, eps, p_drop,
+                                                seed, salt)
         ctx.save_for_backward(s, weight, inv)
         ctx.shape = x.shape
+        ctx.p_drop = p_drop
+        ctx.salt = salt
         return s.view(x.shape), y.view(x.shape)
 
     @staticmethod
     def backward(ctx, ds, dy):
         s, weight, inv = ctx.saved_tensors
         ds2d = ds.contiguous().view_as(s) if ds is not None else None
-        dx, dw = _load_ext().rmsnorm_bwd(dy.contiguous().view_as(s), s, weight,
-                                         inv, ds2d)
-        dx = dx.view(ctx.shape)
-        return dx, dx, dw.to(weight.dtype), None
+        seed = _dropout_seed(s.device) if ctx.p_drop > 0 else None
+        res = _load_ext().rmsnorm_bwd(dy.contiguous().view_as(s), s, weight,
+                                      inv, ds2d, ctx.p_drop, seed, ctx.salt)
+        dx = res[0].view(ctx.shape)
+        ddelta = res[2].view(ctx.shape) if len(res) > 2 else dx
+        return dx, ddelta, res[1].to(weight.dtype), None, None
 
 
-def add_rms_norm(x, r, weight, eps: float = 1e-6):
-    """Returns (s, y): s = x + r (the new residual stream), y = rmsnorm(s)."""
+def add_rms_norm(x, r, weight, eps: float = 1e-6, p_drop: float = 0.0):
+    """Returns (s, y): s = x + dropout(r) (the new residual stream),
+    y = rmsnorm(s). The delta-branch dropout is the pre-norm chain's
+    residual dropout, fused (SURVEY.md K7+K8+K2)."""
     if _use_hip(x):
-        return _AddRMSNorm.apply(x, r, weight, eps)
+        return _AddRMSNorm.apply(x, r, weight, eps, float(p_drop))
+    import torch.nn.functional as F
+
+    if p_drop > 0.0:
+        r = F.dropout(r, p=p_drop, training=True)
     s = x + r
     return s, torch_ref.rms_norm(s, weight, eps)
 
