@@ -73,6 +73,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   __shared__ T k_lds[BK * KLD];
   __shared__ T vt_lds[D * VLD];
   __shared__ T p_lds[4][16 * PLD];
+  __shared__ T bias_lds[HAS_BIAS ? BQ * PLD : 1];  // [q-local][k-local]
 
   const int bh = blockIdx.y;       // b * H + h
   const int b = bh / H, h = bh % H;
@@ -118,6 +119,13 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   constexpr int RPP = 256 / (D / 8);
   constexpr int NPASS = BK / RPP;  // row passes per tile (<= 2 for D<=64)
   VecT<T, 8> kst[NPASS], vst_r[NPASS];
+  // bias tile staging: [64 q rows x 8-col chunks] split (independent of the
+  // D-based K/V split — D=32 has fewer K passes but the same 64 bias rows)
+  const int bias_j = threadIdx.x / (BK / 8);
+  const int bias_c0 = (threadIdx.x % (BK / 8)) * 8;
+  constexpr int BRPP = 256 / (BK / 8);
+  constexpr int NBPASS = BQ / BRPP;
+  VecT<T, 8> bst[NBPASS];
 
   auto stage_load = [&](int k0) {
 #pragma unroll
@@ -131,6 +139,22 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         for (int u = 0; u < 8; ++u) { kst[p].data[u] = T(0.f); vst_r[p].data[u] = T(0.f); }
       }
     }
+    if (HAS_BIAS) {
+#pragma unroll
+      for (int p = 0; p < NBPASS; ++p) {
+        // coalesced vec8 along k of the (q-local, k0+...) bias rows; OOB
+        // rows clamp (the softmax mask discards them anyway)
+        const int qb = min(q0 + bias_j + p * BRPP, Sq - 1);
+        const int kb = k0 + bias_c0;
+        if (kb + 8 <= Sk) {
+          bst[p] = vload<T, 8>(bias_bh + (long)qb * Sk + kb);
+        } else {
+#pragma unroll
+          for (int u = 0; u < 8; ++u)
+            bst[p].data[u] = (kb + u < Sk) ? bias_bh[(long)qb * Sk + kb + u] : T(0.f);
+        }
+      }
+    }
   };
   auto stage_write = [&]() {
 #pragma unroll
@@ -139,6 +163,11 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       vstore<T, 8>(&k_lds[jj * KLD + stage_d0], kst[p]);
 #pragma unroll
       for (int u = 0; u < 8; ++u) vt_lds[(stage_d0 + u) * VLD + jj] = vst_r[p].data[u];
+    }
+    if (HAS_BIAS) {
+#pragma unroll
+      for (int p = 0; p < NBPASS; ++p)
+        vstore<T, 8>(&bias_lds[(bias_j + p * BRPP) * PLD + bias_c0], bst[p]);
     }
   };
 
@@ -178,7 +207,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         const int qr = qrow0 + r;
         float sv = s_acc[t][r];
         if (HAS_BIAS && k_ok && qr < Sq)
-          sv += DT<T>::to_f32(bias_bh[(long)qr * Sk + kc]);
+          sv += DT<T>::to_f32(bias_lds[(wid * 16 + lg * 4 + r) * PLD + t * 16 + lr]);
         sv += kp;
         if (!k_ok || qr >= Sq || (CAUSAL && kc > qr + diag_off)) sv = NEG;
         p_val[t][r] = sv;
