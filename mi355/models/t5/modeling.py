@@ -416,14 +416,26 @@ class T5ForConditionalGeneration(nn.Module):
     # -- greedy generate with preallocated KV cache -------- #
     @torch.no_grad()
     def generate(self, input_ids=None, attention_mask=None, max_new_tokens: int = 128,
-                 min_new_tokens: int = 0, eos_token_id=None, **_ignored):
+                 min_new_tokens: int = 0, eos_token_id=None, use_graph: bool | None = None,
+                 **_ignored):
+        """Greedy decode. On GPU with a fixed token budget
+        (min_new_tokens == max_new_tokens) the whole decode loop is captured
+        ONCE as a hipGraph and replayed per call — the loop is launch-bound
+        (~90 small kernels per token step) and the replay removes the
+        CPU-side gap. Disable with use_graph=False / MI355_GEN_GRAPH=0."""
+        if use_graph is None:
+            use_graph = os.environ.get("MI355_GEN_GRAPH", "1") != "0"
+        if (use_graph and input_ids.is_cuda and min_new_tokens == max_new_tokens
+                and not self.training):
+            try:
+                return self._generate_graphed(input_ids, attention_mask, max_new_tokens)
+            except Exception:
+                pass  # eager fallback below
+        return self._generate_eager(input_ids, attention_mask, max_new_tokens,
+                                    min_new_tokens, eos_token_id)
+
+    def _decode_caches(self, enc, b, max_new_tokens, dev, dt):
         cfg = self.config
-        eos = cfg.eos_token_id if eos_token_id is None else eos_token_id
-        b = input_ids.shape[0]
-        dev = input_ids.device
-        dt = self.shared.weight.dtype
-        kpad = _kpad_from_mask(attention_mask)
-        enc = self.encoder(input_ids=input_ids, kpad=kpad)
         h, dkv = cfg.num_heads, cfg.d_kv
         caches = []
         for blk in self.decoder.block:
@@ -436,6 +448,70 @@ class T5ForConditionalGeneration(nn.Module):
                          "len": 0},
                 "cross": {"k": ck, "v": cv, "static": True},
             })
+        return caches
+
+    @torch.no_grad()
+    def _generate_graphed(self, input_ids, attention_mask, max_new_tokens):
+        cfg = self.config
+        b = input_ids.shape[0]
+        dev = input_ids.device
+        dt = self.shared.weight.dtype
+        key = (b, input_ids.shape[1], max_new_tokens, str(dev))
+        state = getattr(self, "_gen_graph", None)
+        kpad = _kpad_from_mask(attention_mask)
+        enc = self.encoder(input_ids=input_ids, kpad=kpad)
+        if state is None or state["key"] != key or (kpad is None) != (state["kpad"] is None):
+            # build static buffers and capture the 128-step decode loop
+            caches = self._decode_caches(enc, b, max_new_tokens, dev, dt)
+            enc_static = enc.clone()
+            kpad_static = kpad.clone() if kpad is not None else None
+            out_tokens = torch.zeros(b, max_new_tokens, device=dev, dtype=torch.long)
+            attn0 = self.decoder.block[0].layer[0].SelfAttention
+            full_bias = attn0.compute_bias(max_new_tokens, max_new_tokens, dev).float()
+            bias_steps = [full_bias[:, :, t:t + 1, :t + 1].contiguous()
+                          for t in range(max_new_tokens)]
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                tok = torch.full((b, 1), cfg.decoder_start_token_id, device=dev,
+                                 dtype=torch.long)
+                for i in range(len(caches)):
+                    caches[i]["self"]["len"] = 0
+                for t in range(max_new_tokens):
+                    dec = self._decode_step(tok, enc_static, kpad_static, caches,
+                                            t, bias_steps[t].to(dt))
+                    if cfg.tie_word_embeddings:
+                        dec = dec * (self.model_dim**-0.5)
+                    logits = self.lm_head(dec[:, -1])
+                    nxt = logits.argmax(dim=-1)
+                    out_tokens[:, t] = nxt
+                    tok = nxt.unsqueeze(1)
+            state = {"key": key, "graph": g, "enc": enc_static, "kpad": kpad_static,
+                     "caches": caches, "out": out_tokens}
+            self._gen_graph = state
+        # refresh the static inputs, reset self-cache lengths, replay
+        state["enc"].copy_(enc)
+        if kpad is not None:
+            state["kpad"].copy_(kpad)
+        for i, blk in enumerate(self.decoder.block):
+            att = blk.layer[1].EncDecAttention
+            state["caches"][i]["cross"]["k"].copy_(att._shape(att.k(enc), input_ids.shape[0]))
+            state["caches"][i]["cross"]["v"].copy_(att._shape(att.v(enc), input_ids.shape[0]))
+            state["caches"][i]["self"]["len"] = 0
+        state["graph"].replay()
+        return state["out"].clone()
+
+    @torch.no_grad()
+    def _generate_eager(self, input_ids=None, attention_mask=None, max_new_tokens: int = 128,
+                        min_new_tokens: int = 0, eos_token_id=None, **_ignored):
+        cfg = self.config
+        eos = cfg.eos_token_id if eos_token_id is None else eos_token_id
+        b = input_ids.shape[0]
+        dev = input_ids.device
+        dt = self.shared.weight.dtype
+        kpad = _kpad_from_mask(attention_mask)
+        enc = self.encoder(input_ids=input_ids, kpad=kpad)
+        caches = self._decode_caches(enc, b, max_new_tokens, dev, dt)
         tokens = torch.full((b, 1), cfg.decoder_start_token_id, device=dev, dtype=torch.long)
         out_tokens = []
         finished = torch.zeros(b, dtype=torch.bool, device=dev)
