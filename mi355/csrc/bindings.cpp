@@ -39,6 +39,10 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
                                     bool causal, bool need_dbias,
                                     double p_drop, c10::optional<torch::Tensor> seed);
 torch::Tensor seg_postprocess(torch::Tensor logits, long Ho, long Wo);
+std::vector<torch::Tensor> attn_fwd_v2(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                       c10::optional<torch::Tensor> bias,
+                                       c10::optional<torch::Tensor> kpad, bool causal,
+                                       double p_drop, c10::optional<torch::Tensor> seed);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
@@ -57,5 +61,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rel_bias_bwd", &rel_bias_bwd);
   m.def("attn_bwd", &attn_bwd);
   m.def("seg_postprocess", &seg_postprocess);
+  m.def("attn_fwd_v2", &attn_fwd_v2);
   m.attr("__gfx__") = "gfx950";
 }

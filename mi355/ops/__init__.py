@@ -189,7 +189,12 @@ class _Attention(torch.autograd.Function):
         bias_c = bias.contiguous() if bias is not None else None
         kpad_c = kpad.contiguous().float() if kpad is not None else None
         seed = _dropout_seed(q.device) if p_drop > 0 else None
-        o, lse = _load_ext().attn_fwd(q, k, v, bias_c, kpad_c, causal, p_drop, seed)
+        ext = _load_ext()
+        if (q.shape[-1] == 64 and hasattr(ext, "attn_fwd_v2")
+                and os.environ.get("MI355_ATTN_V2", "1") != "0"):
+            o, lse = ext.attn_fwd_v2(q, k, v, bias_c, kpad_c, causal, p_drop, seed)
+        else:
+            o, lse = ext.attn_fwd(q, k, v, bias_c, kpad_c, causal, p_drop, seed)
         ctx.save_for_backward(q, k, v, bias_c, kpad_c, o, lse)
         ctx.causal = causal
         ctx.p_drop = p_drop
