@@ -52,7 +52,7 @@ template <> struct MF32<__half> {
 };
 
 constexpr int BQ2 = 128;  // q rows per block (4 waves x 32)
-constexpr int BK2 = 64;   // keys per staging round (2 MFMA sub-tiles of 32)
+constexpr int BK2 = 32;   // keys per tile
 constexpr float NEG2 = -1e30f;
 
 struct S3 { long sb, sh, ss; };
@@ -111,27 +111,23 @@ __global__ __launch_bounds__(256) void attn_fwd_v2_kernel(
   const long long seed = seed_ptr ? *seed_ptr : 0;
   const float keep_inv = p_drop > 0.f ? 1.f / (1.f - p_drop) : 1.f;
 
-  // T14 staging: K rows ([64][D] image) + V cols (vt [D][64]) + bias
-  const int sj = threadIdx.x / (D / 8);            // k-row (8-elem chunks)
+  // T14 staging: K rows ([32][D] image) + V cols (vt [D][BK2] image) + bias
+  const int sj = threadIdx.x / (D / 8);            // k-row (32 rows, 8 chunks)
   const int sd = (threadIdx.x % (D / 8)) * 8;
-  constexpr int KRPP = 256 / (D / 8);              // 32 k-rows per pass
-  VecT<T, 8> kreg[BK2 / KRPP], vreg[BK2 / KRPP];
-  const int bj = threadIdx.x / (BK2 / 8);          // bias q-row (8-col chunks)
+  VecT<T, 8> kreg, vreg;
+  const int bj = threadIdx.x / (BK2 / 8);          // bias q-row (128 rows, 4 chunks)
   const int bc = (threadIdx.x % (BK2 / 8)) * 8;
-  constexpr int BRPP = 256 / (BK2 / 8);            // 32 bias q-rows per pass
+  constexpr int BRPP = 256 / (BK2 / 8);            // 64 rows per pass
   VecT<T, 8> breg[BQ2 / BRPP];
 
   auto stage_load = [&](int k0) {
+    const int kr = k0 + sj;
+    if (kr < Sk) {
+      kreg = vload<T, 8>(kbase + (long)kr * ks.ss + sd);
+      vreg = vload<T, 8>(vbase + (long)kr * vs.ss + sd);
+    } else {
 #pragma unroll
-    for (int p = 0; p < BK2 / KRPP; ++p) {
-      const int kr = k0 + sj + p * KRPP;
-      if (kr < Sk) {
-        kreg[p] = vload<T, 8>(kbase + (long)kr * ks.ss + sd);
-        vreg[p] = vload<T, 8>(vbase + (long)kr * vs.ss + sd);
-      } else {
-#pragma unroll
-        for (int u = 0; u < 8; ++u) { kreg[p].data[u] = T(0.f); vreg[p].data[u] = T(0.f); }
-      }
+      for (int u = 0; u < 8; ++u) { kreg.data[u] = T(0.f); vreg.data[u] = T(0.f); }
     }
     if (HAS_BIAS) {
 #pragma unroll
@@ -149,13 +145,9 @@ __global__ __launch_bounds__(256) void attn_fwd_v2_kernel(
     }
   };
   auto stage_write = [&]() {
+    vstore<T, 8>(&k_lds[sj * KLD + sd], kreg);
 #pragma unroll
-    for (int p = 0; p < BK2 / KRPP; ++p) {
-      const int jj = sj + p * KRPP;
-      vstore<T, 8>(&k_lds[jj * KLD + sd], kreg[p]);
-#pragma unroll
-      for (int u = 0; u < 8; ++u) vt_lds[(sd + u) * VLD + jj] = vreg[p].data[u];
-    }
+    for (int u = 0; u < 8; ++u) vt_lds[(sd + u) * VLD + sj] = vreg.data[u];
     if (HAS_BIAS) {
 #pragma unroll
       for (int p = 0; p < BQ2 / BRPP; ++p)
@@ -170,18 +162,13 @@ __global__ __launch_bounds__(256) void attn_fwd_v2_kernel(
     __syncthreads();
     if (k0 + BK2 < k_end) stage_load(k0 + BK2);
 
-#pragma unroll
-    for (int sub = 0; sub < 2; ++sub) {
-    const int ks0 = k0 + sub * 32;
-    if (CAUSAL && ks0 >= k_end) break;
     // ---- S^T = K Q^T: A = K rows, B = Q^T ----
     f32x16 st;
 #pragma unroll
     for (int r = 0; r < 16; ++r) st[r] = 0.f;
 #pragma unroll
     for (int s = 0; s < 4; ++s) {
-      FR a = *reinterpret_cast<const FR*>(
-          &k_lds[(sub * 32 + lq) * KLD + s * 16 + half * 8]);
+      FR a = *reinterpret_cast<const FR*>(&k_lds[lq * KLD + s * 16 + half * 8]);
       st = MF32<T>::mma(a, qfrag[s], st);
     }
 
@@ -192,10 +179,9 @@ __global__ __launch_bounds__(256) void attn_fwd_v2_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int kk = (r & 3) + 8 * (r >> 2) + 4 * half;
-      const int kc = ks0 + kk;
+      const int kc = k0 + kk;
       float sv = st[r];
-      if (HAS_BIAS)
-        sv += DT<T>::to_f32(bias_lds[(wid * 32 + lq) * BLD + sub * 32 + kk]);
+      if (HAS_BIAS) sv += DT<T>::to_f32(bias_lds[(wid * 32 + lq) * BLD + kk]);
       if (HAS_KPAD && kc < Sk) sv += kpad_b[kc];
       if (kc >= Sk || !q_ok || (CAUSAL && kc > my_q + diag_off)) sv = NEG2;
       pv[r] = sv;
@@ -223,7 +209,7 @@ __global__ __launch_bounds__(256) void attn_fwd_v2_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int kk = (r & 3) + 8 * (r >> 2) + 4 * half;
-        const long long idx = ((long long)bh * Sq + my_q) * Sk + (ks0 + kk);
+        const long long idx = ((long long)bh * Sq + my_q) * Sk + (k0 + kk);
         pv[r] = (hash_u01(seed, idx) >= p_drop) ? pv[r] * keep_inv : 0.f;
       }
     }
@@ -254,14 +240,13 @@ __global__ __launch_bounds__(256) void attn_fwd_v2_kernel(
 #pragma unroll
       for (int g = 0; g < 2; ++g) {
         FR a = *reinterpret_cast<const FR*>(
-            &vt_lds[(dt * 32 + lq) * VLD + sub * 32 + g * 16 + half * 8]);
+            &vt_lds[(dt * 32 + lq) * VLD + g * 16 + half * 8]);
         struct U4 { unsigned x[4]; } u4{pb[g * 4], pb[g * 4 + 1],
                                         pb[g * 4 + 2], pb[g * 4 + 3]};
         FR bfr = __builtin_bit_cast(FR, u4);
         o_acc[dt] = MF32<T>::mma(a, bfr, o_acc[dt]);
       }
     }
-    }  // sub
   }
 
   // ---- epilogue: normalize, store O rows (4-elem packs) + LSE ----
