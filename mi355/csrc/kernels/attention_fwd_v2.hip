@@ -166,22 +166,30 @@ __global__ __launch_bounds__(256) void attn_fwd_v2_kernel(
     f32x16 st;
 #pragma unroll
     for (int r = 0; r < 16; ++r) st[r] = 0.f;
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int s = 0; s < 4; ++s) {
       FR a = *reinterpret_cast<const FR*>(&k_lds[lq * KLD + s * 16 + half * 8]);
       st = MF32<T>::mma(a, qfrag[s], st);
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // ---- mask + bias; in-lane rows hold keys kk(r) = (r&3)+8*(r>>2)+4*half
     float pv[16];
     float tile_max = -INFINITY;
     const bool q_ok = my_q < Sq;
+    VecT<T, 4> bias4[4];
+    if (HAS_BIAS) {
+#pragma unroll
+      for (int g = 0; g < 4; ++g)  // 4 consecutive keys per reg group -> b64
+        bias4[g] = vload<T, 4>(&bias_lds[(wid * 32 + lq) * BLD + 8 * g + 4 * half]);
+    }
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int kk = (r & 3) + 8 * (r >> 2) + 4 * half;
       const int kc = k0 + kk;
       float sv = st[r];
-      if (HAS_BIAS) sv += DT<T>::to_f32(bias_lds[(wid * 32 + lq) * BLD + kk]);
+      if (HAS_BIAS) sv += DT<T>::to_f32(bias4[r >> 2].data[r & 3]);
       if (HAS_KPAD && kc < Sk) sv += kpad_b[kc];
       if (kc >= Sk || !q_ok || (CAUSAL && kc > my_q + diag_off)) sv = NEG2;
       pv[r] = sv;
@@ -235,6 +243,7 @@ __global__ __launch_bounds__(256) void attn_fwd_v2_kernel(
     }
 
     // ---- O^T += V^T P^T : A = V^T (2 d-tiles), accumulate over 32 keys ----
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int dt = 0; dt < 2; ++dt) {
 #pragma unroll
@@ -247,6 +256,7 @@ __global__ __launch_bounds__(256) void attn_fwd_v2_kernel(
         o_acc[dt] = MF32<T>::mma(a, bfr, o_acc[dt]);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
   }
 
   // ---- epilogue: normalize, store O rows (4-elem packs) + LSE ----
