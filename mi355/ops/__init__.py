@@ -215,8 +215,13 @@ class _Attention(torch.autograd.Function):
             if do.stride(-1) != 1:
                 do = do.contiguous()
             seed = _dropout_seed(q.device) if ctx.p_drop > 0 else None
-            res = ext.attn_bwd(do, q, k, v, o, lse, bias, kpad, ctx.causal,
-                               ctx.bias_requires_grad, ctx.p_drop, seed)
+            if (q.shape[-1] == 64 and hasattr(ext, "attn_bwd_v2")
+                    and os.environ.get("MI355_ATTN_V2", "1") != "0"):
+                res = ext.attn_bwd_v2(do, q, k, v, o, lse, bias, kpad, ctx.causal,
+                                      ctx.bias_requires_grad, ctx.p_drop, seed)
+            else:
+                res = ext.attn_bwd(do, q, k, v, o, lse, bias, kpad, ctx.causal,
+                                   ctx.bias_requires_grad, ctx.p_drop, seed)
             dq, dk, dv = res[0], res[1], res[2]
             dbias = res[3] if len(res) > 3 else None
         else:  # composed bwd via recompute (hipBLASLt batched GEMMs)
