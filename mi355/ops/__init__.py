@@ -215,8 +215,11 @@ class _Attention(torch.autograd.Function):
             if do.stride(-1) != 1:
                 do = do.contiguous()
             seed = _dropout_seed(q.device) if ctx.p_drop > 0 else None
+            # bwd v2 is numerically validated but measures SLOWER than v1
+            # (270 regs -> 1 wave/SIMD + per-wave dQ slab traffic); opt-in
+            # until the round-2 register-pressure work
             if (q.shape[-1] == 64 and hasattr(ext, "attn_bwd_v2")
-                    and os.environ.get("MI355_ATTN_V2", "1") != "0"):
+                    and os.environ.get("MI355_ATTN_BWD_V2", "0") == "1"):
                 res = ext.attn_bwd_v2(do, q, k, v, o, lse, bias, kpad, ctx.causal,
                                       ctx.bias_requires_grad, ctx.p_drop, seed)
             else:
