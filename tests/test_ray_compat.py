@@ -1,0 +1,51 @@
+"""The reference notebooks' `ray.*` call surface resolves through the compat
+namespace (SURVEY.md §2.2 call sites)."""
+import pytest
+
+
+def test_ray_core_surface():
+    import ray
+
+    ray.init(num_workers=2, num_gpus=0)
+    try:
+        assert ray.is_initialized()
+
+        @ray.remote
+        def f(x):
+            return x + 1
+
+        ref = ray.put(41)
+        assert ray.get(f.remote(ray.get(ref))) == 42
+        ready, _ = ray.wait([f.remote(1)], num_returns=1)
+        assert ready
+    finally:
+        ray.shutdown()
+    assert not ray.is_initialized()
+
+
+def test_ray_air_train_tune_imports():
+    from ray import tune
+    from ray.air import Checkpoint, Result, ResultGrid
+    from ray.air.config import CheckpointConfig, RunConfig, ScalingConfig
+    from ray.data import from_items
+    from ray.data.preprocessors import BatchMapper, MinMaxScaler
+    from ray.train.batch_predictor import BatchPredictor
+    from ray.train.huggingface import HuggingFaceCheckpoint, HuggingFaceTrainer
+    from ray.train.predictor import Predictor
+    from ray.train.xgboost import XGBoostPredictor, XGBoostTrainer
+    from ray.tune import ASHAScheduler, TuneConfig, Tuner
+    from ray.util import ActorPool
+
+    sc = ScalingConfig(num_workers=2, use_gpu=False)
+    assert sc.num_workers == 2
+    assert tune.choice([1, 2]).categories == [1, 2]
+    ds = from_items([{"a": 1}, {"a": 2}])
+    assert ds.count() == 2
+
+
+def test_ray_serve_surface():
+    from ray import serve
+
+    d = serve.PredictorDeployment.options(name="X", num_replicas=2,
+                                          route_prefix="/x")
+    assert d.num_replicas == 2 and d.route_prefix == "/x"
