@@ -71,7 +71,7 @@ __global__ void delta_kernel_v2(const T* __restrict__ dout, const T* __restrict_
 }
 
 template <typename T, bool CAUSAL, bool HAS_BIAS, bool HAS_KPAD, bool NEED_DBIAS>
-__global__ __launch_bounds__(256) void attn_bwd_v2_kernel(
+__global__ __launch_bounds__(256, 2) void attn_bwd_v2_kernel(
     const T* __restrict__ dout, const T* __restrict__ q, const T* __restrict__ k,
     const T* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
     const T* __restrict__ bias, const float* __restrict__ kpad,
