@@ -206,6 +206,13 @@ class SegformerDecodeHead(nn.Module):
         self.dropout = nn.Dropout(cfg.classifier_dropout_prob)
         self.classifier = nn.Conv2d(cfg.decoder_hidden_size, cfg.num_labels, 1)
 
+    def _apply(self, fn, recurse=True):
+        # keep BatchNorm params/stats in fp32 even when the model is cast to
+        # bf16/fp16 (its forward runs in fp32 regardless)
+        out = super()._apply(fn, recurse)
+        self.batch_norm.float()
+        return out
+
     def forward(self, features):
         b = features[0].shape[0]
         size = features[0].shape[2:]
