@@ -125,3 +125,35 @@ def test_sort_and_unique(taxi_like):
     s = taxi_like.sort("fare_amount", descending=True).take(2)
     assert s[0]["fare_amount"] >= s[1]["fare_amount"]
     assert taxi_like.unique("passenger_count") == [1, 2, 3, 4]
+
+
+def test_map_batches_via_runtime_tasks():
+    """Block transforms fan out as runtime tasks when mi355.init is active
+    (transforms run as C2 tasks)."""
+    import mi355
+
+    mi355.init(num_workers=2, num_gpus=0)
+    try:
+        ds = mdata.from_items([{"x": float(i)} for i in range(64)]).repartition(4)
+        out = ds.map_batches(lambda df: df.assign(y=df.x * 3),
+                             batch_format="pandas", batch_size=16)
+        assert out.to_pandas()["y"].tolist() == [3.0 * i for i in range(64)]
+    finally:
+        mi355.shutdown()
+
+
+def test_map_batches_actor_pool_strategy():
+    from mi355.data import ActorPoolStrategy
+
+    class AddState:
+        def __init__(self, base):
+            self.base = base
+
+        def __call__(self, df):
+            return df.assign(z=df.x + self.base)
+
+    ds = mdata.from_items([{"x": float(i)} for i in range(20)]).repartition(2)
+    out = ds.map_batches(AddState, compute=ActorPoolStrategy(size=2),
+                         fn_constructor_args=(100,), batch_format="pandas",
+                         batch_size=10)
+    assert sorted(out.to_pandas()["z"].tolist()) == [100.0 + i for i in range(20)]
