@@ -49,6 +49,12 @@ std::vector<torch::Tensor> attn_bwd_v2(torch::Tensor dout, torch::Tensor q, torc
                                        c10::optional<torch::Tensor> kpad,
                                        bool causal, bool need_dbias,
                                        double p_drop, c10::optional<torch::Tensor> seed);
+std::vector<torch::Tensor> attn_bwd_v3(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                                       torch::Tensor v, torch::Tensor o, torch::Tensor lse,
+                                       c10::optional<torch::Tensor> bias,
+                                       c10::optional<torch::Tensor> kpad,
+                                       bool causal, bool need_dbias,
+                                       double p_drop, c10::optional<torch::Tensor> seed);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
@@ -69,5 +75,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("seg_postprocess", &seg_postprocess);
   m.def("attn_fwd_v2", &attn_fwd_v2);
   m.def("attn_bwd_v2", &attn_bwd_v2);
+  m.def("attn_bwd_v3", &attn_bwd_v3);
   m.attr("__gfx__") = "gfx950";
 }

@@ -1,0 +1,569 @@
+// Flash attention backward v3: TWO plain-store kernels (FA2-style split),
+// both on the swapped-operand 32x32 MFMA layout of attention_fwd_v2.hip.
+//
+//  * dkv kernel — block OWNS 128 keys (4 waves x 32). K/V B-fragments live
+//    in registers for the whole kernel; loop over 32-query tiles:
+//      S = Q K^T, dP = dO V^T  (C: q rows in regs, k col = lane)
+//      dV^T += dO^T repack(P), dK^T += Q^T repack(dS); dBias plain stores.
+//    dK/dV written directly (exclusive key ownership) — no slabs.
+//  * dq kernel — block OWNS 128 queries; loop over 32-key tiles:
+//      S^T = K Q^T, dP^T = V dO^T (C: k rows in regs, q col = lane — the
+//      forward orientation, so lse/delta are ONE scalar per lane and the
+//      dS^T repack feeds dQ^T += K^T dS^T directly).
+//    dQ written directly in the input dtype — no fp32 slabs, no atomics.
+// Total MFMA = 4/3 of the monolithic bwd (S recomputed twice) but zero
+// slab traffic and 2+ waves/SIMD.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "../common.h"
+
+namespace {
+
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef _Float16 f16x8 __attribute__((ext_vector_type(8)));
+
+template <typename T> struct M3;
+template <> struct M3<__hip_bfloat16> {
+  using frag = bf16x8;
+  static DEVINL f32x16 mma(frag a, frag b, f32x16 c) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+  }
+  static DEVINL unsigned pack2(float lo, float hi) {
+    union { __hip_bfloat162 h2; unsigned u; } u;
+    u.h2 = __hip_bfloat162(__float2bfloat16(lo), __float2bfloat16(hi));
+    return u.u;
+  }
+};
+template <> struct M3<__half> {
+  using frag = f16x8;
+  static DEVINL f32x16 mma(frag a, frag b, f32x16 c) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_f16(a, b, c, 0, 0, 0);
+  }
+  static DEVINL unsigned pack2(float lo, float hi) {
+    union { __half2 h2; unsigned u; } u;
+    u.h2 = __half2(__float2half(lo), __float2half(hi));
+    return u.u;
+  }
+};
+
+struct S3c { long sb, sh, ss; };
+
+// repack 16 in-reg values (axis layout (r&3)+8*(r>>2)+4*half) into two
+// B-fragments' dwords via cvt_pk + permlane32_swap (guide T12)
+template <typename T>
+DEVINL void repack16(const float* x, unsigned* pb) {
+#pragma unroll
+  for (int g = 0; g < 2; ++g) {
+    const int o = g * 8;
+    unsigned a0 = M3<T>::pack2(x[o + 0], x[o + 1]);
+    unsigned b0 = M3<T>::pack2(x[o + 4], x[o + 5]);
+    unsigned a1 = M3<T>::pack2(x[o + 2], x[o + 3]);
+    unsigned b1 = M3<T>::pack2(x[o + 6], x[o + 7]);
+    auto r0 = __builtin_amdgcn_permlane32_swap(a0, b0, false, false);
+    auto r1 = __builtin_amdgcn_permlane32_swap(a1, b1, false, false);
+    pb[g * 4 + 0] = r0[0];
+    pb[g * 4 + 1] = r1[0];
+    pb[g * 4 + 2] = r0[1];
+    pb[g * 4 + 3] = r1[1];
+  }
+}
+
+template <typename T>
+DEVINL typename M3<T>::frag frag_of(const unsigned* pb) {
+  struct U4 { unsigned x[4]; } u{pb[0], pb[1], pb[2], pb[3]};
+  return __builtin_bit_cast(typename M3<T>::frag, u);
+}
+
+template <typename T>
+__global__ void delta_kernel_v3(const T* __restrict__ dout, const T* __restrict__ o,
+                                float* __restrict__ delta, int H, int Sq, int D,
+                                S3c dos, S3c os) {
+  const long row = (long)blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int b = blockIdx.y;
+  if (row >= (long)H * Sq) return;
+  const int h = row / Sq, qr = row % Sq;
+  const T* dor = dout + b * dos.sb + h * dos.sh + (long)qr * dos.ss;
+  const T* orr = o + b * os.sb + h * os.sh + (long)qr * os.ss;
+  float acc = 0.f;
+  for (int d = lane; d < D; d += WAVE)
+    acc += DT<T>::to_f32(dor[d]) * DT<T>::to_f32(orr[d]);
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) delta[((long)b * H + h) * Sq + qr] = acc;
+}
+
+// ======================== dK / dV / dBias kernel ========================= //
+template <typename T, bool CAUSAL, bool HAS_BIAS, bool HAS_KPAD, bool NEED_DBIAS>
+__global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
+    const T* __restrict__ dout, const T* __restrict__ q, const T* __restrict__ k,
+    const T* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
+    const T* __restrict__ bias, const float* __restrict__ kpad,
+    T* __restrict__ dk, T* __restrict__ dv, float* __restrict__ dbias32,
+    int B, int H, int Sq, int Sk,
+    S3c qs, S3c ks, S3c vs, S3c dos, S3c dks, S3c dvs,
+    long bias_sb, long bias_sh, float p_drop, const long* __restrict__ seed_ptr) {
+  constexpr int D = 64;
+  using FR = typename M3<T>::frag;
+  constexpr int QLD = D + 8;
+  constexpr int TLD = 32 + 8;
+  constexpr int KBk = 128;
+
+  __shared__ T q_img[32 * QLD];
+  __shared__ T do_img[32 * QLD];
+  __shared__ T qT_img[D * TLD];
+  __shared__ T doT_img[D * TLD];
+  __shared__ T bias_img[HAS_BIAS ? 32 * (KBk + 8) : 1];
+  __shared__ float lse_lds[32], delta_lds[32];
+
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int kb0 = blockIdx.x * KBk;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int lk = lane & 31;
+  const int half = lane >> 5;
+  const int my_k = kb0 + wid * 32 + lk;
+  const int my_k_c = min(my_k, Sk - 1);
+
+  const T* qbase = q + (long)b * qs.sb + (long)h * qs.sh;
+  const T* kbase = k + (long)b * ks.sb + (long)h * ks.sh;
+  const T* vbase = v + (long)b * vs.sb + (long)h * vs.sh;
+  const T* dobase = dout + (long)b * dos.sb + (long)h * dos.sh;
+  T* dkbase = dk + (long)b * dks.sb + (long)h * dks.sh;
+  T* dvbase = dv + (long)b * dvs.sb + (long)h * dvs.sh;
+  const float* lse_bh = lse + ((long)b * H + h) * Sq;
+  const float* delta_bh = delta + ((long)b * H + h) * Sq;
+  const T* bias_bh = HAS_BIAS ? bias + (long)b * bias_sb + (long)h * bias_sh : nullptr;
+  float* dbias_h = NEED_DBIAS ? dbias32 + ((long)b * H + h) * (long)Sq * Sk : nullptr;
+
+  FR kfrag[4], vfrag[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s) {
+    kfrag[s] = *reinterpret_cast<const FR*>(kbase + (long)my_k_c * ks.ss + s * 16 + half * 8);
+    vfrag[s] = *reinterpret_cast<const FR*>(vbase + (long)my_k_c * vs.ss + s * 16 + half * 8);
+  }
+
+  f32x16 dv_acc[2], dk_acc[2];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    dv_acc[0][r] = 0.f; dv_acc[1][r] = 0.f;
+    dk_acc[0][r] = 0.f; dk_acc[1][r] = 0.f;
+  }
+
+  const int diag_off = Sk - Sq;
+  int qt_start = 0;
+  if (CAUSAL) qt_start = max(0, (kb0 - diag_off) / 32);
+  const long long seed = seed_ptr ? *seed_ptr : 0;
+  const float keep_inv = p_drop > 0.f ? 1.f / (1.f - p_drop) : 1.f;
+  const float kp = (HAS_KPAD && my_k < Sk) ? kpad[(long)b * Sk + my_k] : 0.f;
+
+  const int sj = threadIdx.x / (D / 8);
+  const int sd = (threadIdx.x % (D / 8)) * 8;
+  const int bj = threadIdx.x / (KBk / 8);
+  const int bc = (threadIdx.x % (KBk / 8)) * 8;
+  constexpr int BRPP = 256 / (KBk / 8);
+
+  if (NEED_DBIAS && CAUSAL && my_k < Sk) {  // zero skipped rows' columns
+    for (int qr = 0; qr < qt_start * 32 && qr < Sq; ++qr)
+      dbias_h[(long)qr * Sk + my_k] = 0.f;
+  }
+
+  for (int qt = qt_start; qt * 32 < Sq; ++qt) {
+    const int q0 = qt * 32;
+    __syncthreads();
+    {
+      const int qr = min(q0 + sj, Sq - 1);
+      VecT<T, 8> qv = vload<T, 8>(qbase + (long)qr * qs.ss + sd);
+      VecT<T, 8> dov = vload<T, 8>(dobase + (long)qr * dos.ss + sd);
+      vstore<T, 8>(&q_img[sj * QLD + sd], qv);
+      vstore<T, 8>(&do_img[sj * QLD + sd], dov);
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        qT_img[(sd + u) * TLD + sj] = qv.data[u];
+        doT_img[(sd + u) * TLD + sj] = dov.data[u];
+      }
+      if (threadIdx.x < 32) {
+        const int qr2 = min(q0 + (int)threadIdx.x, Sq - 1);
+        lse_lds[threadIdx.x] = lse_bh[qr2];
+        delta_lds[threadIdx.x] = delta_bh[qr2];
+      }
+      if (HAS_BIAS) {
+#pragma unroll
+        for (int p = 0; p < 32 / BRPP; ++p) {
+          const int qb = min(q0 + bj + p * BRPP, Sq - 1);
+          const int kbc = kb0 + bc;
+          VecT<T, 8> bv;
+          if (kbc + 8 <= Sk) {
+            bv = vload<T, 8>(bias_bh + (long)qb * Sk + kbc);
+          } else {
+#pragma unroll
+            for (int u = 0; u < 8; ++u)
+              bv.data[u] = (kbc + u < Sk) ? bias_bh[(long)qb * Sk + kbc + u] : T(0.f);
+          }
+          vstore<T, 8>(&bias_img[(bj + p * BRPP) * (KBk + 8) + bc], bv);
+        }
+      }
+    }
+    __syncthreads();
+
+    f32x16 sacc, dpacc;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) { sacc[r] = 0.f; dpacc[r] = 0.f; }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      FR aq = *reinterpret_cast<const FR*>(&q_img[lk * QLD + s * 16 + half * 8]);
+      sacc = M3<T>::mma(aq, kfrag[s], sacc);
+      FR ad = *reinterpret_cast<const FR*>(&do_img[lk * QLD + s * 16 + half * 8]);
+      dpacc = M3<T>::mma(ad, vfrag[s], dpacc);
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    float pp[16], ds[16];
+    const bool k_ok = my_k < Sk;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int qq = (r & 3) + 8 * (r >> 2) + 4 * half;
+      const int qr = q0 + qq;
+      const bool q_ok = qr < Sq;
+      float sv = sacc[r];
+      if (HAS_BIAS) sv += DT<T>::to_f32(bias_img[qq * (KBk + 8) + wid * 32 + lk]);
+      sv += kp;
+      const float l = q_ok ? lse_lds[qq] : -INFINITY;
+      bool masked = !k_ok || !q_ok || (CAUSAL && my_k > qr + diag_off);
+      float p = (masked || l == -INFINITY) ? 0.f : __expf(sv - l);
+      float mfac = 1.f;
+      if (p_drop > 0.f) {
+        const long long idx = ((long long)bh * Sq + qr) * Sk + my_k;
+        mfac = (hash_u01(seed, idx) >= p_drop) ? keep_inv : 0.f;
+      }
+      const float dl = q_ok ? delta_lds[qq] : 0.f;
+      ds[r] = p * (dpacc[r] * mfac - dl);
+      pp[r] = p * mfac;
+      if (NEED_DBIAS && q_ok && k_ok)
+        dbias_h[(long)qr * Sk + my_k] = ds[r];
+    }
+
+    unsigned pbP[8], pbS[8];
+    repack16<T>(pp, pbP);
+    repack16<T>(ds, pbS);
+
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int dt = 0; dt < 2; ++dt) {
+#pragma unroll
+      for (int g = 0; g < 2; ++g) {
+        FR adoT = *reinterpret_cast<const FR*>(
+            &doT_img[(dt * 32 + lk) * TLD + g * 16 + half * 8]);
+        dv_acc[dt] = M3<T>::mma(adoT, frag_of<T>(&pbP[g * 4]), dv_acc[dt]);
+        FR aqT = *reinterpret_cast<const FR*>(
+            &qT_img[(dt * 32 + lk) * TLD + g * 16 + half * 8]);
+        dk_acc[dt] = M3<T>::mma(aqT, frag_of<T>(&pbS[g * 4]), dk_acc[dt]);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+  }
+
+  if (my_k < Sk) {
+#pragma unroll
+    for (int dt = 0; dt < 2; ++dt)
+#pragma unroll
+      for (int grp = 0; grp < 4; ++grp) {
+        const int d0 = dt * 32 + grp * 8 + half * 4;
+        VecT<T, 4> kv, vv;
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          kv.data[u] = DT<T>::from_f32(dk_acc[dt][grp * 4 + u]);
+          vv.data[u] = DT<T>::from_f32(dv_acc[dt][grp * 4 + u]);
+        }
+        vstore<T, 4>(dkbase + (long)my_k * dks.ss + d0, kv);
+        vstore<T, 4>(dvbase + (long)my_k * dvs.ss + d0, vv);
+      }
+  }
+}
+
+// ============================== dQ kernel ================================ //
+template <typename T, bool CAUSAL, bool HAS_BIAS, bool HAS_KPAD>
+__global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
+    const T* __restrict__ dout, const T* __restrict__ q, const T* __restrict__ k,
+    const T* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
+    const T* __restrict__ bias, const float* __restrict__ kpad,
+    T* __restrict__ dq,
+    int B, int H, int Sq, int Sk,
+    S3c qs, S3c ks, S3c vs, S3c dos, S3c dqs,
+    long bias_sb, long bias_sh, float p_drop, const long* __restrict__ seed_ptr) {
+  constexpr int D = 64;
+  using FR = typename M3<T>::frag;
+  constexpr int KLD = D + 8;
+  constexpr int TLD = 32 + 8;
+  constexpr int QB = 128;
+
+  __shared__ T k_img[32 * KLD];    // K rows (A of S^T)
+  __shared__ T v_img[32 * KLD];    // V rows (A of dP^T)
+  __shared__ T ktT_img[D * TLD];   // K^T (A of dQ^T)
+  __shared__ T bias_img[HAS_BIAS ? QB * TLD : 1];  // [q][k-tile]
+  __shared__ float kpad_lds[32];
+
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int q0 = blockIdx.x * QB;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int lq = lane & 31;
+  const int half = lane >> 5;
+  const int my_q = q0 + wid * 32 + lq;
+  const int my_q_c = min(my_q, Sq - 1);
+
+  const T* qbase = q + (long)b * qs.sb + (long)h * qs.sh;
+  const T* kbase = k + (long)b * ks.sb + (long)h * ks.sh;
+  const T* vbase = v + (long)b * vs.sb + (long)h * vs.sh;
+  const T* dobase = dout + (long)b * dos.sb + (long)h * dos.sh;
+  T* dqbase = dq + (long)b * dqs.sb + (long)h * dqs.sh;
+  const T* bias_bh = HAS_BIAS ? bias + (long)b * bias_sb + (long)h * bias_sh : nullptr;
+  const float* kpad_b = HAS_KPAD ? kpad + (long)b * Sk : nullptr;
+
+  FR qfrag[4], dofrag[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s) {
+    qfrag[s] = *reinterpret_cast<const FR*>(qbase + (long)my_q_c * qs.ss + s * 16 + half * 8);
+    dofrag[s] = *reinterpret_cast<const FR*>(dobase + (long)my_q_c * dos.ss + s * 16 + half * 8);
+  }
+  const float my_lse = (my_q < Sq) ? lse[((long)bh) * Sq + my_q] : -INFINITY;
+  const float my_delta = (my_q < Sq) ? delta[((long)bh) * Sq + my_q] : 0.f;
+
+  f32x16 dq_acc[2];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) { dq_acc[0][r] = 0.f; dq_acc[1][r] = 0.f; }
+
+  const int diag_off = Sk - Sq;
+  const int k_end = CAUSAL ? min(Sk, q0 + QB + diag_off) : Sk;
+  const long long seed = seed_ptr ? *seed_ptr : 0;
+  const float keep_inv = p_drop > 0.f ? 1.f / (1.f - p_drop) : 1.f;
+
+  const int sj = threadIdx.x / (D / 8);
+  const int sd = (threadIdx.x % (D / 8)) * 8;
+  const int bj = threadIdx.x / (32 / 8);
+  const int bc = (threadIdx.x % (32 / 8)) * 8;
+  constexpr int BRPP = 256 / (32 / 8);  // 64 bias q-rows per pass
+
+  for (int k0 = 0; k0 < k_end; k0 += 32) {
+    __syncthreads();
+    {
+      const int kr = min(k0 + sj, Sk - 1);
+      VecT<T, 8> kv = vload<T, 8>(kbase + (long)kr * ks.ss + sd);
+      VecT<T, 8> vv = vload<T, 8>(vbase + (long)kr * vs.ss + sd);
+      vstore<T, 8>(&k_img[sj * KLD + sd], kv);
+      vstore<T, 8>(&v_img[sj * KLD + sd], vv);
+#pragma unroll
+      for (int u = 0; u < 8; ++u) ktT_img[(sd + u) * TLD + sj] = kv.data[u];
+      if (HAS_KPAD && threadIdx.x < 32)
+        kpad_lds[threadIdx.x] = (k0 + (int)threadIdx.x < Sk)
+                                    ? kpad_b[k0 + threadIdx.x] : 0.f;
+      if (HAS_BIAS) {
+#pragma unroll
+        for (int p = 0; p < QB / BRPP; ++p) {
+          const int qb = min(q0 + bj + p * BRPP, Sq - 1);
+          const int kbc = k0 + bc;
+          VecT<T, 8> bv;
+          if (kbc + 8 <= Sk) {
+            bv = vload<T, 8>(bias_bh + (long)qb * Sk + kbc);
+          } else {
+#pragma unroll
+            for (int u = 0; u < 8; ++u)
+              bv.data[u] = (kbc + u < Sk) ? bias_bh[(long)qb * Sk + kbc + u] : T(0.f);
+          }
+          vstore<T, 8>(&bias_img[(bj + p * BRPP) * TLD + bc], bv);
+        }
+      }
+    }
+    __syncthreads();
+
+    // S^T = K Q^T, dP^T = V dO^T (C: k rows in regs, q col = lane)
+    f32x16 st, dpt;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) { st[r] = 0.f; dpt[r] = 0.f; }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      FR ak = *reinterpret_cast<const FR*>(&k_img[lq * KLD + s * 16 + half * 8]);
+      st = M3<T>::mma(ak, qfrag[s], st);
+      FR av = *reinterpret_cast<const FR*>(&v_img[lq * KLD + s * 16 + half * 8]);
+      dpt = M3<T>::mma(av, dofrag[s], dpt);
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    // dS^T in regs (k in-lane rows), then repack -> B of dQ^T
+    float dst[16];
+    const bool q_ok = my_q < Sq;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kk = (r & 3) + 8 * (r >> 2) + 4 * half;
+      const int kc = k0 + kk;
+      float sv = st[r];
+      if (HAS_BIAS) sv += DT<T>::to_f32(bias_img[(wid * 32 + lq) * TLD + kk]);
+      if (HAS_KPAD) sv += kpad_lds[kk];
+      bool masked = kc >= Sk || !q_ok || (CAUSAL && kc > my_q + diag_off);
+      float p = (masked || my_lse == -INFINITY) ? 0.f : __expf(sv - my_lse);
+      float mfac = 1.f;
+      if (p_drop > 0.f) {
+        const long long idx = ((long long)bh * Sq + my_q) * Sk + kc;
+        mfac = (hash_u01(seed, idx) >= p_drop) ? keep_inv : 0.f;
+      }
+      dst[r] = p * (dpt[r] * mfac - my_delta);
+    }
+    unsigned pbS[8];
+    repack16<T>(dst, pbS);
+
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int dt = 0; dt < 2; ++dt) {
+#pragma unroll
+      for (int g = 0; g < 2; ++g) {
+        FR akT = *reinterpret_cast<const FR*>(
+            &ktT_img[(dt * 32 + lq) * TLD + g * 16 + half * 8]);
+        dq_acc[dt] = M3<T>::mma(akT, frag_of<T>(&pbS[g * 4]), dq_acc[dt]);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+  }
+
+  if (my_q < Sq) {
+#pragma unroll
+    for (int dt = 0; dt < 2; ++dt)
+#pragma unroll
+      for (int grp = 0; grp < 4; ++grp) {
+        const int d0 = dt * 32 + grp * 8 + half * 4;
+        VecT<T, 4> ov;
+#pragma unroll
+        for (int u = 0; u < 4; ++u) ov.data[u] = DT<T>::from_f32(dq_acc[dt][grp * 4 + u]);
+        vstore<T, 4>(dqbase + (long)my_q * dqs.ss + d0, ov);
+      }
+  }
+}
+
+template <typename T>
+void launch_bwd_v3(const torch::Tensor& dout, const torch::Tensor& q, const torch::Tensor& k,
+                   const torch::Tensor& v, const torch::Tensor& o, const torch::Tensor& lse,
+                   const c10::optional<torch::Tensor>& bias,
+                   const c10::optional<torch::Tensor>& kpad, bool causal, bool need_dbias,
+                   double p_drop, const c10::optional<torch::Tensor>& seed,
+                   torch::Tensor& dq, torch::Tensor& dk, torch::Tensor& dv,
+                   torch::Tensor& dbias32, torch::Tensor& delta, hipStream_t stream) {
+  const int B = q.size(0), H = q.size(1), Sq = q.size(2), Sk = k.size(2);
+  constexpr int D = 64;
+  S3c qs{q.stride(0), q.stride(1), q.stride(2)};
+  S3c ks{k.stride(0), k.stride(1), k.stride(2)};
+  S3c vs{v.stride(0), v.stride(1), v.stride(2)};
+  S3c dos{dout.stride(0), dout.stride(1), dout.stride(2)};
+  S3c oss{o.stride(0), o.stride(1), o.stride(2)};
+  S3c dks{dk.stride(0), dk.stride(1), dk.stride(2)};
+  S3c dvs{dv.stride(0), dv.stride(1), dv.stride(2)};
+  S3c dqs{dq.stride(0), dq.stride(1), dq.stride(2)};
+  {
+    const long rows = (long)H * Sq;
+    dim3 grid((rows + 3) / 4, B);
+    hipLaunchKernelGGL((delta_kernel_v3<T>), grid, dim3(256), 0, stream,
+                       (const T*)dout.data_ptr(), (const T*)o.data_ptr(),
+                       delta.data_ptr<float>(), H, Sq, D, dos, oss);
+    HIP_CHECK_KERNEL();
+  }
+  const bool hb = bias.has_value() && bias->defined();
+  const bool hk = kpad.has_value() && kpad->defined();
+  const T* bias_p = hb ? (const T*)bias->data_ptr() : nullptr;
+  const float* kpad_p = hk ? kpad->data_ptr<float>() : nullptr;
+  long bias_sb = 0, bias_sh = 0;
+  if (hb) {
+    bias_sb = bias->size(0) > 1 ? bias->stride(0) : 0;
+    bias_sh = bias->size(1) > 1 ? bias->stride(1) : 0;
+  }
+  const long* seed_p = (seed.has_value() && seed->defined())
+                           ? seed->data_ptr<long>() : nullptr;
+  {
+    dim3 grid((Sk + 127) / 128, B * H);
+#define LDKV(C, HB, HK, DB)                                                            \
+  hipLaunchKernelGGL((attn_bwd_dkv_kernel<T, C, HB, HK, DB>), grid, dim3(256), 0,      \
+                     stream, (const T*)dout.data_ptr(), (const T*)q.data_ptr(),        \
+                     (const T*)k.data_ptr(), (const T*)v.data_ptr(),                   \
+                     lse.data_ptr<float>(), delta.data_ptr<float>(), bias_p, kpad_p,   \
+                     (T*)dk.data_ptr(), (T*)dv.data_ptr(),                             \
+                     need_dbias ? dbias32.data_ptr<float>() : nullptr, B, H, Sq, Sk,   \
+                     qs, ks, vs, dos, dks, dvs, bias_sb, bias_sh, (float)p_drop, seed_p)
+    if (causal) {
+      if (hb && need_dbias) { if (hk) LDKV(true, true, true, true); else LDKV(true, true, false, true); }
+      else if (hb) { if (hk) LDKV(true, true, true, false); else LDKV(true, true, false, false); }
+      else { if (hk) LDKV(true, false, true, false); else LDKV(true, false, false, false); }
+    } else {
+      if (hb && need_dbias) { if (hk) LDKV(false, true, true, true); else LDKV(false, true, false, true); }
+      else if (hb) { if (hk) LDKV(false, true, true, false); else LDKV(false, true, false, false); }
+      else { if (hk) LDKV(false, false, true, false); else LDKV(false, false, false, false); }
+    }
+#undef LDKV
+    HIP_CHECK_KERNEL();
+  }
+  {
+    dim3 grid((Sq + 127) / 128, B * H);
+#define LDQ(C, HB, HK)                                                                 \
+  hipLaunchKernelGGL((attn_bwd_dq_kernel<T, C, HB, HK>), grid, dim3(256), 0, stream,   \
+                     (const T*)dout.data_ptr(), (const T*)q.data_ptr(),                \
+                     (const T*)k.data_ptr(), (const T*)v.data_ptr(),                   \
+                     lse.data_ptr<float>(), delta.data_ptr<float>(), bias_p, kpad_p,   \
+                     (T*)dq.data_ptr(), B, H, Sq, Sk, qs, ks, vs, dos, dqs,            \
+                     bias_sb, bias_sh, (float)p_drop, seed_p)
+    if (causal) {
+      if (hb && hk) LDQ(true, true, true);
+      else if (hb) LDQ(true, true, false);
+      else if (hk) LDQ(true, false, true);
+      else LDQ(true, false, false);
+    } else {
+      if (hb && hk) LDQ(false, true, true);
+      else if (hb) LDQ(false, true, false);
+      else if (hk) LDQ(false, false, true);
+      else LDQ(false, false, false);
+    }
+#undef LDQ
+    HIP_CHECK_KERNEL();
+  }
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> attn_bwd_v3(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                                       torch::Tensor v, torch::Tensor o, torch::Tensor lse,
+                                       c10::optional<torch::Tensor> bias,
+                                       c10::optional<torch::Tensor> kpad,
+                                       bool causal, bool need_dbias,
+                                       double p_drop, c10::optional<torch::Tensor> seed) {
+  TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1 &&
+              dout.stride(3) == 1 && o.stride(3) == 1 && q.size(3) == 64,
+              "attn_bwd_v3: D=64, d contiguous");
+  const int B = q.size(0), H = q.size(1), Sq = q.size(2), Sk = k.size(2);
+  auto opts = q.options();
+  auto dq = torch::empty({B, Sq, H, 64}, opts).permute({0, 2, 1, 3});
+  auto dk = torch::empty({B, Sk, H, 64}, opts).permute({0, 2, 1, 3});
+  auto dv = torch::empty({B, Sk, H, 64}, opts).permute({0, 2, 1, 3});
+  auto delta = torch::empty({B, H, Sq}, opts.dtype(torch::kFloat));
+  torch::Tensor dbias32;
+  if (need_dbias) {
+    TORCH_CHECK(bias.has_value() && bias->size(0) == 1,
+                "dbias expects batch-broadcast bias (1,H,Sq,Sk)");
+    dbias32 = torch::empty({B, H, Sq, Sk}, opts.dtype(torch::kFloat));
+  }
+  c10::optional<torch::Tensor> bias_c;
+  if (bias.has_value() && bias->defined()) bias_c = bias->to(q.scalar_type()).contiguous();
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  if (q.scalar_type() == torch::kBFloat16)
+    launch_bwd_v3<__hip_bfloat16>(dout, q, k, v, o, lse, bias_c, kpad, causal,
+                                  need_dbias, p_drop, seed, dq, dk, dv, dbias32,
+                                  delta, stream);
+  else if (q.scalar_type() == torch::kHalf)
+    launch_bwd_v3<__half>(dout, q, k, v, o, lse, bias_c, kpad, causal, need_dbias,
+                          p_drop, seed, dq, dk, dv, dbias32, delta, stream);
+  else
+    TORCH_CHECK(false, "attn_bwd_v3: bf16/f16 only");
+  std::vector<torch::Tensor> outs = {dq, dk, dv};
+  if (need_dbias)
+    outs.push_back(dbias32.sum(0, /*keepdim=*/true).to(q.scalar_type()));
+  return outs;
+}

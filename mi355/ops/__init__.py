@@ -215,11 +215,14 @@ class _Attention(torch.autograd.Function):
             if do.stride(-1) != 1:
                 do = do.contiguous()
             seed = _dropout_seed(q.device) if ctx.p_drop > 0 else None
-            # bwd v2 is numerically validated but measures SLOWER than v1
-            # (270 regs -> 1 wave/SIMD + per-wave dQ slab traffic); opt-in
-            # until the round-2 register-pressure work
-            if (q.shape[-1] == 64 and hasattr(ext, "attn_bwd_v2")
-                    and os.environ.get("MI355_ATTN_BWD_V2", "0") == "1"):
+            # three bwd generations in-tree: v3 (FA2-style two plain-store
+            # kernels on the swapped 32x32 layout), v2 (monolithic swapped
+            # layout; slower — slab traffic), v1 (16x16 recompute)
+            bwd_impl = os.environ.get("MI355_ATTN_BWD", "v3" if q.shape[-1] == 64 else "v1")
+            if bwd_impl == "v3" and q.shape[-1] == 64 and hasattr(ext, "attn_bwd_v3"):
+                res = ext.attn_bwd_v3(do, q, k, v, o, lse, bias, kpad, ctx.causal,
+                                      ctx.bias_requires_grad, ctx.p_drop, seed)
+            elif bwd_impl == "v2" and q.shape[-1] == 64 and hasattr(ext, "attn_bwd_v2"):
                 res = ext.attn_bwd_v2(do, q, k, v, o, lse, bias, kpad, ctx.causal,
                                       ctx.bias_requires_grad, ctx.p_drop, seed)
             else:
