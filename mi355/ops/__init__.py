@@ -216,9 +216,16 @@ class _Attention(torch.autograd.Function):
                 do = do.contiguous()
             seed = _dropout_seed(q.device) if ctx.p_drop > 0 else None
             # three bwd generations in-tree: v3 (FA2-style two plain-store
-            # kernels on the swapped 32x32 layout), v2 (monolithic swapped
-            # layout; slower — slab traffic), v1 (16x16 recompute)
-            bwd_impl = os.environ.get("MI355_ATTN_BWD", "v3" if q.shape[-1] == 64 else "v1")
+            # kernels on the swapped 32x32 layout — fastest once its 128-wide
+            # blocks fill the chip), v2 (monolithic swapped layout; slab
+            # traffic), v1 (16x16 recompute — finer 64-wide blocks win when
+            # the grid underfills the 256 CUs, e.g. the B=2 headline config)
+            if q.shape[-1] == 64:
+                nblk = q.shape[0] * q.shape[1] * ((max(q.shape[2], k.shape[2]) + 127) // 128)
+                default = "v3" if nblk >= 256 else "v1"
+            else:
+                default = "v1"
+            bwd_impl = os.environ.get("MI355_ATTN_BWD", default)
             if bwd_impl == "v3" and q.shape[-1] == 64 and hasattr(ext, "attn_bwd_v3"):
                 res = ext.attn_bwd_v3(do, q, k, v, o, lse, bias, kpad, ctx.causal,
                                       ctx.bias_requires_grad, ctx.p_drop, seed)
