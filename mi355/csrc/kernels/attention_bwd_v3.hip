@@ -15,6 +15,7 @@
 // slab traffic and 2+ waves/SIMD.
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
 
 #include "../common.h"
 
@@ -503,9 +504,18 @@ void launch_bwd_v3(const torch::Tensor& dout, const torch::Tensor& q, const torc
     HIP_CHECK_KERNEL();
   }
   {
+    // the dq kernel is independent of dkv — run it on a side stream so the
+    // two grids co-fill the chip (the B=2 train config is 96+96 blocks)
+    auto side = c10::hip::getStreamFromPool(false, q.get_device());
+    hipStream_t sstream = side.stream();
+    hipEvent_t ev_fork, ev_join;
+    hipEventCreateWithFlags(&ev_fork, hipEventDisableTiming);
+    hipEventCreateWithFlags(&ev_join, hipEventDisableTiming);
+    hipEventRecord(ev_fork, stream);
+    hipStreamWaitEvent(sstream, ev_fork, 0);
     dim3 grid((Sq + 127) / 128, B * H);
 #define LDQ(C, HB, HK)                                                                 \
-  hipLaunchKernelGGL((attn_bwd_dq_kernel<T, C, HB, HK>), grid, dim3(256), 0, stream,   \
+  hipLaunchKernelGGL((attn_bwd_dq_kernel<T, C, HB, HK>), grid, dim3(256), 0, sstream,  \
                      (const T*)dout.data_ptr(), (const T*)q.data_ptr(),                \
                      (const T*)k.data_ptr(), (const T*)v.data_ptr(),                   \
                      lse.data_ptr<float>(), delta.data_ptr<float>(), bias_p, kpad_p,   \
@@ -524,6 +534,10 @@ void launch_bwd_v3(const torch::Tensor& dout, const torch::Tensor& q, const torc
     }
 #undef LDQ
     HIP_CHECK_KERNEL();
+    hipEventRecord(ev_join, sstream);
+    hipStreamWaitEvent(stream, ev_join, 0);
+    hipEventDestroy(ev_fork);
+    hipEventDestroy(ev_join);
   }
 }
 
