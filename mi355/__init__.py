@@ -5,10 +5,13 @@ trainer on RCCL-over-xGMI, ASHA tuning, batch + online inference, and a
 model zoo running hand-written HIP/CDNA4 kernels."""
 from .runtime import (ObjectRef, init, is_initialized, shutdown, put, get,
                       wait, remote, ActorPool)
+from .runtime.actor import ActorDiedError
+from . import metrics
 
 __version__ = "0.1.0"
 
 from . import ops  # noqa: E402  (dispatch layer; harmless on CPU)
 
 __all__ = ["init", "is_initialized", "shutdown", "put", "get", "wait",
-           "remote", "ActorPool", "ObjectRef", "ops", "__version__"]
+           "remote", "ActorPool", "ObjectRef", "ActorDiedError", "ops",
+           "metrics", "__version__"]

@@ -20,6 +20,10 @@ from .core import ObjectRef, ObjectStore, _RemoteError, _rt
 _gpu_rr = itertools.count()
 
 
+class ActorDiedError(RuntimeError):
+    """The actor process exited (crash/OOM/os._exit) with calls in flight."""
+
+
 def _actor_main(gpu_id, cls_bytes, init_args, init_kwargs, task_q, done_q):
     if gpu_id is not None:
         os.environ["HIP_VISIBLE_DEVICES"] = str(gpu_id)
@@ -69,29 +73,73 @@ class ActorMethod:
 
 
 class ActorHandle:
-    def __init__(self, cls, init_args, init_kwargs, num_gpus=0):
+    def __init__(self, cls, init_args, init_kwargs, num_gpus=0, max_restarts=0):
         rt = _rt()
-        gpu_id = None
+        self._gpu_id = None
         if num_gpus and rt.num_gpus:
-            gpu_id = next(_gpu_rr) % rt.num_gpus
+            self._gpu_id = next(_gpu_rr) % rt.num_gpus
+        self._cls_bytes = cloudpickle.dumps(cls)
+        self._init_args = list(init_args)
+        self._init_kwargs = dict(init_kwargs)
+        self._restarts_left = max_restarts
+        self._pending: set[str] = set()
+        self._spawn()
+        rt.actors.append(self)
+        self._cls_name = getattr(cls, "__name__", "Actor")
+
+    def _spawn(self):
         ctx = mp.get_context("spawn")
         self._task_q = ctx.Queue()
         self._done_q = ctx.Queue()
         self._proc = ctx.Process(
             target=_actor_main,
-            args=(gpu_id, cloudpickle.dumps(cls), list(init_args), dict(init_kwargs),
-                  self._task_q, self._done_q),
+            args=(self._gpu_id, self._cls_bytes, self._init_args,
+                  self._init_kwargs, self._task_q, self._done_q),
             daemon=True,
         )
         self._proc.start()
         status, info = self._done_q.get()
         if status == "init_error":
             raise RuntimeError(f"actor constructor failed:\n{info}")
-        rt.actors.append(self)
-        self._cls_name = getattr(cls, "__name__", "Actor")
+
+    def _drain_done(self):
+        import queue as _q
+
+        while True:
+            try:
+                kind, ref_id = self._done_q.get_nowait()
+            except (_q.Empty, OSError, EOFError):
+                return
+            if kind == "done":
+                self._pending.discard(ref_id)
+
+    def _check_alive(self):
+        """Detect a crashed actor process; fail in-flight calls and restart it
+        with fresh state (reference semantics: `max_restarts` re-runs the
+        constructor, it does not replay lost calls)."""
+        self._drain_done()
+        if self._proc.is_alive():
+            return
+        store = _rt().store
+        err = ActorDiedError(
+            f"actor {getattr(self, '_cls_name', 'Actor')} died "
+            f"(exitcode {self._proc.exitcode})")
+        for ref_id in self._pending:
+            store.put(_RemoteError(err, str(err)), ObjectRef(ref_id))
+        self._pending.clear()
+        if self._restarts_left > 0:
+            self._restarts_left -= 1
+            from .. import metrics as _prom
+
+            _prom.ACTOR_RESTARTS.labels(actor=self._cls_name).inc()
+            self._spawn()
+        else:
+            raise err
 
     def _call(self, method, args, kwargs) -> ObjectRef:
+        self._check_alive()
         ref = ObjectRef(uuid.uuid4().hex)
+        self._pending.add(ref.id)
         self._task_q.put((ref.id, method, list(args), dict(kwargs)))
         return ref
 
@@ -113,15 +161,18 @@ class ActorHandle:
 
 
 class ActorClass:
-    def __init__(self, cls, num_gpus=0, num_cpus=1, **_):
+    def __init__(self, cls, num_gpus=0, num_cpus=1, max_restarts=0, **_):
         self._cls = cls
         self._num_gpus = num_gpus
+        self._max_restarts = max_restarts
 
     def remote(self, *args, **kwargs) -> ActorHandle:
-        return ActorHandle(self._cls, args, kwargs, num_gpus=self._num_gpus)
+        return ActorHandle(self._cls, args, kwargs, num_gpus=self._num_gpus,
+                           max_restarts=self._max_restarts)
 
     def options(self, **opts):
-        return ActorClass(self._cls, num_gpus=opts.get("num_gpus", self._num_gpus))
+        return ActorClass(self._cls, num_gpus=opts.get("num_gpus", self._num_gpus),
+                          max_restarts=opts.get("max_restarts", self._max_restarts))
 
     def __call__(self, *a, **k):
         raise TypeError("actor classes must be instantiated with .remote()")

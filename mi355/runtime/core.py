@@ -336,6 +336,10 @@ class Runtime:
         for r in lst:
             v = self.store.get(r, timeout=timeout)
             if isinstance(v, _RemoteError):
+                from .actor import ActorDiedError
+
+                if isinstance(v.exc, ActorDiedError):
+                    raise v.exc
                 raise RuntimeError(f"remote task failed:\n{v.tb}") from v.exc
             out.append(v)
         return out[0] if single else out

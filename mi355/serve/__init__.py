@@ -10,8 +10,9 @@ HTTP proxy on :8000 routes by prefix to a round-robin pool of predictor
 replicas. Replicas live in-process (model loaded once each); with the mi355
 runtime initialised and num_gpus>0 they are GPU-pinned actor processes.
 """
-from .deployment import (Deployment, PredictorDeployment, pandas_read_json,
-                         json_request, run, shutdown, get_deployment)
+from .deployment import (Deployment, PredictorDeployment, deployment,
+                         pandas_read_json, json_request, run, shutdown,
+                         get_deployment)
 
-__all__ = ["Deployment", "PredictorDeployment", "pandas_read_json",
+__all__ = ["Deployment", "PredictorDeployment", "deployment", "pandas_read_json",
            "json_request", "run", "shutdown", "get_deployment"]

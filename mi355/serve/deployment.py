@@ -4,10 +4,13 @@ from __future__ import annotations
 import itertools
 import json
 import threading
+import time
 from dataclasses import dataclass, field, replace
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
 
 import pandas as pd
+
+from .. import metrics
 
 
 # -- HTTP adapters (reference: ray.serve.http_adapters.pandas_read_json) ---- #
@@ -64,11 +67,23 @@ class _PredictorReplicaGroup:
 
     def __init__(self, app: Application):
         d = app.deployment
-        predictor_cls, checkpoint = app.args[0], app.args[1]
         self.http_adapter = app.kwargs.get("http_adapter", pandas_read_json)
-        pred_kwargs = {k: v for k, v in app.kwargs.items() if k != "http_adapter"}
-        self.replicas = [predictor_cls.from_checkpoint(checkpoint, **pred_kwargs)
-                         for _ in range(d.num_replicas)]
+        rest = {k: v for k, v in app.kwargs.items() if k != "http_adapter"}
+        if app.args and hasattr(app.args[0], "from_checkpoint"):
+            # PredictorDeployment.bind(PredictorCls, checkpoint, ...)
+            predictor_cls, checkpoint = app.args[0], app.args[1]
+            self.replicas = [predictor_cls.from_checkpoint(checkpoint, **rest)
+                             for _ in range(d.num_replicas)]
+            self._invoke = lambda rep, data, kw: rep.predict(data, **kw)
+        elif isinstance(d.func_or_class, type):
+            # generic @serve.deployment class: instance is the handler
+            self.replicas = [d.func_or_class(*app.args, **rest)
+                             for _ in range(d.num_replicas)]
+            self._invoke = lambda rep, data, kw: rep(data, **kw)
+        else:
+            # @serve.deployment function
+            self.replicas = [d.func_or_class] * max(d.num_replicas, 1)
+            self._invoke = lambda rep, data, kw: rep(data, **kw)
         self._rr = itertools.cycle(range(len(self.replicas)))
         self._locks = [threading.Lock() for _ in self.replicas]
 
@@ -76,7 +91,7 @@ class _PredictorReplicaGroup:
         data = self.http_adapter(body)
         i = next(self._rr)
         with self._locks[i]:
-            out = self.replicas[i].predict(data, **query_kwargs)
+            out = self._invoke(self.replicas[i], data, query_kwargs)
         if isinstance(out, pd.DataFrame):
             return out.to_dict(orient="records")
         import numpy as np
@@ -126,6 +141,7 @@ class _Handler(BaseHTTPRequestHandler):
             self.end_headers()
             self.wfile.write(b'{"error": "no such route"}')
             return
+        t0 = time.perf_counter()
         try:
             result = group.handle(body, {})
             payload = json.dumps(result).encode()
@@ -137,6 +153,10 @@ class _Handler(BaseHTTPRequestHandler):
             self.send_response(500)
             self.end_headers()
             self.wfile.write(json.dumps({"error": str(e)}).encode())
+        finally:
+            metrics.SERVE_REQUESTS.labels(route=route).inc()
+            metrics.SERVE_REQUEST_SECONDS.labels(route=route).inc(
+                time.perf_counter() - t0)
 
     def do_GET(self):
         if self.path in ("/", "/-/routes"):
@@ -144,6 +164,13 @@ class _Handler(BaseHTTPRequestHandler):
             self.end_headers()
             self.wfile.write(json.dumps(
                 {r: d.deployment.name for r, d in _state.apps.items()}).encode())
+        elif self.path.split("?")[0] in ("/-/metrics", "/metrics"):
+            body = metrics.REGISTRY.render().encode()
+            self.send_response(200)
+            self.send_header("Content-Type", metrics.CONTENT_TYPE)
+            self.send_header("Content-Length", str(len(body)))
+            self.end_headers()
+            self.wfile.write(body)
         else:
             self.do_POST()
 

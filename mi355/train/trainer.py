@@ -232,10 +232,17 @@ class DataParallelTrainer:
                     last_metrics = payload
                     metrics_rows.append(payload)
                     # observability: JSONL metrics per run dir
+                    # + Prometheus gauges (mi355.metrics)
                     import json as _json
 
                     with open(os.path.join(run_dir, "metrics.jsonl"), "a") as mf:
                         mf.write(_json.dumps(payload, default=str) + "\n")
+                    from .. import metrics as _prom
+
+                    _prom.record_train_report(self.run_config.name or "train", payload)
+                    _prom.TRAIN_WORKERS.labels(
+                        run=self.run_config.name or "train").set(
+                        sum(p.is_alive() for p in procs))
                     if ckpt_dir:
                         n_ckpt += 1
                         kept = os.path.join(run_dir, f"checkpoint_{n_ckpt:06d}")

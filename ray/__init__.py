@@ -6,7 +6,7 @@ layer only; the implementation is mi355's MI355X-native stack."""
 from mi355 import (ActorPool, ObjectRef, get, init, is_initialized, put,
                    remote, shutdown, wait)
 from mi355 import serve, tune  # noqa: F401
-from . import air, data, train, util  # noqa: F401
+from . import air, data, exceptions, train, util  # noqa: F401
 
 __version__ = "2.3.1+mi355"
 
