@@ -41,8 +41,28 @@ def busy_union(rows):
 
 def analyze(db_path, out_path=None, marker="adamw", n_gaps=20):
     rows = load(db_path)
-    marks = [i for i, (_, _, n) in enumerate(rows) if marker in n]
-    lines = [f"dispatches={len(rows)} step_markers={len(marks)}"]
+    raw = [i for i, (_, _, n) in enumerate(rows) if marker in n]
+    # a step can fire the marker kernel more than once back-to-back (e.g. two
+    # fused optimizer launches) — keep only the last of each cluster (<1ms)
+    marks = [i for j, i in enumerate(raw)
+             if j == len(raw) - 1 or rows[raw[j + 1]][0] - rows[i][1] > 1e6]
+    lines = [f"dispatches={len(rows)} step_markers={len(marks)} (raw {len(raw)})"]
+    lines.append("per-interval (marker->marker = one step incl. host work):")
+    for j in range(max(0, len(marks) - 14), len(marks) - 1):
+        win = rows[marks[j]:marks[j + 1]]
+        span = win[-1][1] - win[0][0]
+        busy = busy_union(win)
+        gaps = []
+        prev_e = win[0][1]
+        for s, e, n in win[1:]:
+            if s > prev_e:
+                gaps.append(s - prev_e)
+            prev_e = max(prev_e, e)
+        big = sorted(gaps, reverse=True)[:3]
+        lines.append(
+            f"  step[{j}]: span={span/1e6:7.3f}ms busy={busy/1e6:7.3f}ms "
+            f"idle={100*(span-busy)/span:4.1f}% ngaps={len(gaps)} "
+            f"top_gaps_us={[round(g/1e3,1) for g in big]}")
     if len(marks) >= 3:
         # steady state: between the 2nd and last marker
         lo, hi = marks[1], marks[-1]
