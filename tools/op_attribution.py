@@ -39,9 +39,30 @@ def main():
         f.write("== by op, self CUDA ==\n")
         f.write(prof.key_averages().table(
             sort_by="self_cuda_time_total", row_limit=40, max_name_column_width=70))
-        f.write("\n\n== by source line (group_by_stack_n=7) ==\n")
-        f.write(prof.key_averages(group_by_stack_n=7).table(
-            sort_by="self_cuda_time_total", row_limit=50, max_name_column_width=60))
+    stacks = out.replace(".txt", "_stacks.txt")
+    prof.export_stacks(stacks, "self_cuda_time_total")
+    # condense: keep only soup ops, aggregate by deepest mi355/bench frame
+    agg = {}
+    with open(stacks) as f:
+        for line in f:
+            parts = line.rsplit(" ", 1)
+            if len(parts) != 2:
+                continue
+            frames, val = parts[0].split(";"), int(parts[1])
+            op = frames[-1]
+            if not any(k in op for k in ("copy_", "add_", "fill_", "cat",
+                                         "sum", "pow", "to ", "aten::mul",
+                                         "aten::div", "contiguous")):
+                continue
+            src = next((fr for fr in reversed(frames)
+                        if ".py(" in fr and ("mi355" in fr or "bench" in fr)),
+                       "?")
+            key = (op.strip(), src.strip())
+            agg[key] = agg.get(key, 0) + val
+    with open(out, "a") as f:
+        f.write("\n\n== soup ops by source (us) ==\n")
+        for (op, src), v in sorted(agg.items(), key=lambda x: -x[1])[:40]:
+            f.write(f"{v/3:10.0f}us/step  {op:<24} {src}\n")
     print("wrote", out)
 
 
