@@ -12,10 +12,15 @@ std::vector<torch::Tensor> add_rmsnorm_fwd(torch::Tensor x, torch::Tensor r,
                                            double p_drop, c10::optional<torch::Tensor> seed,
                                            long salt);
 torch::Tensor geglu_fwd(torch::Tensor gate, torch::Tensor up);
+torch::Tensor geglu_packed_fwd(torch::Tensor x, double p_drop,
+                               c10::optional<torch::Tensor> seed, long salt);
+torch::Tensor geglu_packed_bwd(torch::Tensor dout, torch::Tensor x, double p_drop,
+                               c10::optional<torch::Tensor> seed, long salt);
 std::vector<torch::Tensor> geglu_bwd(torch::Tensor dout, torch::Tensor gate, torch::Tensor up);
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels, long ignore_index);
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels, torch::Tensor lse,
                      long ignore_index, torch::Tensor dloss, torch::Tensor nvalid);
+void sumsq_into(torch::Tensor x, torch::Tensor out);
 void adamw_multi(std::vector<torch::Tensor> masters, std::vector<torch::Tensor> grads,
                  std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
                  std::vector<torch::Tensor> lowps, double lr, double beta1, double beta2,
@@ -61,10 +66,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("add_rmsnorm_fwd", &add_rmsnorm_fwd);
   m.def("geglu_fwd", &geglu_fwd);
+  m.def("geglu_packed_fwd", &geglu_packed_fwd);
+  m.def("geglu_packed_bwd", &geglu_packed_bwd);
   m.def("geglu_bwd", &geglu_bwd);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
   m.def("adamw_multi", &adamw_multi);
+  m.def("sumsq_into", &sumsq_into);
   m.def("decode_attn", &decode_attn);
   m.def("attn_fwd", &attn_fwd);
   m.def("token_embed_fwd", &token_embed_fwd);

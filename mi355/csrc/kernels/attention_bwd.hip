@@ -418,7 +418,107 @@ void launch_bwd(const torch::Tensor& dout, const torch::Tensor& q, const torch::
   HIP_CHECK_KERNEL();
 }
 
+// epilogue kernels (replace the aten sum/copy_/cast chains — census showed
+// them costing ~1 ms/step at B=2): one fused pass each.
+
+// dq = sum over the nkt per-k-tile fp32 slabs, written as T through the
+// (b,h,s) strides of the BSHD-physical dq view. VEC floats along d per thread.
+template <typename T, int VEC>
+__global__ void slab_reduce_kernel(const float* __restrict__ slab,
+                                   T* __restrict__ dq, int nkt, long slab_elems,
+                                   int H, int Sq, int D,
+                                   long s0, long s1, long s2) {
+  long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * VEC;
+  if (i >= slab_elems) return;
+  float acc[VEC];
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) acc[j] = 0.f;
+  for (int t = 0; t < nkt; ++t) {
+    VecT<float, VEC> v = vload<float, VEC>(slab + (long)t * slab_elems + i);
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) acc[j] += v.data[j];
+  }
+  int d = (int)(i % D);
+  long r = i / D;
+  int sq = (int)(r % Sq);
+  r /= Sq;
+  int h = (int)(r % H);
+  long b = r / H;
+  T* out = dq + b * s0 + h * s1 + (long)sq * s2 + d;
+  VecT<T, VEC> ov;
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) ov.data[j] = DT<T>::from_f32(acc[j]);
+  vstore<T, VEC>(out, ov);
+}
+
+// dbias = sum over batch of the fp32 per-batch planes + cast, one pass
+template <typename T, int VEC>
+__global__ void bsum_cast_kernel(const float* __restrict__ in, T* __restrict__ out,
+                                 int B, long plane) {
+  long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * VEC;
+  if (i >= plane) return;
+  float acc[VEC];
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) acc[j] = 0.f;
+  for (int b = 0; b < B; ++b) {
+    VecT<float, VEC> v = vload<float, VEC>(in + (long)b * plane + i);
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) acc[j] += v.data[j];
+  }
+  VecT<T, VEC> ov;
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) ov.data[j] = DT<T>::from_f32(acc[j]);
+  vstore<T, VEC>(out + i, ov);
+}
+
+template <typename T>
+void launch_epilogue(const torch::Tensor& dq32, torch::Tensor& dq,
+                     const torch::Tensor* dbias32, torch::Tensor* dbias,
+                     hipStream_t stream) {
+  const int nkt = dq32.size(0);
+  const long slab = dq32.numel() / nkt;
+  const int H = dq.size(1), Sq = dq.size(2), D = dq.size(3);
+  constexpr int VEC = 4;
+  long blocks = (slab / VEC + 255) / 256;
+  hipLaunchKernelGGL((slab_reduce_kernel<T, VEC>), dim3(blocks), dim3(256), 0,
+                     stream, dq32.data_ptr<float>(), (T*)dq.data_ptr(), nkt,
+                     slab, H, Sq, D, dq.stride(0), dq.stride(1), dq.stride(2));
+  HIP_CHECK_KERNEL();
+  if (dbias32 != nullptr) {
+    const int B = dbias32->size(0);
+    const long plane = dbias32->numel() / B;
+    long bblocks = (plane / VEC + 255) / 256;
+    hipLaunchKernelGGL((bsum_cast_kernel<T, VEC>), dim3(bblocks), dim3(256), 0,
+                       stream, dbias32->data_ptr<float>(), (T*)dbias->data_ptr(),
+                       B, plane);
+    HIP_CHECK_KERNEL();
+  }
+}
+
 }  // namespace
+
+// shared with attention_bwd_v3.hip: fused batch-sum + cast of the fp32
+// per-batch dbias planes
+void attn_dbias_reduce(const torch::Tensor& dbias32, torch::Tensor& dbias,
+                       hipStream_t stream) {
+  const int B = dbias32.size(0);
+  const long plane = dbias32.numel() / B;
+  if (plane % 4 != 0) {  // ragged: fall back to aten
+    dbias.copy_(dbias32.sum(0, /*keepdim=*/true));
+    return;
+  }
+  long blocks = (plane / 4 + 255) / 256;
+  if (dbias.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL((bsum_cast_kernel<__hip_bfloat16, 4>), dim3(blocks),
+                       dim3(256), 0, stream, dbias32.data_ptr<float>(),
+                       (__hip_bfloat16*)dbias.data_ptr(), B, plane);
+  } else {
+    hipLaunchKernelGGL((bsum_cast_kernel<__half, 4>), dim3(blocks), dim3(256),
+                       0, stream, dbias32.data_ptr<float>(),
+                       (__half*)dbias.data_ptr(), B, plane);
+  }
+  HIP_CHECK_KERNEL();
+}
 
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor o, torch::Tensor lse,
@@ -460,11 +560,25 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
   else
     TORCH_CHECK(false, "attn_bwd: unsupported dtype/D=", D);
 
-  // dq: sum the per-k-tile fp32 slabs -> low-precision BSHD-physical view
+  // dq: sum the per-k-tile fp32 slabs -> low-precision BSHD-physical view;
+  // dbias: batch-sum + cast. Single fused pass each (no aten sum/copy/cast).
   auto dq = torch::empty({B, Sq, H, D}, opts).permute({0, 2, 1, 3});
-  dq.copy_(dq32.sum(0));
+  const long plane = need_dbias ? dbias32.numel() / B : 0;
+  const bool fused_ok = ((long)B * H * Sq * D) % 4 == 0 && plane % 4 == 0;
+  torch::Tensor dbias;
+  if (need_dbias) dbias = torch::empty({1, H, Sq, Sk}, opts);
+  if (fused_ok) {
+    if (q.scalar_type() == torch::kBFloat16)
+      launch_epilogue<__hip_bfloat16>(dq32, dq, need_dbias ? &dbias32 : nullptr,
+                                      need_dbias ? &dbias : nullptr, stream);
+    else
+      launch_epilogue<__half>(dq32, dq, need_dbias ? &dbias32 : nullptr,
+                              need_dbias ? &dbias : nullptr, stream);
+  } else {
+    dq.copy_(dq32.sum(0));
+    if (need_dbias) dbias.copy_(dbias32.sum(0, /*keepdim=*/true));
+  }
   std::vector<torch::Tensor> outs = {dq, dk, dv};
-  if (need_dbias)
-    outs.push_back(dbias32.sum(0, /*keepdim=*/true).to(q.scalar_type()));
+  if (need_dbias) outs.push_back(dbias);
   return outs;
 }

@@ -543,6 +543,10 @@ void launch_bwd_v3(const torch::Tensor& dout, const torch::Tensor& q, const torc
 
 }  // namespace
 
+// defined in attention_bwd.hip: fused batch-sum + cast of fp32 dbias planes
+void attn_dbias_reduce(const torch::Tensor& dbias32, torch::Tensor& dbias,
+                       hipStream_t stream);
+
 std::vector<torch::Tensor> attn_bwd_v3(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                                        torch::Tensor v, torch::Tensor o, torch::Tensor lse,
                                        c10::optional<torch::Tensor> bias,
@@ -577,7 +581,10 @@ std::vector<torch::Tensor> attn_bwd_v3(torch::Tensor dout, torch::Tensor q, torc
   else
     TORCH_CHECK(false, "attn_bwd_v3: bf16/f16 only");
   std::vector<torch::Tensor> outs = {dq, dk, dv};
-  if (need_dbias)
-    outs.push_back(dbias32.sum(0, /*keepdim=*/true).to(q.scalar_type()));
+  if (need_dbias) {
+    auto dbias = torch::empty({1, H, Sq, Sk}, opts);
+    attn_dbias_reduce(dbias32, dbias, stream);
+    outs.push_back(dbias);
+  }
   return outs;
 }

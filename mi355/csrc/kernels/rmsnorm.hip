@@ -176,17 +176,16 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy, const T* __restrict
   }
 }
 
-// 2D: x over columns, y over row chunks; one atomic per (col, row-chunk)
-__global__ void colsum_kernel(const float* __restrict__ part, float* __restrict__ out,
+// column sum of the <=256 per-block partials, emitted directly in the weight
+// dtype (no separate fp32->bf16 cast launch, no zero-fill of dw)
+template <typename OT>
+__global__ void colsum_kernel(const float* __restrict__ part, OT* __restrict__ out,
                               int nparts, int D) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= D) return;
-  const int rpb = (nparts + gridDim.y - 1) / gridDim.y;
-  const int r0 = blockIdx.y * rpb, r1 = min(nparts, r0 + rpb);
   float s = 0.f;
-  for (int r = r0; r < r1; ++r) s += part[(long)r * D + c];
-  if (gridDim.y == 1) out[c] = s;
-  else atomicAdd(&out[c], s);
+  for (int r = 0; r < nparts; ++r) s += part[(long)r * D + c];
+  out[c] = DT<OT>::from_f32(s);
 }
 
 template <typename T>
@@ -223,8 +222,8 @@ void launch_bwd(const torch::Tensor& dy, const torch::Tensor& x, const torch::Te
                      ddelta ? (T*)ddelta->data_ptr() : nullptr, p_drop,
                      seed ? seed->data_ptr<long>() : nullptr, salt);
   HIP_CHECK_KERNEL();
-  hipLaunchKernelGGL(colsum_kernel, dim3((D + 255) / 256, 16), dim3(256), 0, stream,
-                     part.data_ptr<float>(), dw.data_ptr<float>(), grid, D);
+  hipLaunchKernelGGL((colsum_kernel<T>), dim3((D + 255) / 256), dim3(256), 0, stream,
+                     part.data_ptr<float>(), (T*)dw.data_ptr(), grid, D);
   HIP_CHECK_KERNEL();
 }
 
@@ -252,7 +251,8 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch:
                                        double p_drop, c10::optional<torch::Tensor> seed,
                                        long salt) {
   auto dx = torch::empty_like(x);
-  auto dw = torch::zeros({x.size(1)}, x.options().dtype(torch::kFloat));
+  TORCH_CHECK(w.scalar_type() == x.scalar_type(), "rmsnorm bwd: w/x dtype mismatch");
+  auto dw = torch::empty({x.size(1)}, w.options());
   auto stream = at::hip::getCurrentHIPStream().stream();
   const torch::Tensor* de = (dextra.has_value() && dextra->defined()) ? &*dextra : nullptr;
   const torch::Tensor* sp = (seed.has_value() && seed->defined()) ? &*seed : nullptr;

@@ -233,9 +233,12 @@ class T5DenseGatedActDense(nn.Module):
         self.dropout = nn.Dropout(config.dropout_rate)
 
     def forward(self, x):
-        gate, up = self.wi(x)
-        h = ops.geglu(gate, up)
-        return self.wo(self.dropout(h))
+        # packed: one fused-wi GEMM -> one GeGLU kernel with the FF dropout
+        # fused on its output (no gate/up split views, no grad cat)
+        p = self.dropout.p if self.training else 0.0
+        h = ops.geglu_packed(F.linear(x, self.wi.weight), p_drop=p,
+                             salt=self.wi.weight.data_ptr())
+        return self.wo(h)
 
 
 class T5LayerFF(nn.Module):

@@ -39,6 +39,13 @@ def hip_available() -> bool:
     return torch.cuda.is_available() and _load_ext() is not None
 
 
+def sumsq_accumulate(x: torch.Tensor, out: torch.Tensor) -> None:
+    """out += sum(x*x) — one fused read of a flat (bf16/fp16/fp32) tensor into
+    a 0-dim fp32 device accumulator (grad-clip total-norm without the
+    .float().pow(2).sum() round trip)."""
+    _load_ext().sumsq_into(x, out)
+
+
 def _use_hip(*tensors: torch.Tensor) -> bool:
     if os.environ.get("MI355_FORCE_REF") == "1":
         return False
@@ -152,6 +159,42 @@ def geglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
     return torch_ref.geglu(gate, up)
 
 
+class _GeGLUPacked(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, p_drop, salt):
+        x = x.contiguous()
+        seed = _dropout_seed(x.device) if p_drop > 0 else None
+        out = _load_ext().geglu_packed_fwd(x, p_drop, seed, salt)
+        ctx.save_for_backward(x)
+        ctx.p_drop, ctx.salt = p_drop, salt
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        (x,) = ctx.saved_tensors
+        seed = _dropout_seed(x.device) if ctx.p_drop > 0 else None
+        dx = _load_ext().geglu_packed_bwd(dout.contiguous(), x, ctx.p_drop,
+                                          seed, ctx.salt)
+        return dx, None, None
+
+
+def geglu_packed(x, p_drop: float = 0.0, salt: int = 0):
+    """GeGLU over a packed (..., 2F) [gate | up] projection output — one
+    kernel straight off the fused wi GEMM (no split views to clone, a single
+    packed grad so autograd never cats). The FF residual dropout is fused on
+    the output (hash mask, graph-safe device seed; same stream as the
+    attention-prob dropout, salted per layer)."""
+    if _use_hip(x):
+        return _GeGLUPacked.apply(x, float(p_drop), salt)
+    gate, up = x.chunk(2, dim=-1)
+    h = torch_ref.geglu(gate, up)
+    if p_drop > 0.0:
+        import torch.nn.functional as F
+
+        h = F.dropout(h, p=p_drop, training=True)
+    return h
+
+
 # --------------------------------------------------------------------------- #
 # Fused attention with additive bias (T5 relative-position bias / masks)      #
 # --------------------------------------------------------------------------- #
@@ -204,7 +247,11 @@ class _Attention(torch.autograd.Function):
     @staticmethod
     def backward(ctx, do):
         q, k, v, bias, kpad, o, lse = ctx.saved_tensors
-        do = do.contiguous()
+        # do usually arrives as a (B,H,S,D)-permuted view of a BSHD-contiguous
+        # grad buffer — the kernels read it through (b,h,s) strides, so only a
+        # non-unit head-dim stride forces a copy
+        if do.stride(-1) != 1:
+            do = do.contiguous()
         ext = _load_ext()
         use_hip_bwd = hasattr(ext, "attn_bwd") and q.shape[-1] in (32, 64) and \
             (not ctx.bias_requires_grad or (bias is not None and bias.shape[0] == 1)) \

@@ -166,6 +166,15 @@ class FlatReplica:
         self._flushed.clear()
 
     def grad_norm(self) -> torch.Tensor:
+        from .. import ops
+
+        if self.groups and self.groups[0][1].flat_grad.is_cuda \
+                and ops.hip_available():
+            dev = self.groups[0][1].flat_grad.device
+            out = torch.zeros((), dtype=torch.float32, device=dev)
+            for _, g in self.groups:
+                ops.sumsq_accumulate(g.flat_grad, out)
+            return out.sqrt()
         sq = None
         for _, g in self.groups:
             s = g.flat_grad.float().pow(2).sum()

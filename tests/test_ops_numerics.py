@@ -305,3 +305,50 @@ def test_native_ext_is_loaded():
     assert ops.hip_available()
     import mi355._C as C
     assert C.__file__.endswith("_C.so") and "/mi355/" in C.__file__
+
+
+@gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_geglu_packed_fwd_bwd(dtype):
+    torch.manual_seed(3)
+    tol = 1e-5 if dtype == torch.float32 else 2e-2
+    x = torch.randn(512, 4096, device="cuda", dtype=dtype, requires_grad=True)
+    out = ops.geglu_packed(x)
+    xr = x.detach().float().requires_grad_(True)
+    outr = torch_ref.geglu(*xr.chunk(2, dim=-1))
+    assert _relerr(out, outr) < tol
+    d = torch.randn_like(outr)
+    out.backward(d.to(dtype))
+    outr.backward(d)
+    assert _relerr(x.grad, xr.grad) < tol
+
+
+@gpu
+def test_geglu_packed_dropout_deterministic():
+    # same device seed + salt -> identical mask fwd/fwd and matching bwd
+    # masking (zero positions in out have zero dgate/dup grads)
+    torch.manual_seed(4)
+    x = torch.randn(256, 1024, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    out1 = ops.geglu_packed(x, p_drop=0.5, salt=123)
+    out2 = ops.geglu_packed(x, p_drop=0.5, salt=123)
+    assert torch.equal(out1, out2)
+    out3 = ops.geglu_packed(x, p_drop=0.5, salt=124)
+    assert not torch.equal(out1, out3)
+    drop_frac = (out1 == 0).float().mean().item()
+    assert 0.4 < drop_frac < 0.62  # ~p plus natural zeros
+    out1.backward(torch.ones_like(out1))
+    g = x.grad[:, 512:]  # dup = d * gelu(gate): zero exactly where dropped
+    assert ((out1 == 0) == (g == 0)).float().mean().item() > 0.95
+
+
+@gpu
+def test_sumsq_accumulate():
+    torch.manual_seed(5)
+    out = torch.zeros((), dtype=torch.float32, device="cuda")
+    a = torch.randn(1_000_003, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(4096, device="cuda", dtype=torch.float32)
+    ops.sumsq_accumulate(a, out)
+    ops.sumsq_accumulate(b, out)
+    ref = a.float().pow(2).sum() + b.pow(2).sum()
+    assert _relerr(out, ref) < 1e-3
