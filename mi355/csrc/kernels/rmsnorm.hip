@@ -213,7 +213,7 @@ void launch_bwd(const torch::Tensor& dy, const torch::Tensor& x, const torch::Te
   const int threads = (D / VEC >= 256) ? 256 : 128;
   TORCH_CHECK(D <= threads * VEC * 4, "rmsnorm bwd: D too large for dw accumulator");
   const int grid = std::min(rows, 256);
-  auto part = torch::empty({grid, D}, dw.options());
+  auto part = torch::empty({grid, D}, dw.options().dtype(torch::kFloat));
   hipLaunchKernelGGL((rmsnorm_bwd_kernel<T, VEC, 4>), dim3(grid), dim3(threads), 0, stream,
                      (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (const T*)w.data_ptr(),
                      inv.data_ptr<float>(),
