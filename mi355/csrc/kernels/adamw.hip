@@ -10,26 +10,52 @@
 
 namespace {
 
-template <typename G, typename P>
+template <typename G, typename P, int VEC>
 __global__ void adamw_kernel(float* __restrict__ master, const G* __restrict__ grad,
                              float* __restrict__ m, float* __restrict__ v,
                              P* __restrict__ lowp, long n, float lr, float beta1,
                              float beta2, float eps, float wd, float bc1, float bc2,
                              float grad_scale, const float* __restrict__ grad_scale_dev) {
   const float gs = grad_scale_dev ? fminf(*grad_scale_dev, 1.0f) : grad_scale;
-  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long stride = (long)gridDim.x * blockDim.x;
-  for (; i < n; i += stride) {
-    float g = DT<G>::to_f32(grad[i]) * gs;
-    float mi = beta1 * m[i] + (1.f - beta1) * g;
-    float vi = beta2 * v[i] + (1.f - beta2) * g * g;
-    m[i] = mi;
-    v[i] = vi;
+  long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * VEC;
+  const long stride = (long)gridDim.x * blockDim.x * VEC;
+  for (; i + VEC <= n; i += stride) {
+    VecT<G, VEC> gv = vload<G, VEC>(grad + i);
+    VecT<float, VEC> mv = vload<float, VEC>(m + i);
+    VecT<float, VEC> vv = vload<float, VEC>(v + i);
+    VecT<float, VEC> pv = vload<float, VEC>(master + i);
+    VecT<P, VEC> lv;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float g = DT<G>::to_f32(gv.data[j]) * gs;
+      float mi = beta1 * mv.data[j] + (1.f - beta1) * g;
+      float vi = beta2 * vv.data[j] + (1.f - beta2) * g * g;
+      mv.data[j] = mi;
+      vv.data[j] = vi;
+      float denom = sqrtf(vi / bc2) + eps;
+      float p = pv.data[j] * (1.f - lr * wd);
+      p -= lr * (mi / bc1) / denom;
+      pv.data[j] = p;
+      lv.data[j] = DT<P>::from_f32(p);
+    }
+    vstore<float, VEC>(m + i, mv);
+    vstore<float, VEC>(v + i, vv);
+    vstore<float, VEC>(master + i, pv);
+    if (lowp) vstore<P, VEC>(lowp + i, lv);
+  }
+  // ragged tail: scalar path on block 0
+  if (blockIdx.x == 0 && threadIdx.x < (n % VEC)) {
+    long t = n - (n % VEC) + threadIdx.x;
+    float g = DT<G>::to_f32(grad[t]) * gs;
+    float mi = beta1 * m[t] + (1.f - beta1) * g;
+    float vi = beta2 * v[t] + (1.f - beta2) * g * g;
+    m[t] = mi;
+    v[t] = vi;
     float denom = sqrtf(vi / bc2) + eps;
-    float p = master[i] * (1.f - lr * wd);
+    float p = master[t] * (1.f - lr * wd);
     p -= lr * (mi / bc1) / denom;
-    master[i] = p;
-    if (lowp) lowp[i] = DT<P>::from_f32(p);
+    master[t] = p;
+    if (lowp) lowp[t] = DT<P>::from_f32(p);
   }
 }
 
@@ -41,8 +67,8 @@ void launch(torch::Tensor& master, const torch::Tensor& grad, torch::Tensor& m,
   long n = master.numel();
   float bc1 = 1.f - powf(b1, (float)step);
   float bc2 = 1.f - powf(b2, (float)step);
-  long blocks = std::min<long>((n + 255) / 256, 4096);
-  hipLaunchKernelGGL((adamw_kernel<G, P>), dim3(blocks), dim3(256), 0, stream,
+  long blocks = std::min<long>((n / 4 + 255) / 256, 8192);
+  hipLaunchKernelGGL((adamw_kernel<G, P, 4>), dim3(blocks), dim3(256), 0, stream,
                      master.data_ptr<float>(), (const G*)grad.data_ptr(),
                      m.data_ptr<float>(), v.data_ptr<float>(),
                      lowp ? (P*)lowp->data_ptr() : nullptr, n, lr, b1, b2, eps, wd,

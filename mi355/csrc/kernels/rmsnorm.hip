@@ -224,7 +224,7 @@ void launch_bwd(const torch::Tensor& dy, const torch::Tensor& x, const torch::Te
   constexpr int VEC = sizeof(T) == 4 ? 4 : 8;
   const int threads = (D / VEC >= 256) ? 256 : 128;
   TORCH_CHECK(D <= threads * VEC * 4, "rmsnorm bwd: D too large for dw accumulator");
-  const int grid = std::min(rows, 256);
+  const int grid = std::min(rows, 1024);
   auto part = torch::empty({grid, D}, dw.options().dtype(torch::kFloat));
   hipLaunchKernelGGL((rmsnorm_bwd_kernel<T, VEC, 4>), dim3(grid), dim3(threads), 0, stream,
                      (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (const T*)w.data_ptr(),
