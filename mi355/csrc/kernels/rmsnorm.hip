@@ -180,22 +180,23 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy, const T* __restrict
 // dtype (no separate fp32->bf16 cast launch, no zero-fill of dw).
 // block = 16 row-lanes x 16 columns so the row loop is 16-way parallel —
 // a 1-thread-per-column version was latency-bound at 59 us/call.
-template <typename OT>
+template <typename OT, int RLANES>
 __global__ void colsum_kernel(const float* __restrict__ part, OT* __restrict__ out,
                               int nparts, int D) {
-  __shared__ float red[16][17];
-  const int cl = threadIdx.x & 15;
-  const int rl = threadIdx.x >> 4;
-  const int c = blockIdx.x * 16 + cl;
+  constexpr int CLANES = 256 / RLANES;
+  __shared__ float red[RLANES][CLANES + 1];
+  const int cl = threadIdx.x % CLANES;
+  const int rl = threadIdx.x / CLANES;
+  const int c = blockIdx.x * CLANES + cl;
   float s = 0.f;
   if (c < D)
-    for (int r = rl; r < nparts; r += 16) s += part[(long)r * D + c];
+    for (int r = rl; r < nparts; r += RLANES) s += part[(long)r * D + c];
   red[rl][cl] = s;
   __syncthreads();
   if (rl == 0 && c < D) {
     float t = 0.f;
 #pragma unroll
-    for (int i = 0; i < 16; ++i) t += red[i][cl];
+    for (int i = 0; i < RLANES; ++i) t += red[i][cl];
     out[c] = DT<OT>::from_f32(t);
   }
 }
@@ -224,7 +225,9 @@ void launch_bwd(const torch::Tensor& dy, const torch::Tensor& x, const torch::Te
   constexpr int VEC = sizeof(T) == 4 ? 4 : 8;
   const int threads = (D / VEC >= 256) ? 256 : 128;
   TORCH_CHECK(D <= threads * VEC * 4, "rmsnorm bwd: D too large for dw accumulator");
-  const int grid = std::min(rows, 1024);
+  // big-batch rows want more blocks; small batches keep the partial slab
+  // (and its colsum) tiny
+  const int grid = rows <= 2048 ? std::min(rows, 256) : 1024;
   auto part = torch::empty({grid, D}, dw.options().dtype(torch::kFloat));
   hipLaunchKernelGGL((rmsnorm_bwd_kernel<T, VEC, 4>), dim3(grid), dim3(threads), 0, stream,
                      (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (const T*)w.data_ptr(),
@@ -234,8 +237,12 @@ void launch_bwd(const torch::Tensor& dy, const torch::Tensor& x, const torch::Te
                      ddelta ? (T*)ddelta->data_ptr() : nullptr, p_drop,
                      seed ? seed->data_ptr<long>() : nullptr, salt);
   HIP_CHECK_KERNEL();
-  hipLaunchKernelGGL((colsum_kernel<T>), dim3((D + 15) / 16), dim3(256), 0, stream,
-                     part.data_ptr<float>(), (T*)dw.data_ptr(), grid, D);
+  if (grid > 256)
+    hipLaunchKernelGGL((colsum_kernel<T, 64>), dim3((D + 3) / 4), dim3(256), 0,
+                       stream, part.data_ptr<float>(), (T*)dw.data_ptr(), grid, D);
+  else
+    hipLaunchKernelGGL((colsum_kernel<T, 16>), dim3((D + 15) / 16), dim3(256), 0,
+                       stream, part.data_ptr<float>(), (T*)dw.data_ptr(), grid, D);
   HIP_CHECK_KERNEL();
 }
 
