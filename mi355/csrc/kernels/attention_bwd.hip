@@ -74,7 +74,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
     const T* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
     const T* __restrict__ bias, const float* __restrict__ kpad,
     float* __restrict__ dq32, T* __restrict__ dk, T* __restrict__ dv,
-    float* __restrict__ dbias32,
+    float* __restrict__ dbias32, float* __restrict__ dkv32 /*q-split slabs*/,
     int B, int H, int Sq, int Sk,
     BStr qs, BStr ks, BStr vs, BStr dos, BStr dks, BStr dvs,
     long bias_sb, long bias_sh,
@@ -153,15 +153,21 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
   }
 
   const int diag_off = Sk - Sq;
+  // q-split (underfilled grids, e.g. B=2): gridDim.z chunks of q-tiles per
+  // (b,h,k-tile); dk/dv then go to per-chunk fp32 slabs reduced afterwards
+  const int ntiles_q = (Sq + BQT - 1) / BQT;
+  const int tpc = (ntiles_q + gridDim.z - 1) / gridDim.z;
+  const int qt_begin = blockIdx.z * tpc;
+  const int qt_end = min(ntiles_q, qt_begin + tpc);
   // causal: this k-tile sees queries with qr >= k0 - diag_off
-  int qt_start = 0;
-  if (CAUSAL) qt_start = max(0, (k0 - diag_off) / BQT);
+  int qt_start = qt_begin;
+  if (CAUSAL) qt_start = max(qt_begin, (k0 - diag_off) / BQT);
 
   T* my_slab = slab + wid * 16 * LDT;
 
   // causal-skipped q-tiles: write the zeros the reductions expect (cheaper
   // than pre-zeroing the whole 25 MB slab on the host)
-  for (int qt = 0; qt < qt_start; ++qt) {
+  for (int qt = qt_begin; qt < min(qt_start, qt_end); ++qt) {
     const int q0z = qt * BQT;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -180,7 +186,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
     }
   }
 
-  for (int qt = qt_start; qt * BQT < Sq; ++qt) {
+  for (int qt = qt_start; qt < qt_end; ++qt) {
     const int q0 = qt * BQT;
     __syncthreads();  // previous iteration's reads done before restage
     // ---- stage Q, dO (+ transposed images) ----
@@ -340,15 +346,31 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
     }
   }
 
-  // ---- write dK, dV (block exclusively owns these k rows) ----
+  // ---- write dK, dV ----
+  if (dkv32 == nullptr) {  // no q-split: block exclusively owns these k rows
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int kr = k0 + wid * 16 + lg * 4 + r;
-    if (kr >= Sk) continue;
+    for (int r = 0; r < 4; ++r) {
+      const int kr = k0 + wid * 16 + lg * 4 + r;
+      if (kr >= Sk) continue;
 #pragma unroll
-    for (int t = 0; t < D / 16; ++t) {
-      dkbase[(long)kr * dks.ss + t * 16 + lr] = DT<T>::from_f32(dk_acc[t][r]);
-      dvbase[(long)kr * dvs.ss + t * 16 + lr] = DT<T>::from_f32(dv_acc[t][r]);
+      for (int t = 0; t < D / 16; ++t) {
+        dkbase[(long)kr * dks.ss + t * 16 + lr] = DT<T>::from_f32(dk_acc[t][r]);
+        dvbase[(long)kr * dvs.ss + t * 16 + lr] = DT<T>::from_f32(dv_acc[t][r]);
+      }
+    }
+  } else {  // q-split: per-chunk fp32 slabs (2, NQC, B, H, Sk, D), plain stores
+    const long plane = (long)B * H * Sk * D;
+    float* dk_s = dkv32 + ((long)blockIdx.z * B * H + bh) * (long)Sk * D;
+    float* dv_s = dk_s + (long)gridDim.z * plane;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int kr = k0 + wid * 16 + lg * 4 + r;
+      if (kr >= Sk) continue;
+#pragma unroll
+      for (int t = 0; t < D / 16; ++t) {
+        dk_s[(long)kr * D + t * 16 + lr] = dk_acc[t][r];
+        dv_s[(long)kr * D + t * 16 + lr] = dv_acc[t][r];
+      }
     }
   }
 }
@@ -360,7 +382,8 @@ void launch_bwd(const torch::Tensor& dout, const torch::Tensor& q, const torch::
                 const c10::optional<torch::Tensor>& kpad, bool causal, bool need_dbias,
                 double p_drop, const c10::optional<torch::Tensor>& seed,
                 torch::Tensor& dq32, torch::Tensor& dk, torch::Tensor& dv,
-                torch::Tensor& dbias32, torch::Tensor& delta, hipStream_t stream) {
+                torch::Tensor& dbias32, torch::Tensor* dkv32, int nqc,
+                torch::Tensor& delta, hipStream_t stream) {
   const int B = q.size(0), H = q.size(1), Sq = q.size(2), Sk = k.size(2);
   BStr qs{q.stride(0), q.stride(1), q.stride(2)};
   BStr ks{k.stride(0), k.stride(1), k.stride(2)};
@@ -393,14 +416,15 @@ void launch_bwd(const torch::Tensor& dout, const torch::Tensor& q, const torch::
   const int ds_extra = (LD == LDT) ? 0 : BQT * LDT;  // ds_img aliases q_img at D==64
   const int smem = sizeof(T) * (BKT * LD * 2 + BQT * LD * 2 + D * LDT * 3 +
                                 ds_extra + 4 * 16 * LDT);
-  dim3 grid((Sk + BKT - 1) / BKT, B * H);
+  dim3 grid((Sk + BKT - 1) / BKT, B * H, nqc);
 #define LB(C, HB, HK, DB)                                                             \
   hipLaunchKernelGGL((attn_bwd_kernel<T, D, C, HB, HK, DB>), grid, dim3(256), smem,   \
                      stream, (const T*)dout.data_ptr(), (const T*)q.data_ptr(),       \
                      (const T*)k.data_ptr(), (const T*)v.data_ptr(),                  \
                      lse.data_ptr<float>(), delta.data_ptr<float>(), bias_p, kpad_p,  \
                      dq32.data_ptr<float>(), (T*)dk.data_ptr(), (T*)dv.data_ptr(),    \
-                     need_dbias ? dbias32.data_ptr<float>() : nullptr, B, H, Sq, Sk,  \
+                     need_dbias ? dbias32.data_ptr<float>() : nullptr,                \
+                     dkv32 ? dkv32->data_ptr<float>() : nullptr, B, H, Sq, Sk,        \
                      qs, ks, vs, dos, dks, dvs, bias_sb, bias_sh, (float)p_drop,       \
                      (seed.has_value() && seed->defined())                             \
                          ? seed->data_ptr<long>() : nullptr)
@@ -533,6 +557,19 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
   auto opts = q.options();
   const long nkt = (Sk + 63) / 64;
   auto dq32 = torch::empty({nkt, B, H, Sq, D}, opts.dtype(torch::kFloat));
+  // q-split factor: fill the 256 CUs (~2 waves of blocks) when B*H*nkt alone
+  // cannot; per-chunk dk/dv fp32 slabs are reduced by the shared epilogue
+  const int ntq = (Sq + 63) / 64;
+  int nqc = 1;
+  {
+    const long base_blocks = nkt * B * H;
+    while (base_blocks * nqc < 512 && nqc * 2 <= ntq) nqc *= 2;
+    const char* env = std::getenv("MI355_ATTN_BWD_NQC");
+    if (env) nqc = std::max(1, std::min(ntq, atoi(env)));
+  }
+  torch::Tensor dkv32;
+  if (nqc > 1)
+    dkv32 = torch::empty({2, (long)nqc, B, H, Sk, D}, opts.dtype(torch::kFloat));
   auto dk = torch::empty({B, Sk, H, D}, opts).permute({0, 2, 1, 3});
   auto dv = torch::empty({B, Sk, H, D}, opts).permute({0, 2, 1, 3});
   auto delta = torch::empty({B, H, Sq}, opts.dtype(torch::kFloat));
@@ -545,20 +582,33 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
   c10::optional<torch::Tensor> bias_c;
   if (bias.has_value() && bias->defined()) bias_c = bias->to(q.scalar_type()).contiguous();
 
+  torch::Tensor* kvs = nqc > 1 ? &dkv32 : nullptr;
   if (q.scalar_type() == torch::kBFloat16 && D == 64)
     launch_bwd<__hip_bfloat16, 64>(dout, q, k, v, o, lse, bias_c, kpad, causal,
-                                   need_dbias, p_drop, seed, dq32, dk, dv, dbias32, delta, stream);
+                                   need_dbias, p_drop, seed, dq32, dk, dv, dbias32, kvs, nqc, delta, stream);
   else if (q.scalar_type() == torch::kHalf && D == 64)
     launch_bwd<__half, 64>(dout, q, k, v, o, lse, bias_c, kpad, causal, need_dbias,
-                           p_drop, seed, dq32, dk, dv, dbias32, delta, stream);
+                           p_drop, seed, dq32, dk, dv, dbias32, kvs, nqc, delta, stream);
   else if (q.scalar_type() == torch::kBFloat16 && D == 32)
     launch_bwd<__hip_bfloat16, 32>(dout, q, k, v, o, lse, bias_c, kpad, causal,
-                                   need_dbias, p_drop, seed, dq32, dk, dv, dbias32, delta, stream);
+                                   need_dbias, p_drop, seed, dq32, dk, dv, dbias32, kvs, nqc, delta, stream);
   else if (q.scalar_type() == torch::kHalf && D == 32)
     launch_bwd<__half, 32>(dout, q, k, v, o, lse, bias_c, kpad, causal, need_dbias,
-                           p_drop, seed, dq32, dk, dv, dbias32, delta, stream);
+                           p_drop, seed, dq32, dk, dv, dbias32, kvs, nqc, delta, stream);
   else
     TORCH_CHECK(false, "attn_bwd: unsupported dtype/D=", D);
+  if (nqc > 1) {
+    // reduce the per-chunk dk/dv slabs (same fused pass as dq)
+    auto dk32 = dkv32.select(0, 0);
+    auto dv32 = dkv32.select(0, 1);
+    if (q.scalar_type() == torch::kBFloat16) {
+      launch_epilogue<__hip_bfloat16>(dk32, dk, nullptr, nullptr, stream);
+      launch_epilogue<__hip_bfloat16>(dv32, dv, nullptr, nullptr, stream);
+    } else {
+      launch_epilogue<__half>(dk32, dk, nullptr, nullptr, stream);
+      launch_epilogue<__half>(dv32, dv, nullptr, nullptr, stream);
+    }
+  }
 
   // dq: sum the per-k-tile fp32 slabs -> low-precision BSHD-physical view;
   // dbias: batch-sum + cast. Single fused pass each (no aten sum/copy/cast).
