@@ -563,7 +563,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
   int nqc = 1;
   {
     const long base_blocks = nkt * B * H;
-    while (base_blocks * nqc < 512 && nqc * 2 <= ntq) nqc *= 2;
+    while (base_blocks * nqc < 384 && nqc * 2 <= ntq) nqc *= 2;
     const char* env = std::getenv("MI355_ATTN_BWD_NQC");
     if (env) nqc = std::max(1, std::min(ntq, atoi(env)));
   }
