@@ -62,6 +62,7 @@ __global__ __launch_bounds__(256) void attn_fwd_v2_kernel(
     const T* __restrict__ q, const T* __restrict__ k, const T* __restrict__ v,
     const T* __restrict__ bias, const float* __restrict__ kpad,
     T* __restrict__ out, float* __restrict__ lse,
+    float* __restrict__ opart, float* __restrict__ mpart, float* __restrict__ lpart,
     int B, int H, int Sq, int Sk,
     S3 qs, S3 ks, S3 vs, S3 os, long bias_sb, long bias_sh,
     float p_drop, const long* __restrict__ seed_ptr) {
@@ -155,12 +156,19 @@ __global__ __launch_bounds__(256) void attn_fwd_v2_kernel(
     }
   };
 
-  if (k_end > 0) stage_load(0);
-  for (int k0 = 0; k0 < k_end; k0 += BK2) {
+  // k-split (underfilled grids): gridDim.z chunks of k-tiles; partial
+  // (o_acc, m, l) per chunk are combined by attn_fwd_merge_kernel
+  const int nkt_all = (Sk + BK2 - 1) / BK2;
+  const int tpk = (nkt_all + gridDim.z - 1) / gridDim.z;
+  const int kz0 = blockIdx.z * tpk * BK2;
+  const int kz_end = min(k_end, kz0 + tpk * BK2);
+
+  if (kz_end > kz0) stage_load(kz0);
+  for (int k0 = kz0; k0 < kz_end; k0 += BK2) {
     __syncthreads();  // previous tile's LDS consumers done
     stage_write();
     __syncthreads();
-    if (k0 + BK2 < k_end) stage_load(k0 + BK2);
+    if (k0 + BK2 < kz_end) stage_load(k0 + BK2);
 
     // ---- S^T = K Q^T: A = K rows, B = Q^T ----
     f32x16 st;
@@ -261,23 +269,86 @@ __global__ __launch_bounds__(256) void attn_fwd_v2_kernel(
 
   // ---- epilogue: normalize, store O rows (4-elem packs) + LSE ----
   if (my_q < Sq) {
-    const float invl = (l_i > 0.f) ? 1.f / l_i : 0.f;
-    T* orow = obase + (long)my_q * os.ss;
+    if (opart == nullptr) {  // no k-split: final store
+      const float invl = (l_i > 0.f) ? 1.f / l_i : 0.f;
+      T* orow = obase + (long)my_q * os.ss;
 #pragma unroll
-    for (int dt = 0; dt < 2; ++dt) {
+      for (int dt = 0; dt < 2; ++dt) {
 #pragma unroll
-      for (int grp = 0; grp < 4; ++grp) {
-        const int d0 = dt * 32 + grp * 8 + half * 4;
-        VecT<T, 4> ov;
+        for (int grp = 0; grp < 4; ++grp) {
+          const int d0 = dt * 32 + grp * 8 + half * 4;
+          VecT<T, 4> ov;
 #pragma unroll
-        for (int u = 0; u < 4; ++u)
-          ov.data[u] = DT<T>::from_f32(o_acc[dt][grp * 4 + u] * invl);
-        vstore<T, 4>(orow + d0, ov);
+          for (int u = 0; u < 4; ++u)
+            ov.data[u] = DT<T>::from_f32(o_acc[dt][grp * 4 + u] * invl);
+          vstore<T, 4>(orow + d0, ov);
+        }
+      }
+      if (half == 0)
+        lse[(long)bh * Sq + my_q] = (l_i > 0.f) ? m_i + logf(l_i) : -INFINITY;
+    } else {  // k-split: raw fp32 partial accumulator + (m, l) per chunk
+      const long zrow = ((long)blockIdx.z * gridDim.y + bh) * Sq + my_q;
+      float* op = opart + zrow * D;
+#pragma unroll
+      for (int dt = 0; dt < 2; ++dt) {
+#pragma unroll
+        for (int grp = 0; grp < 4; ++grp) {
+          const int d0 = dt * 32 + grp * 8 + half * 4;
+          VecT<float, 4> ov;
+#pragma unroll
+          for (int u = 0; u < 4; ++u) ov.data[u] = o_acc[dt][grp * 4 + u];
+          vstore<float, 4>(op + d0, ov);
+        }
+      }
+      if (half == 0) {
+        mpart[zrow] = m_i;
+        lpart[zrow] = l_i;
       }
     }
-    if (half == 0)
-      lse[(long)bh * Sq + my_q] = (l_i > 0.f) ? m_i + logf(l_i) : -INFINITY;
   }
+}
+
+// combine k-split partials: o = sum_c exp(m_c - M) * o_c / L,
+// L = sum_c l_c * exp(m_c - M), lse = M + log L. One thread per (bh, q, 16
+// d-quads); M/L recomputed per quad (tiny vs the fp32 partial traffic).
+template <typename T>
+__global__ void attn_fwd_merge_kernel(const float* __restrict__ opart,
+                                      const float* __restrict__ mpart,
+                                      const float* __restrict__ lpart,
+                                      T* __restrict__ out, float* __restrict__ lse,
+                                      int nkc, long BHSq, int H, int Sq,
+                                      S3 os) {
+  constexpr int D = 64;
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= BHSq * (D / 4)) return;
+  const int dq = (int)(idx % (D / 4));
+  const long row = idx / (D / 4);  // bh * Sq + q
+  float M = -INFINITY;
+  for (int c = 0; c < nkc; ++c) M = fmaxf(M, mpart[c * BHSq + row]);
+  float L = 0.f;
+  for (int c = 0; c < nkc; ++c) {
+    const float mc = mpart[c * BHSq + row];
+    if (mc > -INFINITY) L += lpart[c * BHSq + row] * __expf(mc - M);
+  }
+  const float invl = (L > 0.f) ? 1.f / L : 0.f;
+  float acc[4] = {0.f, 0.f, 0.f, 0.f};
+  for (int c = 0; c < nkc; ++c) {
+    const float mc = mpart[c * BHSq + row];
+    if (!(mc > -INFINITY)) continue;
+    const float w = __expf(mc - M) * invl;
+    VecT<float, 4> ov = vload<float, 4>(opart + (c * BHSq + row) * D + dq * 4);
+#pragma unroll
+    for (int u = 0; u < 4; ++u) acc[u] += w * ov.data[u];
+  }
+  const long bh = row / Sq;
+  const int qr = (int)(row % Sq);
+  const int b = (int)(bh / H), h = (int)(bh % H);
+  T* orow = out + (long)b * os.sb + (long)h * os.sh + (long)qr * os.ss;
+  VecT<T, 4> tv;
+#pragma unroll
+  for (int u = 0; u < 4; ++u) tv.data[u] = DT<T>::from_f32(acc[u]);
+  vstore<T, 4>(orow + dq * 4, tv);
+  if (dq == 0) lse[row] = (L > 0.f) ? M + logf(L) : -INFINITY;
 }
 
 template <typename T>
@@ -287,7 +358,29 @@ void launch_v2(const torch::Tensor& q, const torch::Tensor& k, const torch::Tens
                double p_drop, const c10::optional<torch::Tensor>& seed,
                torch::Tensor& out, torch::Tensor& lse, hipStream_t stream) {
   const int B = q.size(0), H = q.size(1), Sq = q.size(2), Sk = k.size(2);
-  dim3 grid((Sq + BQ2 - 1) / BQ2, B * H);
+  // k-split factor (flash-decoding style) for underfilled grids
+  const int qblocks = (Sq + BQ2 - 1) / BQ2;
+  const int nkt = (Sk + BK2 - 1) / BK2;
+  int nkc = 1;
+  {
+    const long base_blocks = (long)qblocks * B * H;
+    while (base_blocks * nkc < 384 && nkc * 2 <= nkt / 4) nkc *= 2;
+    const char* env = std::getenv("MI355_ATTN_FWD_NKC");
+    if (env) nkc = std::max(1, std::min(nkt, atoi(env)));
+  }
+  torch::Tensor opart, mpart, lpart;
+  float *opart_p = nullptr, *mpart_p = nullptr, *lpart_p = nullptr;
+  if (nkc > 1) {
+    const long BHSq = (long)B * H * Sq;
+    auto f32 = q.options().dtype(torch::kFloat);
+    opart = torch::empty({(long)nkc, BHSq, 64}, f32);
+    mpart = torch::empty({(long)nkc, BHSq}, f32);
+    lpart = torch::empty({(long)nkc, BHSq}, f32);
+    opart_p = opart.data_ptr<float>();
+    mpart_p = mpart.data_ptr<float>();
+    lpart_p = lpart.data_ptr<float>();
+  }
+  dim3 grid(qblocks, B * H, nkc);
   S3 qst{q.stride(0), q.stride(1), q.stride(2)};
   S3 kst{k.stride(0), k.stride(1), k.stride(2)};
   S3 vst{v.stride(0), v.stride(1), v.stride(2)};
@@ -305,7 +398,8 @@ void launch_v2(const torch::Tensor& q, const torch::Tensor& k, const torch::Tens
   hipLaunchKernelGGL((attn_fwd_v2_kernel<T, C, HB, HK>), grid, dim3(256), 0,        \
                      stream, (const T*)q.data_ptr(), (const T*)k.data_ptr(),        \
                      (const T*)v.data_ptr(), bias_p, kpad_p, (T*)out.data_ptr(),    \
-                     lse.data_ptr<float>(), B, H, Sq, Sk, qst, kst, vst, ost,       \
+                     lse.data_ptr<float>(), opart_p, mpart_p, lpart_p,              \
+                     B, H, Sq, Sk, qst, kst, vst, ost,                              \
                      bias_sb, bias_sh, (float)p_drop,                               \
                      (seed.has_value() && seed->defined()) ? seed->data_ptr<long>() \
                                                            : nullptr)
@@ -322,6 +416,15 @@ void launch_v2(const torch::Tensor& q, const torch::Tensor& k, const torch::Tens
   }
 #undef LV2
   HIP_CHECK_KERNEL();
+  if (nkc > 1) {
+    const long BHSq = (long)B * H * Sq;
+    const long total = BHSq * 16;  // one thread per d-quad
+    hipLaunchKernelGGL((attn_fwd_merge_kernel<T>), dim3((total + 255) / 256),
+                       dim3(256), 0, stream, opart_p, mpart_p, lpart_p,
+                       (T*)out.data_ptr(), lse.data_ptr<float>(), nkc, BHSq,
+                       H, Sq, ost);
+    HIP_CHECK_KERNEL();
+  }
 }
 
 }  // namespace
