@@ -51,11 +51,7 @@ __global__ void attn_delta_kernel(const T* __restrict__ dout, const T* __restric
                                   float* __restrict__ delta, int H, int Sq, int D,
                                   BStr dos, BStr os) {
   const long row = (long)blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
-  const long total = (long)gridDim.y * 0 + 0;  // unused
-  (void)total;
   const int lane = threadIdx.x & (WAVE - 1);
-  const long nrows = (long)Sq * H * gridDim.y;  // gridDim.y == B
-  (void)nrows;
   const int b = blockIdx.y;
   if (row >= (long)H * Sq) return;
   const int h = row / Sq, qr = row % Sq;
