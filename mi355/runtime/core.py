@@ -288,9 +288,12 @@ class Runtime:
             p.start()
             self.workers.append((p, tq, gpu_id))
             self._idle.put(wid)
-        self._reaper = threading.Thread(target=self._reap, daemon=True)
+        self._pending: "queue.Queue[tuple]" = queue.Queue()
         self._alive = True
+        self._reaper = threading.Thread(target=self._reap, daemon=True)
         self._reaper.start()
+        self._dispatcher = threading.Thread(target=self._dispatch, daemon=True)
+        self._dispatcher.start()
         self.actors: list = []
         atexit.register(self.shutdown)
 
@@ -307,23 +310,29 @@ class Runtime:
             self._idle.put(wid)
 
     # -- tasks ---------------------------------------------------------- #
+    # Shared-backlog pull scheduling: submit never blocks (Ray semantics) —
+    # tasks land in one queue and the dispatcher hands each to the next idle
+    # worker (idle workers "steal" from the common backlog, so an unlucky
+    # long task never strands work behind it).
     def submit(self, fn, args, kwargs, num_gpus: float = 0) -> ObjectRef:
         ref = ObjectRef(uuid.uuid4().hex)
-        wid = self._pick_worker(num_gpus)
         fn_bytes = fn if isinstance(fn, bytes) else cloudpickle.dumps(fn)
         with self._lock:
-            self._inflight[ref.id] = wid
-        self.workers[wid][1].put(("task", ref.id, fn_bytes, list(args), dict(kwargs)))
+            self._inflight[ref.id] = -1  # queued, not yet on a worker
+        self._pending.put((ref.id, fn_bytes, list(args), dict(kwargs)))
         return ref
 
-    def _pick_worker(self, num_gpus: float = 0) -> int:
-        # workers are GPU-pinned round-robin; any worker satisfies a CPU task
-        while True:
-            wid = self._idle.get()
-            if num_gpus <= 0 or self.workers[wid][2] is not None or self.num_gpus == 0:
-                return wid
-            self._idle.put(wid)  # CPU-only worker, GPU requested: rotate
-            time.sleep(0.0005)
+    def _dispatch(self):
+        while self._alive:
+            try:
+                ref_id, fn_bytes, args, kwargs = self._pending.get(timeout=0.2)
+            except queue.Empty:
+                continue
+            wid = self._idle.get()  # next worker to go idle takes the task
+            with self._lock:
+                if ref_id in self._inflight:
+                    self._inflight[ref_id] = wid
+            self.workers[wid][1].put(("task", ref_id, fn_bytes, args, kwargs))
 
     # -- store ---------------------------------------------------------- #
     def put(self, obj) -> ObjectRef:

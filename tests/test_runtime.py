@@ -95,3 +95,26 @@ def test_actor_pool_map_ordered():
 
     pool = mi355.ActorPool([Worker.remote() for _ in range(2)])
     assert pool.map(lambda a, v: a.process.remote(v), [3, 1, 2]) == [6, 2, 4]
+
+
+def test_submit_nonblocking_backlog():
+    """submit returns immediately even with more tasks than workers (shared
+    backlog, pull scheduling); all tasks still complete with correct results."""
+    import time as _t
+
+    mi355.init(num_workers=2)
+    try:
+        @mi355.remote
+        def slow(i):
+            import time
+
+            time.sleep(0.2)
+            return i * i
+
+        t0 = _t.monotonic()
+        refs = [slow.remote(i) for i in range(8)]
+        submit_time = _t.monotonic() - t0
+        assert submit_time < 1.0, f"submit blocked: {submit_time:.2f}s"
+        assert mi355.get(refs) == [i * i for i in range(8)]
+    finally:
+        mi355.shutdown()
