@@ -127,3 +127,20 @@ def test_ddp_grad_equals_full_batch_grad():
     assert torch.allclose(res[0], res[1], atol=1e-6)
     assert torch.allclose(res[0].float(), ref.float(), atol=1e-4), \
         (res[0] - ref).abs().max()
+
+
+@pytest.mark.timeout(300)
+def test_flat_replica_four_ranks_stay_in_sync():
+    """collective correctness at world sizes beyond 2 —
+    4 gloo ranks, bucketed all-reduce, identical params after a step."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29537
+    procs = [ctx.Process(target=_worker, args=(r, 4, port, q)) for r in range(4)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=240) for _ in range(4)]
+    for p in procs:
+        p.join(timeout=60)
+    sums = [r[3] for r in results]
+    assert all(s == pytest.approx(sums[0], rel=1e-9) for s in sums), sums
