@@ -78,38 +78,61 @@ def run_train(args, device, dtype):
     optimizer = FlatAdamW(replica, lr=2e-5, weight_decay=0.01, max_grad_norm=1.0)
     input_ids, labels = synthetic_batch(args, device, seed=1000 + get_rank())
 
-    # auto mode: capture only single-GPU steps (RCCL-inside-hipGraph is not
-    # validated on the 8-GPU path; eager overlap is the safe default there)
-    use_graphs = False
-    if device.type == "cuda" and (args.graphs == "on" or
-                                  (args.graphs == "auto" and get_world_size() <= 1)):
-        use_graphs = True
+    # graph strategy:
+    #  * world == 1: ONE full-step hipGraph (fwd+bwd+adamw)
+    #  * world  > 1: TWO graphs (g1 = zero+fwd+bwd, g2 = clip+adamw) with the
+    #    bucketed RCCL all-reduce EAGER between the replays — no collective is
+    #    ever captured (RCCL-inside-hipGraph stays a round-2 experiment), but
+    #    the ~700 per-step launches that made the eager path ~60% slower at
+    #    per-GPU batch 2 are gone. MI355_DIST_GRAPHS=0 forces eager.
+    use_graphs = device.type == "cuda" and args.graphs != "off"
+    hybrid = use_graphs and get_world_size() > 1 \
+        and os.environ.get("MI355_DIST_GRAPHS", "1") != "0"
+    if device.type == "cuda" and os.environ.get("MI355_HYBRID_GRAPH") == "1":
+        hybrid = use_graphs = True  # 1-GPU validation knob for the hybrid path
 
     step_fn = lambda: train_step(replica, optimizer, input_ids, labels)
 
     for _ in range(max(args.warmup, 1)):
         step_fn()
 
-    graph = None
+    graph = graph2 = None
     if use_graphs:
         try:
+            barrier()
             torch.cuda.synchronize()
-            g = torch.cuda.CUDAGraph()
-            # capture one full step (fwd+bwd+allreduce+adamw) as a hipGraph
-            with torch.cuda.graph(g):
-                step_fn()
-            graph = g
-            graph.replay()  # warm replay
+            if hybrid:
+                g1 = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g1):
+                    optimizer.zero_grad()
+                    out = replica(input_ids=input_ids, labels=labels)
+                    out.loss.backward()  # hooks skip collectives while capturing
+                g2 = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g2):
+                    optimizer.step()
+                graph, graph2 = g1, g2
+            else:
+                g = torch.cuda.CUDAGraph()
+                # capture one full step (fwd+bwd+allreduce+adamw) as a hipGraph
+                with torch.cuda.graph(g):
+                    step_fn()
+                graph = g
             torch.cuda.synchronize()
+            barrier()
         except Exception as e:
             if args.graphs == "on":
                 raise
             if get_rank() == 0:
                 print(f"# graph capture unavailable, eager fallback: {e!r}")
-            graph = None
+            graph = graph2 = None
 
     def timed_step():
-        if graph is not None:
+        if graph is not None and graph2 is not None:
+            graph.replay()
+            replica.all_reduce_grads()
+            replica.finish_grad_sync()
+            graph2.replay()
+        elif graph is not None:
             graph.replay()
         else:
             step_fn()

@@ -116,6 +116,11 @@ class FlatReplica:
 
     def _make_hook(self, gi, pi):
         def hook(_param):
+            # hipGraph capture of the backward (two-graph hybrid, bench.py):
+            # collectives must NOT be recorded into the graph — the replayed
+            # backward is followed by an explicit all_reduce_grads() instead
+            if torch.cuda.is_available() and torch.cuda.is_current_stream_capturing():
+                return
             _, g = self.groups[gi]
             bi = g.param_bucket[pi]
             g.bucket_ready[bi] += 1
@@ -132,6 +137,23 @@ class FlatReplica:
         work = dist.all_reduce(view, op=dist.ReduceOp.SUM, group=self.pg, async_op=True)
         self._works.append(work)
         self._flushed.add((gi, bi))
+
+    def all_reduce_grads(self):
+        """Explicit bucketed all-reduce of the flat grad buffers. Used by the
+        two-graph hybrid step (backward replayed from a hipGraph, so the
+        post-accumulate hooks never fire): g1.replay() -> all_reduce_grads()
+        -> finish_grad_sync() -> g2.replay(). The flat buffers are static
+        addresses, so this is safe across graph replays."""
+        if not is_distributed():
+            return
+        for gi, (_, g) in enumerate(self.groups):
+            for bi, (bs, be, _last) in enumerate(g.buckets):
+                if (gi, bi) in self._flushed:
+                    continue
+                work = dist.all_reduce(g.flat_grad[bs:be], op=dist.ReduceOp.SUM,
+                                       group=self.pg, async_op=True)
+                self._works.append(work)
+                self._flushed.add((gi, bi))
 
     # ------------------------------------------------------------------ #
     def finish_grad_sync(self):
