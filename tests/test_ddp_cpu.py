@@ -144,3 +144,59 @@ def test_flat_replica_four_ranks_stay_in_sync():
         p.join(timeout=60)
     sums = [r[3] for r in results]
     assert all(s == pytest.approx(sums[0], rel=1e-9) for s in sums), sums
+
+
+def _worker_explicit(rank, world, port, q):
+    """Hook-less backward + explicit all_reduce_grads(): the path the
+    two-graph hybrid step uses after a hipGraph backward replay (hooks never
+    fire) — must produce the same averaged grads as the hook path."""
+    os.environ.update({
+        "RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": str(world),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    from mi355.models.t5.configuration import t5_config
+    from mi355.models.t5.modeling import T5ForConditionalGeneration
+    from mi355.parallel import FlatReplica, init_distributed
+
+    init_distributed(backend="gloo")
+    torch.manual_seed(7)
+    cfg = t5_config("flan-t5-small", num_layers=1, num_decoder_layers=1, d_model=32,
+                    d_ff=64, num_heads=2, d_kv=16, vocab_size=128, dropout_rate=0.0)
+    model = T5ForConditionalGeneration(cfg)
+    replica = FlatReplica(model, bucket_mb=0.01)
+    for h in replica._hooks:  # emulate graph replay: hooks don't run
+        h.remove()
+    torch.manual_seed(55)
+    ids = torch.randint(0, 128, (4, 8))
+    my = ids[rank * 2:(rank + 1) * 2]
+    replica.zero_grads()
+    out = replica(input_ids=my, labels=my.clone())
+    out.loss.backward()
+    replica.all_reduce_grads()
+    replica.finish_grad_sync()
+    grad = torch.cat([g.flat_grad for _, g in replica.groups]).clone()
+    q.put((rank, grad.numpy().copy()))
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_explicit_allreduce_matches_hook_path():
+    ctx = mp.get_context("spawn")
+    q1, q2 = ctx.Queue(), ctx.Queue()
+    procs = [ctx.Process(target=_worker_equiv, args=(r, 2, 29541, q1)) for r in range(2)]
+    procs += [ctx.Process(target=_worker_explicit, args=(r, 2, 29543, q2)) for r in range(2)]
+    for p in procs[:2]:
+        p.start()
+    hook_grads = {r: g for r, g in (q1.get(timeout=240), q1.get(timeout=240))}
+    for p in procs[:2]:
+        p.join(timeout=60)
+    for p in procs[2:]:
+        p.start()
+    expl_grads = {r: g for r, g in (q2.get(timeout=240), q2.get(timeout=240))}
+    for p in procs[2:]:
+        p.join(timeout=60)
+    import numpy as np
+
+    for r in (0, 1):
+        np.testing.assert_allclose(hook_grads[r], expl_grads[r], rtol=1e-6, atol=1e-7)
