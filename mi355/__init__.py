@@ -4,7 +4,7 @@ shared-memory object store, Arrow-block datasets + preprocessors, a DDP
 trainer on RCCL-over-xGMI, ASHA tuning, batch + online inference, and a
 model zoo running hand-written HIP/CDNA4 kernels."""
 from .runtime import (ObjectRef, init, is_initialized, shutdown, put, get,
-                      wait, remote, ActorPool)
+                      wait, remote, ActorPool, timeline)
 from .runtime.actor import ActorDiedError
 from . import metrics
 
@@ -14,4 +14,4 @@ from . import ops  # noqa: E402  (dispatch layer; harmless on CPU)
 
 __all__ = ["init", "is_initialized", "shutdown", "put", "get", "wait",
            "remote", "ActorPool", "ObjectRef", "ActorDiedError", "ops",
-           "metrics", "__version__"]
+           "metrics", "timeline", "__version__"]

@@ -240,8 +240,11 @@ def _worker_main(wid, gpu_id, task_q, done_q):
         if kind == "task":
             _, ref_id, fn_bytes, args, kwargs = msg
             ref = ObjectRef(ref_id)
+            t0 = time.monotonic()
+            name = "task"
             try:
                 fn = cloudpickle.loads(fn_bytes)
+                name = getattr(fn, "__name__", "task")
                 a, kw = _resolve_args(store, args, kwargs)
                 result = fn(*a, **kw)
                 store.put(result, ref)
@@ -249,7 +252,7 @@ def _worker_main(wid, gpu_id, task_q, done_q):
                 import traceback
 
                 store.put(_RemoteError(e, traceback.format_exc()), ref)
-            done_q.put((wid, ref_id))
+            done_q.put((wid, ref_id, name, t0, time.monotonic()))
     store.cleanup()
 
 
@@ -289,6 +292,7 @@ class Runtime:
             self.workers.append((p, tq, gpu_id))
             self._idle.put(wid)
         self._pending: "queue.Queue[tuple]" = queue.Queue()
+        self._events: list[tuple] = []  # (wid, task_name, t_start, t_end)
         self._alive = True
         self._reaper = threading.Thread(target=self._reap, daemon=True)
         self._reaper.start()
@@ -300,13 +304,14 @@ class Runtime:
     def _reap(self):
         while self._alive:
             try:
-                wid, ref_id = self.done_q.get(timeout=0.2)
+                wid, ref_id, name, t0, t1 = self.done_q.get(timeout=0.2)
             except queue.Empty:
                 continue
             except (EOFError, OSError):
                 return
             with self._lock:
                 self._inflight.pop(ref_id, None)
+                self._events.append((wid, name, t0, t1))
             self._idle.put(wid)
 
     # -- tasks ---------------------------------------------------------- #
@@ -333,6 +338,22 @@ class Runtime:
                 if ref_id in self._inflight:
                     self._inflight[ref_id] = wid
             self.workers[wid][1].put(("task", ref_id, fn_bytes, args, kwargs))
+
+    def timeline(self, filename: str | None = None):
+        """Chrome-trace JSON of completed task events per worker (SURVEY.md
+        §5.1; reference surface: ray.timeline()). Load in chrome://tracing
+        or Perfetto."""
+        with self._lock:
+            events = list(self._events)
+        trace = [{"name": name, "ph": "X", "pid": 0, "tid": wid,
+                  "ts": int(t0 * 1e6), "dur": int((t1 - t0) * 1e6),
+                  "cat": "task"} for wid, name, t0, t1 in events]
+        if filename:
+            import json
+
+            with open(filename, "w") as f:
+                json.dump(trace, f)
+        return trace
 
     # -- store ---------------------------------------------------------- #
     def put(self, obj) -> ObjectRef:
@@ -422,6 +443,11 @@ def _rt() -> Runtime:
     if _runtime is None:
         init()
     return _runtime
+
+
+def timeline(filename: str | None = None):
+    """Chrome-trace JSON of completed task events (ray.timeline parity)."""
+    return _rt().timeline(filename)
 
 
 def put(obj) -> ObjectRef:

@@ -3,7 +3,7 @@ notebooks' call sites keep working verbatim — `ray.init()`, `ray.put/get`,
 `@ray.remote`, `ray.data.from_huggingface`, `ray.train.huggingface.
 HuggingFaceTrainer`, `ray.tune`, `ray.air`, `ray.serve`. This is an alias
 layer only; the implementation is mi355's MI355X-native stack."""
-from mi355 import (ActorPool, ObjectRef, get, init, is_initialized, put,
+from mi355 import (ActorPool, ObjectRef, get, init, is_initialized, put, timeline,
                    remote, shutdown, wait)
 from mi355 import serve, tune  # noqa: F401
 from . import air, data, exceptions, train, util  # noqa: F401

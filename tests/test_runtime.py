@@ -118,3 +118,24 @@ def test_submit_nonblocking_backlog():
         assert mi355.get(refs) == [i * i for i in range(8)]
     finally:
         mi355.shutdown()
+
+
+def test_timeline_chrome_trace(tmp_path):
+    """ray.timeline parity: completed task events exported as Chrome trace."""
+    mi355.init(num_workers=2)
+    try:
+        @mi355.remote
+        def named_work(i):
+            return i + 1
+
+        mi355.get([named_work.remote(i) for i in range(4)])
+        out = tmp_path / "trace.json"
+        events = mi355.timeline(str(out))
+        assert len(events) >= 4
+        assert all(e["ph"] == "X" and "dur" in e for e in events)
+        import json
+
+        assert json.load(open(out)) == events
+        assert any(e["name"] == "named_work" for e in events)
+    finally:
+        mi355.shutdown()
