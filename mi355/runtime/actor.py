@@ -24,13 +24,16 @@ class ActorDiedError(RuntimeError):
     """The actor process exited (crash/OOM/os._exit) with calls in flight."""
 
 
-def _actor_main(gpu_id, cls_bytes, init_args, init_kwargs, task_q, done_q):
+def _actor_main(gpu_id, cls_bytes, init_bytes, task_q, done_q):
     if gpu_id is not None:
         os.environ["HIP_VISIBLE_DEVICES"] = str(gpu_id)
         os.environ["CUDA_VISIBLE_DEVICES"] = str(gpu_id)
     store = ObjectStore()
     try:
         cls = cloudpickle.loads(cls_bytes)
+        # init args cloudpickled as one blob: plain pickle (mp spawn) cannot
+        # carry local classes / closures that actor constructors often take
+        init_args, init_kwargs = cloudpickle.loads(init_bytes)
         a, kw = _resolve(store, init_args, init_kwargs)
         instance = cls(*a, **kw)
         done_q.put(("ready", None))
@@ -79,8 +82,7 @@ class ActorHandle:
         if num_gpus and rt.num_gpus:
             self._gpu_id = next(_gpu_rr) % rt.num_gpus
         self._cls_bytes = cloudpickle.dumps(cls)
-        self._init_args = list(init_args)
-        self._init_kwargs = dict(init_kwargs)
+        self._init_bytes = cloudpickle.dumps((list(init_args), dict(init_kwargs)))
         self._restarts_left = max_restarts
         self._pending: set[str] = set()
         self._spawn()
@@ -93,8 +95,8 @@ class ActorHandle:
         self._done_q = ctx.Queue()
         self._proc = ctx.Process(
             target=_actor_main,
-            args=(self._gpu_id, self._cls_bytes, self._init_args,
-                  self._init_kwargs, self._task_q, self._done_q),
+            args=(self._gpu_id, self._cls_bytes, self._init_bytes,
+                  self._task_q, self._done_q),
             daemon=True,
         )
         self._proc.start()

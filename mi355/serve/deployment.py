@@ -62,6 +62,28 @@ class Application:
     kwargs: dict
 
 
+def _runtime_up() -> bool:
+    import mi355
+
+    return mi355.is_initialized()
+
+
+class _ActorReplica:
+    """Serve replica hosted in its own actor process (loads the model once)."""
+
+    def __init__(self, predictor_cls, checkpoint, kwargs):
+        self.predictor = predictor_cls.from_checkpoint(checkpoint, **kwargs)
+
+    def predict(self, data, kw):
+        return self.predictor.predict(data, **kw)
+
+
+def _invoke_actor(rep, data, kw):
+    import mi355
+
+    return mi355.get(rep.predict.remote(data, kw))
+
+
 class _PredictorReplicaGroup:
     """Round-robin pool of predictor replicas for a PredictorDeployment."""
 
@@ -72,9 +94,20 @@ class _PredictorReplicaGroup:
         if app.args and hasattr(app.args[0], "from_checkpoint"):
             # PredictorDeployment.bind(PredictorCls, checkpoint, ...)
             predictor_cls, checkpoint = app.args[0], app.args[1]
-            self.replicas = [predictor_cls.from_checkpoint(checkpoint, **rest)
-                             for _ in range(d.num_replicas)]
-            self._invoke = lambda rep, data, kw: rep.predict(data, **kw)
+            if d.ray_actor_options and _runtime_up():
+                # replica = dedicated actor process (GPU-pinned round-robin
+                # when num_gpus>0) — the reference's serve replica model
+                from ..runtime.actor import ActorClass
+
+                num_gpus = d.ray_actor_options.get("num_gpus", 0)
+                cls = ActorClass(_ActorReplica, num_gpus=num_gpus)
+                self.replicas = [cls.remote(predictor_cls, checkpoint, rest)
+                                 for _ in range(d.num_replicas)]
+                self._invoke = _invoke_actor
+            else:
+                self.replicas = [predictor_cls.from_checkpoint(checkpoint, **rest)
+                                 for _ in range(d.num_replicas)]
+                self._invoke = lambda rep, data, kw: rep.predict(data, **kw)
         elif isinstance(d.func_or_class, type):
             # generic @serve.deployment class: instance is the handler
             self.replicas = [d.func_or_class(*app.args, **rest)

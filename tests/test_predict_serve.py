@@ -99,3 +99,55 @@ def test_serve_http_roundtrip():
             assert e.code == 404
     finally:
         serve.shutdown()
+
+
+def test_serve_actor_backed_replicas():
+    """ray_actor_options + initialized runtime -> each replica is its own
+    actor process (the reference's serve replica model)."""
+    import json
+    import socket
+    import urllib.request
+
+    import mi355
+    from mi355 import serve
+    from mi355.air.checkpoint import Checkpoint
+    from mi355.predict.predictor import Predictor
+
+    class EchoPredictor(Predictor):
+        def __init__(self, scale, preprocessor=None):
+            super().__init__(preprocessor)
+            self.scale = scale
+
+        @classmethod
+        def from_checkpoint(cls, checkpoint, **kw):
+            return cls(checkpoint.to_dict()["scale"])
+
+        def predict(self, df, **kw):
+            return df.assign(y=df["x"] * self.scale)
+
+    mi355.init(num_workers=1)
+    try:
+        ckpt = Checkpoint.from_dict({"scale": 3})
+        s = socket.socket(); s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]; s.close()
+        app = serve.PredictorDeployment.options(
+            name="actor-echo", route_prefix="/actor-echo", num_replicas=2,
+            ray_actor_options={"num_gpus": 0},
+        ).bind(EchoPredictor, ckpt, http_adapter=serve.pandas_read_json)
+        serve.run(app, port=port)
+        import sys
+
+        dmod = sys.modules["mi355.serve.deployment"]
+        group = dmod._state.routes["/actor-echo"]
+        from mi355.runtime.actor import ActorHandle
+
+        assert all(isinstance(r, ActorHandle) for r in group.replicas)
+        req = urllib.request.Request(
+            f"http://127.0.0.1:{port}/actor-echo",
+            data=json.dumps({"x": [1, 2]}).encode(), method="POST")
+        with urllib.request.urlopen(req) as r:
+            body = json.loads(r.read())
+        assert [row["y"] for row in body] == [3, 6]
+    finally:
+        serve.shutdown()
+        mi355.shutdown()
