@@ -148,9 +148,14 @@ class Trainer:
                                             args.per_device_train_batch_size,
                                             device, shuffle_seed=args.seed + epoch):
                 inputs = self._model_inputs(batch)
+                # zero BEFORE forward: zero_grads also advances the fused
+                # dropout seed, and fwd+bwd of one step must see the same masks
+                if fused:
+                    optimizer.zero_grad()
+                else:
+                    replica.zero_grads()
                 out = self.model(**inputs)
                 loss = out.loss if hasattr(out, "loss") else out["loss"]
-                replica.zero_grads() if not fused else optimizer.zero_grad()
                 loss.backward()
                 replica.finish_grad_sync()
                 if fused:
@@ -224,9 +229,10 @@ class Trainer:
         if is_distributed():  # M4: aggregate eval metric across workers
             import torch.distributed as dist
 
-            t = torch.tensor([mean * nb, float(nb)])
+            # tensor must live on the backend's device (RCCL = cuda)
+            t = torch.tensor([mean * nb, float(nb)], device=device)
             dist.all_reduce(t)
-            mean = float(t[0] / max(t[1], 1))
+            mean = float(t[0] / max(float(t[1]), 1.0))
         return {"eval_loss": mean}
 
     def save_model(self, output_dir: str):
