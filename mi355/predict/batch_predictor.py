@@ -34,6 +34,13 @@ class BatchPredictor:
         from .. import runtime
 
         n_workers = max_scoring_workers or min_scoring_workers
+        if (max_scoring_workers is None and num_gpus_per_worker
+                and runtime.is_initialized()):
+            # reference default: one replica per GPU (8 shards on a full node)
+            from ..runtime.core import _rt
+
+            n_workers = max(n_workers,
+                            int(_rt().num_gpus // max(num_gpus_per_worker, 1)))
 
         if runtime.is_initialized() and n_workers > 1:
             return self._predict_actor_pool(data, batch_size, n_workers,
