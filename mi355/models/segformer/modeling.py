@@ -245,6 +245,12 @@ class SegformerForSemanticSegmentation(nn.Module):
 
     def __init__(self, config: SegformerConfig):
         super().__init__()
+        # MIOpen immediate mode falls back to naive_conv_* kernels (double-
+        # accumulate, 77% of the train step measured) for several of these
+        # bf16 conv shapes; benchmark/find mode selects the CK implicit-GEMM
+        # solvers instead. Convs only exist on this model family.
+        if torch.cuda.is_available():
+            torch.backends.cudnn.benchmark = True
         self.config = config
         self.segformer = SegformerEncoder(config)
         self.decode_head = SegformerDecodeHead(config)
