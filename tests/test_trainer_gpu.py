@@ -55,3 +55,23 @@ def test_hf_trainer_single_gpu(tmp_path):
     assert res.error is None, res.error
     assert res.metrics["epoch"] == 2 and "eval_loss" in res.metrics
     assert res.checkpoint is not None
+
+
+@pytest.mark.gpu
+def test_bench_hybrid_two_graph_path():
+    """The multi-GPU step strategy (graphed fwd+bwd / eager all-reduce /
+    graphed optimizer) must work at world=1 via the validation knob."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    env = dict(os.environ, MI355_HYBRID_GRAPH="1")
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "3", "--warmup", "2"],
+        capture_output=True, text=True, timeout=300, env=env,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["config"]["graphs"] is True, out.stdout + out.stderr
+    assert d["value"] > 0
