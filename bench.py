@@ -126,6 +126,17 @@ def run_train(args, device, dtype):
                 print(f"# graph capture unavailable, eager fallback: {e!r}")
             graph = graph2 = None
 
+    if get_world_size() > 1 and device.type == "cuda":
+        # all-or-nothing: a rank whose capture failed must not run eager
+        # while others replay — the collective ORDER would diverge across
+        # ranks (hook-driven bucket flushes vs sequential explicit flushes)
+        import torch.distributed as dist
+
+        ok = torch.tensor([0.0 if graph is None else 1.0], device=device)
+        dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+        if ok.item() < 0.5:
+            graph = graph2 = None
+
     def timed_step():
         if graph is not None and graph2 is not None:
             graph.replay()
