@@ -67,6 +67,8 @@ def train_step(replica, optimizer, input_ids, labels):
     optimizer.zero_grad()
     out = replica(input_ids=input_ids, labels=labels)
     out.loss.backward()
+    replica.collect_grads()      # deferred mode: foreach-copy into flat bufs
+    replica.all_reduce_grads()   # deferred mode: explicit ordered flush
     replica.finish_grad_sync()
     optimizer.step()
     return out.loss
@@ -74,7 +76,10 @@ def train_step(replica, optimizer, input_ids, labels):
 
 def run_train(args, device, dtype):
     model = build_model(args, device, dtype)
-    replica = FlatReplica(model)
+    # deferred grads on GPU: the engine stores fresh grad tensors (no
+    # per-param accumulate add_, no flat-buffer zero-fill), batch-copied into
+    # the flat buffers after backward; explicit deterministic bucket flushes
+    replica = FlatReplica(model, deferred_grads=device.type == "cuda")
     optimizer = FlatAdamW(replica, lr=2e-5, weight_decay=0.01, max_grad_norm=1.0)
     input_ids, labels = synthetic_batch(args, device, seed=1000 + get_rank())
 
@@ -106,7 +111,8 @@ def run_train(args, device, dtype):
                 with torch.cuda.graph(g1):
                     optimizer.zero_grad()
                     out = replica(input_ids=input_ids, labels=labels)
-                    out.loss.backward()  # hooks skip collectives while capturing
+                    out.loss.backward()
+                    replica.collect_grads()
                 g2 = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(g2):
                     optimizer.step()
