@@ -42,7 +42,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
                                     c10::optional<torch::Tensor> bias,
                                     c10::optional<torch::Tensor> kpad,
                                     bool causal, bool need_dbias,
-                                    double p_drop, c10::optional<torch::Tensor> seed);
+                                    double p_drop, c10::optional<torch::Tensor> seed,
+                                    c10::optional<torch::Tensor> packed);
 torch::Tensor seg_postprocess(torch::Tensor logits, long Ho, long Wo);
 std::vector<torch::Tensor> attn_fwd_v2(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                        c10::optional<torch::Tensor> bias,
@@ -59,7 +60,8 @@ std::vector<torch::Tensor> attn_bwd_v3(torch::Tensor dout, torch::Tensor q, torc
                                        c10::optional<torch::Tensor> bias,
                                        c10::optional<torch::Tensor> kpad,
                                        bool causal, bool need_dbias,
-                                       double p_drop, c10::optional<torch::Tensor> seed);
+                                       double p_drop, c10::optional<torch::Tensor> seed,
+                                       c10::optional<torch::Tensor> packed);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
@@ -79,10 +81,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("token_embed_bwd", &token_embed_bwd);
   m.def("rel_bias_fwd", &rel_bias_fwd);
   m.def("rel_bias_bwd", &rel_bias_bwd);
-  m.def("attn_bwd", &attn_bwd);
+  m.def("attn_bwd", &attn_bwd, py::arg("dout"), py::arg("q"), py::arg("k"),
+        py::arg("v"), py::arg("o"), py::arg("lse"), py::arg("bias"),
+        py::arg("kpad"), py::arg("causal"), py::arg("need_dbias"),
+        py::arg("p_drop"), py::arg("seed"), py::arg("packed") = c10::nullopt);
   m.def("seg_postprocess", &seg_postprocess);
   m.def("attn_fwd_v2", &attn_fwd_v2);
   m.def("attn_bwd_v2", &attn_bwd_v2);
-  m.def("attn_bwd_v3", &attn_bwd_v3);
+  m.def("attn_bwd_v3", &attn_bwd_v3, py::arg("dout"), py::arg("q"), py::arg("k"),
+        py::arg("v"), py::arg("o"), py::arg("lse"), py::arg("bias"),
+        py::arg("kpad"), py::arg("causal"), py::arg("need_dbias"),
+        py::arg("p_drop"), py::arg("seed"), py::arg("packed") = c10::nullopt);
   m.attr("__gfx__") = "gfx950";
 }

@@ -545,7 +545,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
                                     c10::optional<torch::Tensor> bias,
                                     c10::optional<torch::Tensor> kpad,
                                     bool causal, bool need_dbias,
-                                    double p_drop, c10::optional<torch::Tensor> seed) {
+                                    double p_drop, c10::optional<torch::Tensor> seed,
+                                    c10::optional<torch::Tensor> packed) {
   TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1 &&
               dout.stride(3) == 1 && o.stride(3) == 1, "head dim must be contiguous");
   const int B = q.size(0), H = q.size(1), Sq = q.size(2), Sk = k.size(2), D = q.size(3);
@@ -553,6 +554,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
   auto opts = q.options();
   const long nkt = (Sk + 63) / 64;
   auto dq32 = torch::empty({nkt, B, H, Sq, D}, opts.dtype(torch::kFloat));
+  const bool has_packed = packed.has_value() && packed->defined();
   // q-split factor: fill the 256 CUs (~2 waves of blocks) when B*H*nkt alone
   // cannot; per-chunk dk/dv fp32 slabs are reduced by the shared epilogue
   const int ntq = (Sq + 63) / 64;
@@ -566,8 +568,10 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
   torch::Tensor dkv32;
   if (nqc > 1)
     dkv32 = torch::empty({2, (long)nqc, B, H, Sk, D}, opts.dtype(torch::kFloat));
-  auto dk = torch::empty({B, Sk, H, D}, opts).permute({0, 2, 1, 3});
-  auto dv = torch::empty({B, Sk, H, D}, opts).permute({0, 2, 1, 3});
+  auto dk = has_packed ? packed->select(2, 1).permute({0, 2, 1, 3})
+                       : torch::empty({B, Sk, H, D}, opts).permute({0, 2, 1, 3});
+  auto dv = has_packed ? packed->select(2, 2).permute({0, 2, 1, 3})
+                       : torch::empty({B, Sk, H, D}, opts).permute({0, 2, 1, 3});
   auto delta = torch::empty({B, H, Sq}, opts.dtype(torch::kFloat));
   torch::Tensor dbias32;
   if (need_dbias) {
@@ -608,7 +612,14 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
 
   // dq: sum the per-k-tile fp32 slabs -> low-precision BSHD-physical view;
   // dbias: batch-sum + cast. Single fused pass each (no aten sum/copy/cast).
-  auto dq = torch::empty({B, Sq, H, D}, opts).permute({0, 2, 1, 3});
+  // With `packed` ((B,Sq,3,H,D), self-attention) dq/dk/dv are views into the
+  // caller's buffer, so autograd sees ONE fused qkv grad and never cats.
+  torch::Tensor dq;
+  if (packed.has_value() && packed->defined()) {
+    TORCH_CHECK(Sq == Sk, "packed qkv grads require self-attention");
+    dq = packed->select(2, 0).permute({0, 2, 1, 3});
+  } else
+    dq = torch::empty({B, Sq, H, D}, opts).permute({0, 2, 1, 3});
   const long plane = need_dbias ? dbias32.numel() / B : 0;
   const bool fused_ok = ((long)B * H * Sq * D) % 4 == 0 && plane % 4 == 0;
   torch::Tensor dbias;

@@ -552,15 +552,22 @@ std::vector<torch::Tensor> attn_bwd_v3(torch::Tensor dout, torch::Tensor q, torc
                                        c10::optional<torch::Tensor> bias,
                                        c10::optional<torch::Tensor> kpad,
                                        bool causal, bool need_dbias,
-                                       double p_drop, c10::optional<torch::Tensor> seed) {
+                                       double p_drop, c10::optional<torch::Tensor> seed,
+                                       c10::optional<torch::Tensor> packed) {
   TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1 &&
               dout.stride(3) == 1 && o.stride(3) == 1 && q.size(3) == 64,
               "attn_bwd_v3: D=64, d contiguous");
   const int B = q.size(0), H = q.size(1), Sq = q.size(2), Sk = k.size(2);
   auto opts = q.options();
-  auto dq = torch::empty({B, Sq, H, 64}, opts).permute({0, 2, 1, 3});
-  auto dk = torch::empty({B, Sk, H, 64}, opts).permute({0, 2, 1, 3});
-  auto dv = torch::empty({B, Sk, H, 64}, opts).permute({0, 2, 1, 3});
+  // packed ((B,Sq,3,H,64)): outputs are views into one fused qkv grad buffer
+  const bool has_packed = packed.has_value() && packed->defined();
+  if (has_packed) TORCH_CHECK(Sq == Sk, "packed qkv grads require self-attention");
+  auto dq = has_packed ? packed->select(2, 0).permute({0, 2, 1, 3})
+                       : torch::empty({B, Sq, H, 64}, opts).permute({0, 2, 1, 3});
+  auto dk = has_packed ? packed->select(2, 1).permute({0, 2, 1, 3})
+                       : torch::empty({B, Sk, H, 64}, opts).permute({0, 2, 1, 3});
+  auto dv = has_packed ? packed->select(2, 2).permute({0, 2, 1, 3})
+                       : torch::empty({B, Sk, H, 64}, opts).permute({0, 2, 1, 3});
   auto delta = torch::empty({B, H, Sq}, opts.dtype(torch::kFloat));
   torch::Tensor dbias32;
   if (need_dbias) {

@@ -162,8 +162,20 @@ class T5Attention(nn.Module):
         or {"k","v" precomputed static} for cross-attn."""
         b, sq, _ = hidden.shape
         static = kv_cache is not None and kv_cache.get("static", False)
+        if kv_hidden is None and kv_cache is None:
+            # self-attention without a cache (training / encoder): ONE fused
+            # qkv GEMM consumed PACKED — the attention grad comes back as one
+            # (B,S,3,H,D) tensor, so autograd never cats dq/dk/dv
+            qkv5 = F.linear(hidden, self.qkv.weight).view(
+                b, sq, 3, self.n_heads, self.d_kv)
+            p_drop = self.dropout if self.training else 0.0
+            out = ops.attention_qkv_packed(qkv5, bias=position_bias,
+                                           causal=causal, kpad=kpad,
+                                           p_drop=p_drop)
+            out = out.transpose(1, 2).reshape(b, sq, self.inner_dim)
+            return self.o(out)
         if kv_hidden is None and not static:
-            # self-attention (train or decode): ONE fused qkv GEMM
+            # self-attention with a decode cache: fused qkv GEMM, split views
             q3, k3, v3 = self.qkv(hidden)
             qs = self._shape(q3, b)
             ks = self._shape(k3, b)

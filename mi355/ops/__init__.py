@@ -221,6 +221,66 @@ def bump_dropout_seed(device):
         _dropout_seed(device).add_(1)
 
 
+def _hip_attn_fwd(q, k, v, bias_c, kpad_c, causal, p_drop, seed):
+    ext = _load_ext()
+    if (q.shape[-1] == 64 and hasattr(ext, "attn_fwd_v2")
+            and os.environ.get("MI355_ATTN_V2", "1") != "0"):
+        return ext.attn_fwd_v2(q, k, v, bias_c, kpad_c, causal, p_drop, seed)
+    return ext.attn_fwd(q, k, v, bias_c, kpad_c, causal, p_drop, seed)
+
+
+def _hip_attn_bwd(do, q, k, v, o, lse, bias, kpad, causal, need_dbias, p_drop,
+                  packed=None):
+    """Dispatch over the three bwd generations: v3 (FA2-style two plain-store
+    kernels on the swapped 32x32 layout — fastest once its 128-wide blocks
+    fill the chip), v2 (monolithic swapped layout; slab traffic), v1 (16x16
+    recompute with a q-split — finer blocks win when the grid underfills the
+    256 CUs, e.g. the B=2 headline config). `packed` ((B,S,3,H,D)) makes
+    dq/dk/dv views into one fused qkv grad buffer."""
+    ext = _load_ext()
+    use_hip_bwd = hasattr(ext, "attn_bwd") and q.shape[-1] in (32, 64) and \
+        (not need_dbias or (bias is not None and bias.shape[0] == 1)) \
+        and os.environ.get("MI355_ATTN_BWD_REF") != "1"
+    if p_drop > 0 and not use_hip_bwd:
+        raise RuntimeError("attention-prob dropout bwd requires the HIP kernel")
+    if use_hip_bwd:
+        if do.stride(-1) != 1:
+            do = do.contiguous()
+        seed = _dropout_seed(q.device) if p_drop > 0 else None
+        if q.shape[-1] == 64:
+            nblk = q.shape[0] * q.shape[1] * ((max(q.shape[2], k.shape[2]) + 127) // 128)
+            default = "v3" if nblk >= 256 else "v1"
+        else:
+            default = "v1"
+        bwd_impl = os.environ.get("MI355_ATTN_BWD", default)
+        if packed is not None and bwd_impl == "v2":
+            bwd_impl = default  # v2 (gated off) has no packed-out support
+        if bwd_impl == "v3" and q.shape[-1] == 64 and hasattr(ext, "attn_bwd_v3"):
+            res = ext.attn_bwd_v3(do, q, k, v, o, lse, bias, kpad, causal,
+                                  need_dbias, p_drop, seed, packed)
+        elif bwd_impl == "v2" and q.shape[-1] == 64 and hasattr(ext, "attn_bwd_v2"):
+            res = ext.attn_bwd_v2(do, q, k, v, o, lse, bias, kpad, causal,
+                                  need_dbias, p_drop, seed)
+        else:
+            res = ext.attn_bwd(do, q, k, v, o, lse, bias, kpad, causal,
+                               need_dbias, p_drop, seed, packed)
+        dq, dk, dv = res[0], res[1], res[2]
+        dbias = res[3] if len(res) > 3 else None
+    else:  # composed bwd via recompute (hipBLASLt batched GEMMs)
+        dq, dk, dv, dbias = _attn_bwd_composed(do, q, k, v, lse, bias, kpad,
+                                               causal, need_dbias)
+        if packed is not None:
+            packed.select(2, 0).copy_(dq.transpose(1, 2))
+            packed.select(2, 1).copy_(dk.transpose(1, 2))
+            packed.select(2, 2).copy_(dv.transpose(1, 2))
+    if dbias is not None and bias is not None and dbias.shape != bias.shape:
+        # reduce over broadcast dims (bias is usually (1,H,Sq,Sk))
+        for d in range(dbias.dim()):
+            if bias.shape[d] == 1 and dbias.shape[d] != 1:
+                dbias = dbias.sum(dim=d, keepdim=True)
+    return dq, dk, dv, dbias
+
+
 class _Attention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, bias, causal, kpad, p_drop):
@@ -232,12 +292,7 @@ class _Attention(torch.autograd.Function):
         bias_c = bias.contiguous() if bias is not None else None
         kpad_c = kpad.contiguous().float() if kpad is not None else None
         seed = _dropout_seed(q.device) if p_drop > 0 else None
-        ext = _load_ext()
-        if (q.shape[-1] == 64 and hasattr(ext, "attn_fwd_v2")
-                and os.environ.get("MI355_ATTN_V2", "1") != "0"):
-            o, lse = ext.attn_fwd_v2(q, k, v, bias_c, kpad_c, causal, p_drop, seed)
-        else:
-            o, lse = ext.attn_fwd(q, k, v, bias_c, kpad_c, causal, p_drop, seed)
+        o, lse = _hip_attn_fwd(q, k, v, bias_c, kpad_c, causal, p_drop, seed)
         ctx.save_for_backward(q, k, v, bias_c, kpad_c, o, lse)
         ctx.causal = causal
         ctx.p_drop = p_drop
@@ -252,48 +307,46 @@ class _Attention(torch.autograd.Function):
         # non-unit head-dim stride forces a copy
         if do.stride(-1) != 1:
             do = do.contiguous()
-        ext = _load_ext()
-        use_hip_bwd = hasattr(ext, "attn_bwd") and q.shape[-1] in (32, 64) and \
-            (not ctx.bias_requires_grad or (bias is not None and bias.shape[0] == 1)) \
-            and os.environ.get("MI355_ATTN_BWD_REF") != "1"
-        if ctx.p_drop > 0 and not use_hip_bwd:
-            raise RuntimeError("attention-prob dropout bwd requires the HIP kernel")
-        if use_hip_bwd:
-            if do.stride(-1) != 1:
-                do = do.contiguous()
-            seed = _dropout_seed(q.device) if ctx.p_drop > 0 else None
-            # three bwd generations in-tree: v3 (FA2-style two plain-store
-            # kernels on the swapped 32x32 layout — fastest once its 128-wide
-            # blocks fill the chip), v2 (monolithic swapped layout; slab
-            # traffic), v1 (16x16 recompute — finer 64-wide blocks win when
-            # the grid underfills the 256 CUs, e.g. the B=2 headline config)
-            if q.shape[-1] == 64:
-                nblk = q.shape[0] * q.shape[1] * ((max(q.shape[2], k.shape[2]) + 127) // 128)
-                default = "v3" if nblk >= 256 else "v1"
-            else:
-                default = "v1"
-            bwd_impl = os.environ.get("MI355_ATTN_BWD", default)
-            if bwd_impl == "v3" and q.shape[-1] == 64 and hasattr(ext, "attn_bwd_v3"):
-                res = ext.attn_bwd_v3(do, q, k, v, o, lse, bias, kpad, ctx.causal,
-                                      ctx.bias_requires_grad, ctx.p_drop, seed)
-            elif bwd_impl == "v2" and q.shape[-1] == 64 and hasattr(ext, "attn_bwd_v2"):
-                res = ext.attn_bwd_v2(do, q, k, v, o, lse, bias, kpad, ctx.causal,
-                                      ctx.bias_requires_grad, ctx.p_drop, seed)
-            else:
-                res = ext.attn_bwd(do, q, k, v, o, lse, bias, kpad, ctx.causal,
-                                   ctx.bias_requires_grad, ctx.p_drop, seed)
-            dq, dk, dv = res[0], res[1], res[2]
-            dbias = res[3] if len(res) > 3 else None
-        else:  # composed bwd via recompute (hipBLASLt batched GEMMs)
-            dq, dk, dv, dbias = _attn_bwd_composed(
-                do, q, k, v, lse, bias, kpad, ctx.causal, ctx.bias_requires_grad
-            )
-        if dbias is not None and bias is not None and dbias.shape != bias.shape:
-            # reduce over broadcast dims (bias is usually (1,H,Sq,Sk))
-            for d in range(dbias.dim()):
-                if bias.shape[d] == 1 and dbias.shape[d] != 1:
-                    dbias = dbias.sum(dim=d, keepdim=True)
+        dq, dk, dv, dbias = _hip_attn_bwd(do, q, k, v, o, lse, bias, kpad,
+                                          ctx.causal, ctx.bias_requires_grad,
+                                          ctx.p_drop)
         return dq, dk, dv, dbias, None, None, None
+
+
+class _AttentionQKVPacked(torch.autograd.Function):
+    """Self-attention straight off the fused qkv projection: input is the
+    (B, S, 3, H, D) view of one GEMM output, and backward writes ONE packed
+    (B, S, 3, H, D) grad (dq/dk/dv are strided views into it) so autograd
+    reshapes it for free instead of cat-ing three grads per layer."""
+
+    @staticmethod
+    def forward(ctx, qkv, bias, causal, kpad, p_drop):
+        q = qkv.select(2, 0).permute(0, 2, 1, 3)
+        k = qkv.select(2, 1).permute(0, 2, 1, 3)
+        v = qkv.select(2, 2).permute(0, 2, 1, 3)
+        bias_c = bias.contiguous() if bias is not None else None
+        kpad_c = kpad.contiguous().float() if kpad is not None else None
+        seed = _dropout_seed(qkv.device) if p_drop > 0 else None
+        o, lse = _hip_attn_fwd(q, k, v, bias_c, kpad_c, causal, p_drop, seed)
+        ctx.save_for_backward(qkv, bias_c, kpad_c, o, lse)
+        ctx.causal = causal
+        ctx.p_drop = p_drop
+        ctx.bias_requires_grad = bias is not None and bias.requires_grad
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        qkv, bias, kpad, o, lse = ctx.saved_tensors
+        q = qkv.select(2, 0).permute(0, 2, 1, 3)
+        k = qkv.select(2, 1).permute(0, 2, 1, 3)
+        v = qkv.select(2, 2).permute(0, 2, 1, 3)
+        if do.stride(-1) != 1:
+            do = do.contiguous()
+        dqkv = torch.empty_like(qkv)
+        _, _, _, dbias = _hip_attn_bwd(do, q, k, v, o, lse, bias, kpad,
+                                       ctx.causal, ctx.bias_requires_grad,
+                                       ctx.p_drop, packed=dqkv)
+        return dqkv, dbias, None, None, None
 
 
 def _attn_bwd_composed(do, q, k, v, lse, bias, kpad, causal, need_dbias):
@@ -332,6 +385,21 @@ def attention(q, k, v, bias=None, causal=False, scale: float = 1.0, kpad=None,
         return _Attention.apply(q, k, v, bias, causal, kpad, float(p_drop))
     return torch_ref.attention(q, k, v, bias=bias, causal=causal, scale=scale,
                                kpad=kpad, p_drop=p_drop)
+
+
+def attention_qkv_packed(qkv, bias=None, causal=False, kpad=None,
+                         p_drop: float = 0.0):
+    """Self-attention over the fused-projection output: ``qkv`` is the
+    (B, S, 3, H, D) view of one GEMM result. The backward produces ONE packed
+    grad, so the wi-style split-cat never appears in autograd."""
+    hip_ok = (qkv.dtype in (torch.bfloat16, torch.float16)
+              and qkv.shape[-1] in (32, 64))
+    if hip_ok and _use_hip(qkv):
+        return _AttentionQKVPacked.apply(qkv, bias, causal, kpad, float(p_drop))
+    q = qkv.select(2, 0).permute(0, 2, 1, 3)
+    k = qkv.select(2, 1).permute(0, 2, 1, 3)
+    v = qkv.select(2, 2).permute(0, 2, 1, 3)
+    return attention(q, k, v, bias=bias, causal=causal, kpad=kpad, p_drop=p_drop)
 
 
 # --------------------------------------------------------------------------- #

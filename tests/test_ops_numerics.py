@@ -352,3 +352,31 @@ def test_sumsq_accumulate():
     ops.sumsq_accumulate(b, out)
     ref = a.float().pow(2).sum() + b.pow(2).sum()
     assert _relerr(out, ref) < 1e-3
+
+
+@gpu
+@pytest.mark.parametrize("causal", [False, True])
+def test_attention_qkv_packed_matches_split(causal):
+    """Packed self-attention (one (B,S,3,H,D) input/grad) must equal the
+    split q/k/v path, including the fused qkv grad vs the cat of dq/dk/dv."""
+    torch.manual_seed(11)
+    B, S, H, D = 2, 192, 4, 64
+    qkv = torch.randn(B, S, 3, H, D, device="cuda", dtype=torch.bfloat16,
+                      requires_grad=True)
+    bias = (torch.randn(1, H, S, S, device="cuda") * 0.5).to(torch.bfloat16)
+    bias.requires_grad_(True)
+    out_p = ops.attention_qkv_packed(qkv, bias=bias, causal=causal)
+
+    qkv2 = qkv.detach().clone().requires_grad_(True)
+    bias2 = bias.detach().clone().requires_grad_(True)
+    q = qkv2.select(2, 0).permute(0, 2, 1, 3)
+    k = qkv2.select(2, 1).permute(0, 2, 1, 3)
+    v = qkv2.select(2, 2).permute(0, 2, 1, 3)
+    out_s = ops.attention(q, k, v, bias=bias2, causal=causal)
+    assert torch.equal(out_p, out_s)
+
+    d = torch.randn_like(out_p)
+    out_p.backward(d)
+    out_s.backward(d)
+    assert torch.equal(qkv.grad, qkv2.grad)
+    assert torch.equal(bias.grad, bias2.grad)
