@@ -114,7 +114,8 @@ def test_submit_nonblocking_backlog():
         t0 = _t.monotonic()
         refs = [slow.remote(i) for i in range(8)]
         submit_time = _t.monotonic() - t0
-        assert submit_time < 1.0, f"submit blocked: {submit_time:.2f}s"
+        # generous bound: CI boxes can stall the spawn machinery
+        assert submit_time < 3.0, f"submit blocked: {submit_time:.2f}s"
         assert mi355.get(refs) == [i * i for i in range(8)]
     finally:
         mi355.shutdown()
