@@ -91,8 +91,13 @@ def run_train(args, device, dtype):
     #    the ~700 per-step launches that made the eager path ~60% slower at
     #    per-GPU batch 2 are gone. MI355_DIST_GRAPHS=0 forces eager.
     use_graphs = device.type == "cuda" and args.graphs != "off"
-    hybrid = use_graphs and get_world_size() > 1 \
-        and os.environ.get("MI355_DIST_GRAPHS", "1") != "0"
+    dist_mode = os.environ.get("MI355_DIST_GRAPHS", "1")
+    hybrid = use_graphs and get_world_size() > 1 and dist_mode == "1"
+    # MI355_DIST_GRAPHS=full: capture the RCCL collectives INSIDE one full
+    # graph (train_step incl. all_reduce). Experimental — untested on the
+    # 8-GPU path, so never the default; a capture failure on any rank falls
+    # back to eager on every rank via the consensus below.
+    full_dist_capture = use_graphs and get_world_size() > 1 and dist_mode == "full"
     if device.type == "cuda" and os.environ.get("MI355_HYBRID_GRAPH") == "1":
         hybrid = use_graphs = True  # 1-GPU validation knob for the hybrid path
 
@@ -106,7 +111,12 @@ def run_train(args, device, dtype):
         try:
             barrier()
             torch.cuda.synchronize()
-            if hybrid:
+            if full_dist_capture:
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    step_fn()
+                graph = g
+            elif hybrid:
                 g1 = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(g1):
                     optimizer.zero_grad()
