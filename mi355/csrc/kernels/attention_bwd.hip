@@ -568,9 +568,10 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
   torch::Tensor dkv32;
   if (nqc > 1)
     dkv32 = torch::empty({2, (long)nqc, B, H, Sk, D}, opts.dtype(torch::kFloat));
-  auto dk = has_packed ? packed->select(2, 1).permute({0, 2, 1, 3})
+  const int kslot = has_packed ? packed->size(2) - 2 : 0;  // 1 for qkv, 0 for kv
+  auto dk = has_packed ? packed->select(2, kslot).permute({0, 2, 1, 3})
                        : torch::empty({B, Sk, H, D}, opts).permute({0, 2, 1, 3});
-  auto dv = has_packed ? packed->select(2, 2).permute({0, 2, 1, 3})
+  auto dv = has_packed ? packed->select(2, kslot + 1).permute({0, 2, 1, 3})
                        : torch::empty({B, Sk, H, D}, opts).permute({0, 2, 1, 3});
   auto delta = torch::empty({B, H, Sq}, opts.dtype(torch::kFloat));
   torch::Tensor dbias32;
@@ -614,8 +615,10 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
   // dbias: batch-sum + cast. Single fused pass each (no aten sum/copy/cast).
   // With `packed` ((B,Sq,3,H,D), self-attention) dq/dk/dv are views into the
   // caller's buffer, so autograd sees ONE fused qkv grad and never cats.
+  // packed slots: 3 = fused qkv grad (self-attention, Sq==Sk), 2 = fused
+  // kv grad (cross-attention: dq is its own tensor)
   torch::Tensor dq;
-  if (packed.has_value() && packed->defined()) {
+  if (has_packed && packed->size(2) == 3) {
     TORCH_CHECK(Sq == Sk, "packed qkv grads require self-attention");
     dq = packed->select(2, 0).permute({0, 2, 1, 3});
   } else

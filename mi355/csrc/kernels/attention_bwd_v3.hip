@@ -559,14 +559,17 @@ std::vector<torch::Tensor> attn_bwd_v3(torch::Tensor dout, torch::Tensor q, torc
               "attn_bwd_v3: D=64, d contiguous");
   const int B = q.size(0), H = q.size(1), Sq = q.size(2), Sk = k.size(2);
   auto opts = q.options();
-  // packed ((B,Sq,3,H,64)): outputs are views into one fused qkv grad buffer
+  // packed: (B,Sq,3,H,64) = fused qkv grad (self-attn) or (B,Sk,2,H,64) =
+  // fused kv grad (cross-attn; dq separate)
   const bool has_packed = packed.has_value() && packed->defined();
-  if (has_packed) TORCH_CHECK(Sq == Sk, "packed qkv grads require self-attention");
-  auto dq = has_packed ? packed->select(2, 0).permute({0, 2, 1, 3})
-                       : torch::empty({B, Sq, H, 64}, opts).permute({0, 2, 1, 3});
-  auto dk = has_packed ? packed->select(2, 1).permute({0, 2, 1, 3})
+  const bool pq = has_packed && packed->size(2) == 3;
+  if (pq) TORCH_CHECK(Sq == Sk, "packed qkv grads require self-attention");
+  const int kslot = has_packed ? packed->size(2) - 2 : 0;
+  auto dq = pq ? packed->select(2, 0).permute({0, 2, 1, 3})
+               : torch::empty({B, Sq, H, 64}, opts).permute({0, 2, 1, 3});
+  auto dk = has_packed ? packed->select(2, kslot).permute({0, 2, 1, 3})
                        : torch::empty({B, Sk, H, 64}, opts).permute({0, 2, 1, 3});
-  auto dv = has_packed ? packed->select(2, 2).permute({0, 2, 1, 3})
+  auto dv = has_packed ? packed->select(2, kslot + 1).permute({0, 2, 1, 3})
                        : torch::empty({B, Sk, H, 64}, opts).permute({0, 2, 1, 3});
   auto delta = torch::empty({B, H, Sq}, opts.dtype(torch::kFloat));
   torch::Tensor dbias32;

@@ -191,9 +191,17 @@ class T5Attention(nn.Module):
             qs = self._shape(self.q(hidden), b)
             ks, vs = kv_cache["k"], kv_cache["v"]
         else:
+            # cross-attention (training): ONE fused kv GEMM over the last two
+            # thirds of the qkv weight, consumed packed (one dkv grad, and a
+            # single contiguous weight-slice grad instead of two)
             qs = self._shape(self.q(hidden), b)
-            ks = self._shape(self.k(kv_hidden), b)
-            vs = self._shape(self.v(kv_hidden), b)
+            kv5 = F.linear(kv_hidden, self.qkv.weight[self.inner_dim:]).view(
+                b, -1, 2, self.n_heads, self.d_kv)
+            p_drop = self.dropout if self.training else 0.0
+            out = ops.attention_kv_packed(qs, kv5, bias=position_bias,
+                                          kpad=kpad, p_drop=p_drop)
+            out = out.transpose(1, 2).reshape(b, sq, self.inner_dim)
+            return self.o(out)
         p_drop = self.dropout if self.training else 0.0
         if kv_cache is not None and sq == 1 and not static:
             out = ops.decode_attention(qs, kv_cache["k"], kv_cache["v"],

@@ -387,6 +387,51 @@ def attention(q, k, v, bias=None, causal=False, scale: float = 1.0, kpad=None,
                                kpad=kpad, p_drop=p_drop)
 
 
+class _AttentionKVPacked(torch.autograd.Function):
+    """Cross-attention with the fused kv projection consumed packed: kv is
+    the (B, Sk, 2, H, D) view of one GEMM output; backward writes one packed
+    (B, Sk, 2, H, D) grad (dk/dv are views into it) plus a separate dq."""
+
+    @staticmethod
+    def forward(ctx, q, kv, bias, kpad, p_drop):
+        if q.stride(-1) != 1:
+            q = q.contiguous()
+        k = kv.select(2, 0).permute(0, 2, 1, 3)
+        v = kv.select(2, 1).permute(0, 2, 1, 3)
+        bias_c = bias.contiguous() if bias is not None else None
+        kpad_c = kpad.contiguous().float() if kpad is not None else None
+        seed = _dropout_seed(q.device) if p_drop > 0 else None
+        o, lse = _hip_attn_fwd(q, k, v, bias_c, kpad_c, False, p_drop, seed)
+        ctx.save_for_backward(q, kv, bias_c, kpad_c, o, lse)
+        ctx.p_drop = p_drop
+        ctx.bias_requires_grad = bias is not None and bias.requires_grad
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, kv, bias, kpad, o, lse = ctx.saved_tensors
+        k = kv.select(2, 0).permute(0, 2, 1, 3)
+        v = kv.select(2, 1).permute(0, 2, 1, 3)
+        if do.stride(-1) != 1:
+            do = do.contiguous()
+        dkv = torch.empty_like(kv)
+        dq, _, _, dbias = _hip_attn_bwd(do, q, k, v, o, lse, bias, kpad,
+                                        False, ctx.bias_requires_grad,
+                                        ctx.p_drop, packed=dkv)
+        return dq, dkv, dbias, None, None
+
+
+def attention_kv_packed(q, kv, bias=None, kpad=None, p_drop: float = 0.0):
+    """Cross-attention over a packed (B, Sk, 2, H, D) fused-kv projection."""
+    hip_ok = (kv.dtype in (torch.bfloat16, torch.float16)
+              and kv.shape[-1] in (32, 64))
+    if hip_ok and _use_hip(kv):
+        return _AttentionKVPacked.apply(q, kv, bias, kpad, float(p_drop))
+    k = kv.select(2, 0).permute(0, 2, 1, 3)
+    v = kv.select(2, 1).permute(0, 2, 1, 3)
+    return attention(q, k, v, bias=bias, causal=False, kpad=kpad, p_drop=p_drop)
+
+
 def attention_qkv_packed(qkv, bias=None, causal=False, kpad=None,
                          p_drop: float = 0.0):
     """Self-attention over the fused-projection output: ``qkv`` is the

@@ -380,3 +380,26 @@ def test_attention_qkv_packed_matches_split(causal):
     out_s.backward(d)
     assert torch.equal(qkv.grad, qkv2.grad)
     assert torch.equal(bias.grad, bias2.grad)
+
+
+@gpu
+def test_attention_kv_packed_matches_split():
+    torch.manual_seed(12)
+    B, Sq, Sk, H, D = 2, 96, 160, 4, 64
+    q = torch.randn(B, H, Sq, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    kv = torch.randn(B, Sk, 2, H, D, device="cuda", dtype=torch.bfloat16,
+                     requires_grad=True)
+    out_p = ops.attention_kv_packed(q, kv)
+
+    q2 = q.detach().clone().requires_grad_(True)
+    kv2 = kv.detach().clone().requires_grad_(True)
+    k = kv2.select(2, 0).permute(0, 2, 1, 3)
+    v = kv2.select(2, 1).permute(0, 2, 1, 3)
+    out_s = ops.attention(q2, k, v)
+    assert torch.equal(out_p, out_s)
+    d = torch.randn_like(out_p)
+    out_p.backward(d)
+    out_s.backward(d)
+    assert torch.equal(q.grad, q2.grad)
+    assert torch.equal(kv.grad, kv2.grad)
