@@ -19,7 +19,7 @@ def main():
     m.eval()
     pix = torch.randn(16, 3, 512, 512, device="cuda", dtype=torch.bfloat16)
     with torch.no_grad():
-        for _ in range(3):
+        for _ in range(5):
             m(pixel_values=pix)
         torch.cuda.synchronize()
         t0 = time.perf_counter()
@@ -44,7 +44,7 @@ def main():
         out = m(pixel_values=pix8, labels=labels)
         opt.zero_grad(); out.loss.backward(); opt.step()
         return out.loss
-    for _ in range(3):
+    for _ in range(6):  # MIOpen find mode autotunes on first calls
         step()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
