@@ -32,10 +32,14 @@ FLAGS = [
          "FlatReplica gradient all-reduce bucket size in MiB. xGMI is 7 "
          "point-to-point links x ~153 GB/s, so fewer/larger flushes beat "
          "NVSwitch-tuned 25 MB buckets."),
+    Flag("MI355_EXPLICIT_FLUSH", "whole",
+         "Explicit (non-hook) gradient flush granularity: whole = one "
+         "collective per flat group, buckets = MI355_BUCKET_MB slices."),
     Flag("MI355_DIST_GRAPHS", "1",
          "bench.py multi-GPU path: 1 = two-graph hybrid step (hipGraph "
-         "fwd+bwd and optimizer around an eager bucketed RCCL all-reduce), "
-         "0 = fully eager."),
+         "fwd+bwd and optimizer around an eager RCCL all-reduce), 0 = fully "
+         "eager, full = capture the collectives inside one graph "
+         "(experimental)."),
     Flag("MI355_HYBRID_GRAPH", "0",
          "Force the two-graph hybrid step at world_size 1 (validation knob)."),
     # -- kernels -------------------------------------------------------------

@@ -175,14 +175,29 @@ class FlatReplica:
                 torch._foreach_copy_(slices, grads)
 
     def all_reduce_grads(self):
-        """Explicit bucketed all-reduce of the flat grad buffers. Used by the
+        """Explicit all-reduce of the flat grad buffers. Used by the
         two-graph hybrid step (backward replayed from a hipGraph, so the
         post-accumulate hooks never fire): g1.replay() -> all_reduce_grads()
         -> finish_grad_sync() -> g2.replay(). The flat buffers are static
-        addresses, so this is safe across graph replays."""
+        addresses, so this is safe across graph replays.
+
+        Unlike the hook path there is no backward left to overlap with, so
+        the default is ONE whole-buffer collective per group (RCCL pipelines
+        a large ring message better than 8 sequential 64 MB calls);
+        MI355_EXPLICIT_FLUSH=buckets restores bucketed flushes."""
         if not is_distributed():
             return
+        whole = os.environ.get("MI355_EXPLICIT_FLUSH", "whole") != "buckets"
         for gi, (_, g) in enumerate(self.groups):
+            if whole:
+                if not any((gi, bi) in self._flushed
+                           for bi in range(len(g.buckets))):
+                    work = dist.all_reduce(g.flat_grad, op=dist.ReduceOp.SUM,
+                                           group=self.pg, async_op=True)
+                    self._works.append(work)
+                    self._flushed.update((gi, bi)
+                                         for bi in range(len(g.buckets)))
+                continue
             for bi, (bs, be, _last) in enumerate(g.buckets):
                 if (gi, bi) in self._flushed:
                     continue
