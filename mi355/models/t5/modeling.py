@@ -166,14 +166,14 @@ class T5Attention(nn.Module):
             # self-attention without a cache (training / encoder): ONE fused
             # qkv GEMM consumed PACKED — the attention grad comes back as one
             # (B,S,3,H,D) tensor, so autograd never cats dq/dk/dv
-            qkv5 = F.linear(hidden, self.qkv.weight).view(
+            qkv5 = ops.linear(hidden, self.qkv.weight).view(
                 b, sq, 3, self.n_heads, self.d_kv)
             p_drop = self.dropout if self.training else 0.0
             out = ops.attention_qkv_packed(qkv5, bias=position_bias,
                                            causal=causal, kpad=kpad,
                                            p_drop=p_drop)
             out = out.transpose(1, 2).reshape(b, sq, self.inner_dim)
-            return self.o(out)
+            return ops.linear(out, self.o.weight)
         if kv_hidden is None and not static:
             # self-attention with a decode cache: fused qkv GEMM, split views
             q3, k3, v3 = self.qkv(hidden)
@@ -195,13 +195,13 @@ class T5Attention(nn.Module):
             # thirds of the qkv weight, consumed packed (one dkv grad, and a
             # single contiguous weight-slice grad instead of two)
             qs = self._shape(self.q(hidden), b)
-            kv5 = F.linear(kv_hidden, self.qkv.weight[self.inner_dim:]).view(
+            kv5 = ops.linear(kv_hidden, self.qkv.weight[self.inner_dim:]).view(
                 b, -1, 2, self.n_heads, self.d_kv)
             p_drop = self.dropout if self.training else 0.0
             out = ops.attention_kv_packed(qs, kv5, bias=position_bias,
                                           kpad=kpad, p_drop=p_drop)
             out = out.transpose(1, 2).reshape(b, sq, self.inner_dim)
-            return self.o(out)
+            return ops.linear(out, self.o.weight)
         p_drop = self.dropout if self.training else 0.0
         if kv_cache is not None and sq == 1 and not static:
             out = ops.decode_attention(qs, kv_cache["k"], kv_cache["v"],
@@ -214,7 +214,7 @@ class T5Attention(nn.Module):
             out = ops.attention(qs, ks, vs, bias=position_bias, causal=causal,
                                 kpad=kpad, p_drop=p_drop)
         out = out.transpose(1, 2).reshape(b, sq, self.inner_dim)
-        return self.o(out)
+        return ops.linear(out, self.o.weight)
 
 
 class T5LayerSelfAttention(nn.Module):
@@ -256,9 +256,9 @@ class T5DenseGatedActDense(nn.Module):
         # packed: one fused-wi GEMM -> one GeGLU kernel with the FF dropout
         # fused on its output (no gate/up split views, no grad cat)
         p = self.dropout.p if self.training else 0.0
-        h = ops.geglu_packed(F.linear(x, self.wi.weight), p_drop=p,
+        h = ops.geglu_packed(ops.linear(x, self.wi.weight), p_drop=p,
                              salt=self.wi.weight.data_ptr())
-        return self.wo(h)
+        return ops.linear(h, self.wo.weight)
 
 
 class T5LayerFF(nn.Module):

@@ -135,6 +135,59 @@ def add_rms_norm(x, r, weight, eps: float = 1e-6, p_drop: float = 0.0):
 
 
 # --------------------------------------------------------------------------- #
+# Two-stream linear: overlap the dW and dX backward GEMMs                     #
+# --------------------------------------------------------------------------- #
+_gemm_side_streams: dict = {}
+
+
+def _gemm_side_stream(device):
+    key = str(device)
+    st = _gemm_side_streams.get(key)
+    if st is None:
+        st = _gemm_side_streams[key] = torch.cuda.Stream(device)
+    return st
+
+
+class _Linear2S(torch.autograd.Function):
+    """Bias-free linear whose backward computes dW on a side stream while dX
+    runs on the main stream (fork/join events — hipGraph capture turns them
+    into graph dependencies). At per-GPU batch 2 each of the two backward
+    GEMMs underfills the 256 CUs on its own (~190 blocks); co-scheduling
+    them fills the chip."""
+
+    @staticmethod
+    def forward(ctx, x, w):
+        ctx.save_for_backward(x, w)
+        return torch.nn.functional.linear(x, w)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy2 = dy.reshape(-1, dy.shape[-1])
+        x2 = x.reshape(-1, x.shape[-1])
+        main = torch.cuda.current_stream()
+        side = _gemm_side_stream(dy.device)
+        fork = torch.cuda.Event()
+        fork.record(main)
+        side.wait_event(fork)
+        with torch.cuda.stream(side):
+            dw = dy2.t().mm(x2)
+        join = torch.cuda.Event()
+        join.record(side)
+        dx = dy2.mm(w).view(x.shape)
+        main.wait_event(join)
+        dw.record_stream(main)
+        return dx, dw
+
+
+def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """F.linear (no bias) with the two-stream backward on GPU."""
+    if x.is_cuda and os.environ.get("MI355_GEMM_2S", "1") != "0":
+        return _Linear2S.apply(x, w)
+    return torch.nn.functional.linear(x, w)
+
+
+# --------------------------------------------------------------------------- #
 # Gated-GELU elementwise combine                                              #
 # --------------------------------------------------------------------------- #
 class _GeGLU(torch.autograd.Function):
