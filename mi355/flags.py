@@ -43,6 +43,9 @@ FLAGS = [
     Flag("MI355_HYBRID_GRAPH", "0",
          "Force the two-graph hybrid step at world_size 1 (validation knob)."),
     # -- kernels -------------------------------------------------------------
+    Flag("MI355_GEMM_2S", "0",
+         "1 = two-stream linear backward (dW on a side stream). Off by "
+         "default: measured slower at B=2 (event overhead)."),
     Flag("MI355_FORCE_REF", "0",
          "1 = route every op through the plain-PyTorch reference "
          "implementation instead of the HIP kernels (numerics baselines)."),

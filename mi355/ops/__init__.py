@@ -181,8 +181,11 @@ class _Linear2S(torch.autograd.Function):
 
 
 def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
-    """F.linear (no bias) with the two-stream backward on GPU."""
-    if x.is_cuda and os.environ.get("MI355_GEMM_2S", "1") != "0":
+    """F.linear (no bias); MI355_GEMM_2S=1 opts into the two-stream backward
+    (measured a ~3% LOSS at B=2 — the per-call fork/join events cost more
+    than the co-scheduling wins — so it stays off; revisit under a
+    pre-instantiated graph in round 2)."""
+    if x.is_cuda and os.environ.get("MI355_GEMM_2S", "0") == "1":
         return _Linear2S.apply(x, w)
     return torch.nn.functional.linear(x, w)
 
