@@ -273,6 +273,10 @@ class SegformerForSemanticSegmentation(nn.Module):
         return next(self.parameters()).device
 
     def forward(self, pixel_values=None, labels=None, **_):
+        # data loaders hand over fp32 images; run in the model's dtype
+        pdtype = next(self.parameters()).dtype
+        if pixel_values.dtype != pdtype:
+            pixel_values = pixel_values.to(pdtype)
         features = self.segformer(pixel_values)
         logits = self.decode_head(features)  # (B, L, H/4, W/4)
         loss = None
