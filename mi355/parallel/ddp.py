@@ -100,24 +100,26 @@ class FlatReplica:
         device = next(model.parameters()).device
         dtype = next(model.parameters()).dtype
 
+        # partition by (decay, dtype): a model can keep some params in fp32
+        # (e.g. SegFormer's BatchNorm under a bf16 cast) — those get their own
+        # flat group instead of being silently re-pointed into a bf16 buffer
         seen: dict[int, str] = {}
-        decay, nodecay = [], []
+        parts: dict[tuple, list] = {}
         for name, p in model.named_parameters():
             if not p.requires_grad or id(p) in seen:
                 continue
             seen[id(p)] = name
-            (nodecay if _no_decay(name, p) else decay).append((name, p))
-        decay.reverse()
-        nodecay.reverse()
+            key = ("no_decay" if _no_decay(name, p) else "decay", p.dtype)
+            parts.setdefault(key, []).append((name, p))
         self.groups = []
-        if decay:
-            self.groups.append(("decay", _FlatGroup(decay, dtype, device,
-                                                    bucket_bytes,
-                                                    assign_grads=not deferred_grads)))
-        if nodecay:
-            self.groups.append(("no_decay", _FlatGroup(nodecay, dtype, device,
-                                                       bucket_bytes,
-                                                       assign_grads=not deferred_grads)))
+        for decay_name in ("decay", "no_decay"):
+            for (dn, dt), named in parts.items():
+                if dn != decay_name:
+                    continue
+                named.reverse()
+                self.groups.append((decay_name,
+                                    _FlatGroup(named, dt, device, bucket_bytes,
+                                               assign_grads=not deferred_grads)))
 
         self._works = []
         self._flushed = set()

@@ -72,3 +72,33 @@ def test_segformer_trainer_two_cpu_workers(tmp_path):
 
     model = res.checkpoint.get_model(SegformerForSemanticSegmentation)
     assert model.config.num_labels == 12
+
+
+def test_flat_replica_preserves_batchnorm_dtype():
+    """bf16-cast SegFormer keeps BatchNorm fp32 (its own flat group) and the
+    fused optimizer steps both dtype groups."""
+    import torch
+
+    from mi355.models.segformer import SegformerConfig
+    from mi355.models.segformer.modeling import SegformerForSemanticSegmentation
+    from mi355.parallel import FlatAdamW, FlatReplica
+
+    torch.manual_seed(3)
+    model = SegformerForSemanticSegmentation(
+        SegformerConfig(**TINY)).to(torch.bfloat16)
+    assert model.decode_head.batch_norm.weight.dtype == torch.float32
+    replica = FlatReplica(model)
+    assert model.decode_head.batch_norm.weight.dtype == torch.float32
+    dtypes = {g.flat_param.dtype for _, g in replica.groups}
+    assert dtypes == {torch.bfloat16, torch.float32}
+    opt = FlatAdamW(replica, lr=1e-3)
+    pix = torch.randn(2, 3, 32, 32)
+    labels = torch.randint(0, 12, (2, 32, 32))
+    before = model.decode_head.batch_norm.weight.detach().clone()
+    opt.zero_grad()
+    out = model(pixel_values=pix, labels=labels)
+    out.loss.backward()
+    replica.finish_grad_sync()
+    opt.step()
+    assert out.loss.isfinite()
+    assert not torch.equal(before, model.decode_head.batch_norm.weight)
