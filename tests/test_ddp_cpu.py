@@ -227,3 +227,51 @@ def test_deferred_grads_match_inplace():
 
     a, b = run(False), run(True)
     assert torch.allclose(a, b, atol=1e-7), (a - b).abs().max()
+
+
+def _worker_mixed_dtype(rank, world, port, q):
+    """Mixed-dtype model (bf16 body + fp32 BatchNorm-style params): the
+    per-dtype flat groups must broadcast and all-reduce independently."""
+    os.environ.update({
+        "RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": str(world),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    from mi355.parallel import FlatReplica, init_distributed
+
+    init_distributed(backend="gloo")
+    torch.manual_seed(100 + rank)  # DIFFERENT init per rank: broadcast must fix
+    model = torch.nn.Sequential(
+        torch.nn.Linear(8, 8, bias=False).to(torch.bfloat16),
+        torch.nn.Linear(8, 4, bias=True),  # fp32
+    )
+    replica = FlatReplica(model, bucket_mb=0.001)
+    dtypes = sorted(str(g.flat_param.dtype) for _, g in replica.groups)
+    x = torch.randn(4, 8, generator=torch.Generator().manual_seed(9))
+    replica.zero_grads()
+    out = model[1](model[0](x.to(torch.bfloat16)).float())
+    out.square().mean().backward()
+    replica.finish_grad_sync()
+    flat = torch.cat([g.flat_grad.float() for _, g in replica.groups])
+    par = torch.cat([g.flat_param.float() for _, g in replica.groups])
+    q.put((rank, dtypes, flat.numpy().copy(), par.numpy().copy()))
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_mixed_dtype_groups_sync():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_mixed_dtype, args=(r, 2, 29547, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    res = {r: (d, g, pa) for r, d, g, pa in (q.get(timeout=240), q.get(timeout=240))}
+    for p in procs:
+        p.join(timeout=60)
+    import numpy as np
+
+    # decay-bf16, decay-fp32 (weight), no_decay-fp32 (bias)
+    assert res[0][0] == ["torch.bfloat16", "torch.float32", "torch.float32"]
+    np.testing.assert_allclose(res[0][2], res[1][2])  # params broadcast from rank0
+    np.testing.assert_allclose(res[0][1], res[1][1], rtol=1e-5)  # grads reduced
