@@ -72,6 +72,9 @@ FLAGS = [
          "examples/: tiny model + CPU (defaults to 1 when no GPU present)."),
     Flag("MI355_NUM_WORKERS", "2",
          "examples/: trainer worker count (the reference uses 2)."),
+    Flag("MI355_SANITIZE", "<unset>",
+         "Build the native store/tokenizer with -fsanitize=<value> "
+         "(address|thread) — §5.2 sanitizer CI target."),
     Flag("MI355_JOB_NAME", "<entrypoint>",
          "mi355 job submit: override the run name recorded in job status."),
 ]
