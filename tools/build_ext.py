@@ -119,6 +119,9 @@ def build_cpp_module(src_rel: str, name: str, force: bool = False) -> Path:
     cmd = ["g++", "-O2", "-std=c++17", "-shared", "-fPIC",
            f"-I{pybind11.get_include()}", f"-I{sysconfig.get_paths()['include']}",
            str(src), "-o", str(out)]
+    san = os.environ.get("MI355_SANITIZE")
+    if san:
+        cmd[1:1] = [f"-fsanitize={san}", "-g", "-fno-omit-frame-pointer"]
     r = subprocess.run(cmd, capture_output=True, text=True)
     if r.returncode != 0:
         sys.stderr.write(r.stdout[-2000:] + r.stderr[-6000:])
