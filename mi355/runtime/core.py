@@ -262,6 +262,10 @@ class _RemoteError:
     tb: str
 
 
+class WorkerCrashedError(RuntimeError):
+    """A stateless-task worker process died while running a task."""
+
+
 # --------------------------------------------------------------------------- #
 # Runtime (driver side)                                                       #
 # --------------------------------------------------------------------------- #
@@ -302,16 +306,47 @@ class Runtime:
         atexit.register(self.shutdown)
 
     def _reap(self):
+        last_liveness = time.monotonic()
         while self._alive:
             try:
                 wid, ref_id, name, t0, t1 = self.done_q.get(timeout=0.2)
             except queue.Empty:
+                if time.monotonic() - last_liveness > 1.0:
+                    self._check_workers()
+                    last_liveness = time.monotonic()
                 continue
             except (EOFError, OSError):
                 return
             with self._lock:
                 self._inflight.pop(ref_id, None)
                 self._events.append((wid, name, t0, t1))
+            self._idle.put(wid)
+
+    def _check_workers(self):
+        """Failure detection: a worker that died mid-task
+        (crash / OOM-kill / os._exit) fails its in-flight ref with a
+        WorkerCrashedError and is respawned with fresh queues, so the pool
+        keeps its capacity and a blocked get() raises instead of hanging."""
+        ctx = mp.get_context("spawn")
+        for wid, (p, tq, gpu_id) in enumerate(list(self.workers)):
+            if p.is_alive():
+                continue
+            dead_refs = []
+            with self._lock:
+                for ref_id, w in list(self._inflight.items()):
+                    if w == wid:
+                        dead_refs.append(ref_id)
+                        self._inflight.pop(ref_id)
+            for ref_id in dead_refs:
+                err = WorkerCrashedError(
+                    f"worker {wid} died (exitcode {p.exitcode}) while running "
+                    f"task {ref_id}")
+                self.store.put(_RemoteError(err, str(err)), ObjectRef(ref_id))
+            ntq = ctx.Queue()
+            np_ = ctx.Process(target=_worker_main,
+                              args=(wid, gpu_id, ntq, self.done_q), daemon=True)
+            np_.start()
+            self.workers[wid] = (np_, ntq, gpu_id)
             self._idle.put(wid)
 
     # -- tasks ---------------------------------------------------------- #
@@ -368,7 +403,7 @@ class Runtime:
             if isinstance(v, _RemoteError):
                 from .actor import ActorDiedError
 
-                if isinstance(v.exc, ActorDiedError):
+                if isinstance(v.exc, (ActorDiedError, WorkerCrashedError)):
                     raise v.exc
                 raise RuntimeError(f"remote task failed:\n{v.tb}") from v.exc
             out.append(v)

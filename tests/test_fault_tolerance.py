@@ -53,3 +53,33 @@ def test_failure_config_retries_from_checkpoint(tmp_path):
     res = t.fit()
     assert res.error is None, res.error
     assert res.metrics["epoch"] == 4  # resumed past the crash to completion
+
+
+def test_task_worker_crash_detected_and_respawned():
+    """A stateless-task worker that dies mid-task fails its ref with
+    WorkerCrashedError and is respawned, so later tasks still run."""
+    import pytest
+
+    import mi355
+    from mi355.runtime.core import WorkerCrashedError
+
+    mi355.init(num_workers=2)
+    try:
+        @mi355.remote
+        def die_hard():
+            import os
+
+            os._exit(23)
+
+        @mi355.remote
+        def fine(x):
+            return x * 2
+
+        ref = die_hard.remote()
+        with pytest.raises(WorkerCrashedError):
+            mi355.get(ref, timeout=30)
+        # pool recovered: full fan-out still works
+        assert mi355.get([fine.remote(i) for i in range(6)]) == \
+            [0, 2, 4, 6, 8, 10]
+    finally:
+        mi355.shutdown()
