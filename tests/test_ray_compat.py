@@ -49,3 +49,14 @@ def test_ray_serve_surface():
     d = serve.PredictorDeployment.options(name="X", num_replicas=2,
                                           route_prefix="/x")
     assert d.num_replicas == 2 and d.route_prefix == "/x"
+
+
+def test_ray_timeline_and_exceptions():
+    import ray
+    import ray.exceptions
+
+    assert callable(ray.timeline)
+    assert issubclass(ray.exceptions.RayActorError, RuntimeError)
+    from mi355 import ActorDiedError
+
+    assert ray.exceptions.RayActorError is ActorDiedError
