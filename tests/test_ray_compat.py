@@ -60,3 +60,26 @@ def test_ray_timeline_and_exceptions():
     from mi355 import ActorDiedError
 
     assert ray.exceptions.RayActorError is ActorDiedError
+
+
+def test_reference_import_forms():
+    """Every `from ray...` import statement that appears in the reference
+    notebooks must resolve (SURVEY.md §2.2 C-table citations)."""
+    from ray import serve, tune  # noqa: F401
+    from ray.air import Checkpoint  # noqa: F401
+    from ray.air.config import (CheckpointConfig, RunConfig,  # noqa: F401
+                                ScalingConfig)
+    from ray.data.preprocessors import BatchMapper, MinMaxScaler  # noqa: F401
+    from ray.serve import PredictorDeployment  # noqa: F401
+    from ray.serve.http_adapters import pandas_read_json  # noqa: F401
+    from ray.train.batch_predictor import BatchPredictor  # noqa: F401
+    from ray.train.huggingface import (HuggingFaceCheckpoint,  # noqa: F401
+                                       HuggingFacePredictor,
+                                       HuggingFaceTrainer)
+    from ray.train.predictor import Predictor  # noqa: F401
+    from ray.train.xgboost import (XGBoostPredictor,  # noqa: F401
+                                   XGBoostTrainer)
+    from ray.tune import Tuner  # noqa: F401
+    from ray.tune.schedulers.async_hyperband import ASHAScheduler  # noqa: F401
+    from ray.tune.tuner import TuneConfig  # noqa: F401
+    from ray.util.actor_pool import ActorPool  # noqa: F401
