@@ -1,3 +1,5 @@
+"""ASHA tuner smoke on real hardware (falls back to CPU): 2 trials,
+tiny T5, epoch-wise eval_loss reports driving the lockstep early stop."""
 import sys
 sys.path.insert(0, ".")
 import torch
@@ -34,7 +36,7 @@ def init_worker(train_ds, eval_ds=None, **cfg):
                              num_train_epochs=cfg.get("epochs", 2),
                              per_device_train_batch_size=4,
                              evaluation_strategy="epoch", save_strategy="epoch",
-                             bf16=True, disable_tqdm=True)
+                             bf16=torch.cuda.is_available(), disable_tqdm=True)
     return Trainer(model=model, args=args, train_dataset=train_ds,
                    eval_dataset=eval_ds, tokenizer=tok)
 
@@ -43,7 +45,7 @@ ds = synthetic_alpaca(96, seed=3)
 tr, ev = ds.train_test_split(test_size=0.25, shuffle=True, seed=3)
 trainer = HuggingFaceTrainer(
     trainer_init_per_worker=init_worker,
-    scaling_config=ScalingConfig(num_workers=1, use_gpu=True),
+    scaling_config=ScalingConfig(num_workers=1, use_gpu=torch.cuda.is_available()),
     datasets={"train": tr, "evaluation": ev},
     preprocessor=BatchMapper(prep, batch_format="pandas", batch_size=64),
 )
