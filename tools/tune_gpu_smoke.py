@@ -40,22 +40,27 @@ def init_worker(train_ds, eval_ds=None, **cfg):
     return Trainer(model=model, args=args, train_dataset=train_ds,
                    eval_dataset=eval_ds, tokenizer=tok)
 
-ray.init()
-ds = synthetic_alpaca(96, seed=3)
-tr, ev = ds.train_test_split(test_size=0.25, shuffle=True, seed=3)
-trainer = HuggingFaceTrainer(
-    trainer_init_per_worker=init_worker,
-    scaling_config=ScalingConfig(num_workers=1, use_gpu=torch.cuda.is_available()),
-    datasets={"train": tr, "evaluation": ev},
-    preprocessor=BatchMapper(prep, batch_format="pandas", batch_size=64),
-)
-tuner = Tuner(trainer,
-              param_space={"trainer_init_config": {
-                  "learning_rate": tune.choice([1e-4, 2e-5]), "epochs": 3}},
-              tune_config=TuneConfig(metric="eval_loss", mode="min",
-                                     num_samples=2,
-                                     scheduler=ASHAScheduler(max_t=3)),
-              run_config=RunConfig(name="asha-gpu-smoke"))
-grid = tuner.fit()
-best = grid.get_best_result()
-print("ASHA GPU smoke OK, best eval_loss:", best.metrics.get("eval_loss"))
+def main():
+    ray.init()
+    ds = synthetic_alpaca(96, seed=3)
+    tr, ev = ds.train_test_split(test_size=0.25, shuffle=True, seed=3)
+    trainer = HuggingFaceTrainer(
+        trainer_init_per_worker=init_worker,
+        scaling_config=ScalingConfig(num_workers=1, use_gpu=torch.cuda.is_available()),
+        datasets={"train": tr, "evaluation": ev},
+        preprocessor=BatchMapper(prep, batch_format="pandas", batch_size=64),
+    )
+    tuner = Tuner(trainer,
+                  param_space={"trainer_init_config": {
+                      "learning_rate": tune.choice([1e-4, 2e-5]), "epochs": 3}},
+                  tune_config=TuneConfig(metric="eval_loss", mode="min",
+                                         num_samples=2,
+                                         scheduler=ASHAScheduler(max_t=3)),
+                  run_config=RunConfig(name="asha-gpu-smoke"))
+    grid = tuner.fit()
+    best = grid.get_best_result()
+    print("ASHA GPU smoke OK, best eval_loss:", best.metrics.get("eval_loss"))
+
+
+if __name__ == "__main__":
+    main()
