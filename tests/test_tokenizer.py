@@ -97,3 +97,39 @@ def test_against_sentencepiece_library_if_available(tok, tmp_path):
         theirs = proc.EncodeAsIds(text)
         assert ours == theirs, (tok.convert_ids_to_tokens(ours),
                                 proc.EncodeAsPieces(text))
+
+
+def test_tokenizer_properties_hypothesis():
+    """Property-based checks over arbitrary text (hypothesis): ids in range,
+    eos-terminated, deterministic, pad/truncation shape contract."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from mi355.tokenize import T5Tokenizer
+
+    tok = T5Tokenizer.train_new(
+        ["the quick brown fox jumps over the lazy dog"] * 4 +
+        ["pack my box with five dozen liquor jugs"] * 4,
+        vocab_size=200, model_max_length=16)
+
+    @settings(max_examples=60, deadline=None)
+    @given(st.text(alphabet=st.characters(codec="ascii",
+                                          exclude_categories=("Cc", "Cs")),
+                   max_size=60))
+    def check(text):
+        enc = tok([text], padding="max_length", truncation=True,
+                  max_length=16, return_tensors="np")
+        ids = enc["input_ids"][0]
+        mask = enc["attention_mask"][0]
+        assert ids.shape == (16,) and mask.shape == (16,)
+        assert ids.max() < len(tok) and ids.min() >= 0
+        nvalid = int(mask.sum())
+        assert (ids[nvalid:] == tok.pad_token_id).all()
+        assert nvalid >= 1 and ids[nvalid - 1] == tok.eos_token_id
+        enc2 = tok([text], padding="max_length", truncation=True,
+                   max_length=16, return_tensors="np")
+        assert (enc2["input_ids"][0] == ids).all()
+        # decode drops specials and never raises
+        tok.batch_decode(ids[None], skip_special_tokens=True)
+
+    check()
