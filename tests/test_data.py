@@ -157,3 +157,36 @@ def test_map_batches_actor_pool_strategy():
                          fn_constructor_args=(100,), batch_format="pandas",
                          batch_size=10)
     assert sorted(out.to_pandas()["z"].tolist()) == [100.0 + i for i in range(20)]
+
+
+def test_dataset_roundtrip_properties_hypothesis():
+    """Property: any mixed-typed list-of-dicts survives
+    from_items -> blocks -> to_pandas with values intact and ops composing."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    import mi355.data as mdata
+
+    row = st.fixed_dictionaries({
+        "i": st.integers(min_value=-10**6, max_value=10**6),
+        "f": st.floats(allow_nan=False, allow_infinity=False, width=32),
+        "s": st.text(alphabet=st.characters(codec="ascii",
+                                            exclude_categories=("Cc", "Cs")),
+                     max_size=12),
+    })
+
+    @settings(max_examples=25, deadline=None)
+    @given(st.lists(row, min_size=1, max_size=24))
+    def check(rows):
+        ds = mdata.Dataset.from_items(rows)
+        assert ds.count() == len(rows)
+        df = ds.to_pandas()
+        assert list(df["i"]) == [r["i"] for r in rows]
+        assert list(df["s"]) == [r["s"] for r in rows]
+        lim = ds.limit(3)
+        assert lim.count() == min(3, len(rows))
+        doubled = ds.map_batches(lambda d: d.assign(i=d["i"] * 2),
+                                 batch_format="pandas", batch_size=8)
+        assert list(doubled.to_pandas()["i"]) == [r["i"] * 2 for r in rows]
+
+    check()
