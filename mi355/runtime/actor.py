@@ -127,7 +127,10 @@ class ActorHandle:
             f"actor {getattr(self, '_cls_name', 'Actor')} died "
             f"(exitcode {self._proc.exitcode})")
         for ref_id in self._pending:
-            store.put(_RemoteError(err, str(err)), ObjectRef(ref_id))
+            try:
+                store.put(_RemoteError(err, str(err)), ObjectRef(ref_id))
+            except Exception:
+                pass  # result landed before the crash: keep it
         self._pending.clear()
         if self._restarts_left > 0:
             self._restarts_left -= 1

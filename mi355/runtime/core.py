@@ -341,7 +341,10 @@ class Runtime:
                 err = WorkerCrashedError(
                     f"worker {wid} died (exitcode {p.exitcode}) while running "
                     f"task {ref_id}")
-                self.store.put(_RemoteError(err, str(err)), ObjectRef(ref_id))
+                try:
+                    self.store.put(_RemoteError(err, str(err)), ObjectRef(ref_id))
+                except Exception:
+                    pass  # worker finished the write before dying: keep it
             ntq = ctx.Queue()
             np_ = ctx.Process(target=_worker_main,
                               args=(wid, gpu_id, ntq, self.done_q), daemon=True)
