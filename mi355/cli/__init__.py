@@ -56,6 +56,16 @@ def cmd_status(args):
     return 0
 
 
+def cmd_flags(_args):
+    from ..flags import FLAGS, snapshot
+
+    snap = snapshot()
+    for f in FLAGS:
+        mark = "*" if f.name in os.environ else " "
+        print(f"{mark} {f.name:<24} {snap[f.name]:<10} {f.help}")
+    return 0
+
+
 def cmd_gpus(_args):
     try:
         out = subprocess.run(["rocm-smi", "--showuse", "--showmemuse"],
@@ -79,6 +89,8 @@ def main(argv=None):
     st = sub.add_parser("status")
     st.add_argument("--dir", default=None)
     st.set_defaults(fn=cmd_status)
+    fl = sub.add_parser("flags", help="list every MI355_* env knob")
+    fl.set_defaults(fn=cmd_flags)
     g = sub.add_parser("gpus")
     g.set_defaults(fn=cmd_gpus)
     args = ap.parse_args(argv)
