@@ -80,7 +80,7 @@ class ActorHandle:
         rt = _rt()
         self._gpu_id = None
         if num_gpus and rt.num_gpus:
-            self._gpu_id = next(_gpu_rr) % rt.num_gpus
+            self._gpu_id = rt.gpu_ids[next(_gpu_rr) % rt.num_gpus]
         self._cls_bytes = cloudpickle.dumps(cls)
         self._init_bytes = cloudpickle.dumps((list(init_args), dict(init_kwargs)))
         self._restarts_left = max_restarts

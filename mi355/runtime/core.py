@@ -275,7 +275,13 @@ class Runtime:
 
         _session_prefix()  # pin the session id BEFORE spawn so workers inherit it
 
-        self.num_gpus = torch.cuda.device_count() if num_gpus is None else num_gpus
+        vis = os.environ.get("MI355_VISIBLE_DEVICES")
+        self.gpu_ids = ([int(x) for x in vis.split(",") if x != ""]
+                        if vis is not None
+                        else list(range(torch.cuda.device_count())))
+        if num_gpus is not None:
+            self.gpu_ids = self.gpu_ids[:num_gpus]
+        self.num_gpus = len(self.gpu_ids)
         if num_workers is None:
             num_workers = max(2, min(os.cpu_count() or 4, 16))
             if self.num_gpus:
@@ -288,7 +294,7 @@ class Runtime:
         self._inflight: dict[str, int] = {}
         self._lock = threading.Lock()
         for wid in range(num_workers):
-            gpu_id = wid % self.num_gpus if self.num_gpus else None
+            gpu_id = self.gpu_ids[wid % self.num_gpus] if self.num_gpus else None
             tq = ctx.Queue()
             p = ctx.Process(target=_worker_main, args=(wid, gpu_id, tq, self.done_q),
                             daemon=True)

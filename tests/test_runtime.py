@@ -165,3 +165,18 @@ def test_parallel_rf_demo_equivalence():
         assert sorted(seq) == sorted(par)
     finally:
         mi355.shutdown()
+
+
+def test_visible_devices_restricts_gpu_pinning(monkeypatch):
+    """MI355_VISIBLE_DEVICES limits which GPU ids workers/actors pin to."""
+    monkeypatch.setenv("MI355_VISIBLE_DEVICES", "5,7")
+    mi355.init(num_workers=2)
+    try:
+        from mi355.runtime.core import _rt
+
+        rt = _rt()
+        assert rt.gpu_ids == [5, 7] and rt.num_gpus == 2
+        assert {g for _, _, g in rt.workers} <= {5, 7}
+    finally:
+        mi355.shutdown()
+        monkeypatch.delenv("MI355_VISIBLE_DEVICES")
