@@ -403,3 +403,33 @@ def test_attention_kv_packed_matches_split():
     out_s.backward(d)
     assert torch.equal(q.grad, q2.grad)
     assert torch.equal(kv.grad, kv2.grad)
+
+
+@gpu
+def test_attention_dropout_consistent_across_bwd_impls(monkeypatch):
+    """The fused attention-prob dropout mask derives from absolute (q,k)
+    indices, so the v1 (q-split) and v3 (split-kernel) backwards must see the
+    SAME mask and produce matching grads for one seed."""
+    torch.manual_seed(21)
+    B, S, H, D = 2, 256, 4, 64
+    make = lambda: torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    q0, k0, v0 = make(), make(), make()
+    d = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+
+    def run(impl):
+        monkeypatch.setenv("MI355_ATTN_BWD", impl)
+        torch.manual_seed(5)
+        ops._dropout_seeds.clear()  # same device seed for both runs
+        q = q0.clone().requires_grad_(True)
+        k = k0.clone().requires_grad_(True)
+        v = v0.clone().requires_grad_(True)
+        out = ops.attention(q, k, v, p_drop=0.3)
+        out.backward(d)
+        return out.detach(), q.grad, k.grad, v.grad
+
+    o1, dq1, dk1, dv1 = run("v1")
+    o3, dq3, dk3, dv3 = run("v3")
+    monkeypatch.delenv("MI355_ATTN_BWD")
+    assert torch.equal(o1, o3)  # fwd identical (same kernel + seed)
+    for a, b in ((dq1, dq3), (dk1, dk3), (dv1, dv3)):
+        assert _relerr(a.float(), b.float()) < 3e-2
