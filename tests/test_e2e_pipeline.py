@@ -160,3 +160,19 @@ def test_full_pipeline(tokenizer, tmp_path):
     df = out.to_pandas()
     assert "generated_output" in df.columns
     assert len(df) == valid_ds.count()
+
+
+def test_checkpoint_as_directory_context():
+    """Checkpoint dict<->dir duality: as_directory() materializes a dict
+    checkpoint and cleans up after."""
+    import os
+
+    from mi355.air import Checkpoint
+
+    ck = Checkpoint.from_dict({"weights": [1, 2, 3], "epoch": 7})
+    with ck.as_directory() as d:
+        assert os.path.isdir(d)
+        back = Checkpoint.from_directory(d).to_dict()
+        assert back["epoch"] == 7 and back["weights"] == [1, 2, 3]
+    roundtrip = Checkpoint.from_dict(ck.to_dict())
+    assert roundtrip.to_dict()["epoch"] == 7

@@ -151,3 +151,33 @@ def test_serve_actor_backed_replicas():
     finally:
         serve.shutdown()
         mi355.shutdown()
+
+
+def test_serve_routes_listing_and_metrics_route():
+    """GET / lists deployed routes (the proxy's route map); /-/metrics serves
+    the Prometheus registry."""
+    import json
+    import socket
+    import urllib.request
+
+    import pandas as pd
+
+    from mi355 import serve
+
+    @serve.deployment(name="lister", route_prefix="/lister")
+    class Echo:
+        def __call__(self, df: pd.DataFrame):
+            return df
+
+    s = socket.socket(); s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]; s.close()
+    serve.run(Echo.bind(), host="127.0.0.1", port=port)
+    try:
+        with urllib.request.urlopen(f"http://127.0.0.1:{port}/-/routes") as r:
+            routes = json.loads(r.read())
+        assert routes == {"/lister": "lister"}
+        with urllib.request.urlopen(f"http://127.0.0.1:{port}/-/metrics") as r:
+            assert "mi355_serve_requests_total" in r.read().decode()
+        assert serve.get_deployment("lister").name == "lister"
+    finally:
+        serve.shutdown()
