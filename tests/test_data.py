@@ -190,3 +190,27 @@ def test_dataset_roundtrip_properties_hypothesis():
         assert list(doubled.to_pandas()["i"]) == [r["i"] * 2 for r in rows]
 
     check()
+
+
+def test_filter_add_rename_select(taxi_like):
+    ds = taxi_like
+    flt = ds.filter(lambda row: row["passenger_count"] > 2)
+    assert all(r["passenger_count"] > 2 for r in flt.take_all())
+    plus = ds.add_column("double_fare", lambda df: df["fare_amount"] * 2)
+    df = plus.to_pandas()
+    assert (df["double_fare"] == df["fare_amount"] * 2).all()
+    ren = plus.rename_columns({"double_fare": "fare2"})
+    assert "fare2" in ren.columns() and "double_fare" not in ren.columns()
+    sel = ren.select_columns(["fare2"])
+    assert sel.columns() == ["fare2"]
+
+
+def test_random_shuffle_preserves_rows(taxi_like):
+    ds = taxi_like
+    sh = ds.random_shuffle(seed=3)
+    assert sh.count() == ds.count()
+    assert sorted(r["fare_amount"] for r in sh.take_all()) == \
+        sorted(r["fare_amount"] for r in ds.take_all())
+    sh2 = ds.random_shuffle(seed=3)
+    assert [r["fare_amount"] for r in sh.take_all()] == \
+        [r["fare_amount"] for r in sh2.take_all()]  # seed-deterministic
