@@ -4,8 +4,9 @@
 // Given dO, Q, K, V, O, LSE: recompute P tile-by-tile and produce
 //   dV = P^T dO
 //   dP = dO V^T ;  dS = P * (dP - rowsum(dO*O))
-//   dK = dS^T Q ;  dQ = dS K  (fp32 atomic accumulation over k-tiles)
-//   dBias = sum_b dS  (fp32 atomic scatter, bias broadcast over batch)
+//   dK = dS^T Q ;  dQ = dS K  (per-k-tile fp32 slabs, plain stores, fused
+//                              slab-reduce epilogue — no atomics)
+//   dBias = sum_b dS  (per-batch fp32 planes + fused batch-sum/cast)
 //
 // Geometry: block = 4 waves owns one (b, h, k-tile of 64 keys); waves own 16
 // key rows each; loop over q-tiles of 64. All MFMA v_mfma_f32_16x16x32.

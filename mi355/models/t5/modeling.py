@@ -90,7 +90,8 @@ def _register_fused_statedict_compat(parent: nn.Module, attr: str):
         if key in state_dict:
             w = state_dict.pop(key)
             for i, n in enumerate(names):
-                state_dict[f"{prefix}{n}.weight"] =                     w[i * fused.d_out:(i + 1) * fused.d_out]
+                state_dict[f"{prefix}{n}.weight"] = \
+                    w[i * fused.d_out:(i + 1) * fused.d_out]
         return state_dict
 
     def load_pre_hook(state_dict, prefix, local_metadata, strict, missing_keys,
