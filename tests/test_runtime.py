@@ -180,3 +180,38 @@ def test_visible_devices_restricts_gpu_pinning(monkeypatch):
     finally:
         mi355.shutdown()
         monkeypatch.delenv("MI355_VISIBLE_DEVICES")
+
+
+def test_object_store_concurrent_stress():
+    """§5.2 race-freedom: many concurrent producers/consumers over the
+    sealed-object store — every object is read back intact (seal word =
+    release/acquire, objects immutable after seal)."""
+    import numpy as np
+
+    mi355.init(num_workers=4)
+    try:
+        @mi355.remote
+        def produce(seed):
+            import numpy as _np
+
+            rng = _np.random.default_rng(seed)
+            return rng.integers(0, 255, size=4096, dtype=_np.uint8)
+
+        @mi355.remote
+        def check(arr, seed):
+            import numpy as _np
+
+            rng = _np.random.default_rng(seed)
+            return bool((arr == rng.integers(0, 255, size=4096,
+                                             dtype=_np.uint8)).all())
+
+        refs = [produce.remote(s) for s in range(24)]
+        oks = mi355.get([check.remote(r, s) for s, r in enumerate(refs)])
+        assert all(oks)
+        # driver-side reads of the same sealed objects concurrently
+        for s, r in enumerate(refs):
+            rng = np.random.default_rng(s)
+            assert (mi355.get(r) == rng.integers(0, 255, size=4096,
+                                                 dtype=np.uint8)).all()
+    finally:
+        mi355.shutdown()
