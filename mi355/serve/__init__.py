@@ -7,8 +7,10 @@ Introduction_to_Ray_AI_Runtime.ipynb:cell-70-78):
     requests.post("http://localhost:8000/rayair", json=[{...}])
 
 HTTP proxy on :8000 routes by prefix to a round-robin pool of predictor
-replicas. Replicas live in-process (model loaded once each); with the mi355
-runtime initialised and num_gpus>0 they are GPU-pinned actor processes.
+replicas (GET /-/routes lists them, /-/metrics serves Prometheus text).
+Replicas live in-process by default (model loaded once each); pass
+``ray_actor_options={"num_gpus": 1}`` with the mi355 runtime initialised and
+each replica becomes its own GPU-pinned actor process.
 """
 from .deployment import (Deployment, PredictorDeployment, deployment,
                          pandas_read_json, json_request, run, shutdown,
