@@ -356,6 +356,9 @@ class Runtime:
                               args=(wid, gpu_id, ntq, self.done_q), daemon=True)
             np_.start()
             self.workers[wid] = (np_, ntq, gpu_id)
+            # if the worker was idle when it died its token is still in
+            # _idle; the extra token only lets the fresh worker be handed two
+            # tasks back-to-back once (they queue; accounting self-corrects)
             self._idle.put(wid)
 
     # -- tasks ---------------------------------------------------------- #
