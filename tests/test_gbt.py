@@ -85,3 +85,42 @@ def test_xgboost_trainer_two_workers(tmp_path):
     out = bp.predict(valid.drop_columns(["is_big_tip"]), batch_size=64).to_pandas()
     assert len(out) == valid.count()
     assert out["predictions"].between(0, 1).all()
+
+
+def test_serve_xgboost_service(tmp_path):
+    """The Intro notebook's exact serve cell: PredictorDeployment.options(
+    name='XGBoostService', num_replicas=2, route_prefix='/rayair').bind(
+    XGBoostPredictor, checkpoint, http_adapter=pandas_read_json) + POST."""
+    import json
+    import socket
+    import urllib.request
+
+    from ray import serve
+    from ray.serve import PredictorDeployment
+    from ray.serve.http_adapters import pandas_read_json
+
+    df = make_data(200)
+    m = GBTModel({"objective": "binary:logistic", "max_depth": 3, "eta": 0.3})
+    m.fit(df[["x1", "x2"]].to_numpy(), df["is_big_tip"].to_numpy(),
+          num_boost_round=8)
+    ckpt_dir = str(tmp_path / "gbt")
+    m.save_pretrained(ckpt_dir)
+    from mi355.air import Checkpoint
+
+    ckpt = Checkpoint.from_directory(ckpt_dir)
+
+    s = socket.socket(); s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]; s.close()
+    serve.run(PredictorDeployment.options(
+        name="XGBoostService", num_replicas=2, route_prefix="/rayair",
+    ).bind(XGBoostPredictor, ckpt, http_adapter=pandas_read_json), port=port)
+    try:
+        body = json.dumps({"x1": [0.1, 0.9], "x2": [0.2, 0.8]}).encode()
+        req = urllib.request.Request(f"http://127.0.0.1:{port}/rayair",
+                                     data=body, method="POST")
+        with urllib.request.urlopen(req) as r:
+            rows = json.loads(r.read())
+        assert len(rows) == 2
+        assert all(0.0 <= row["predictions"] <= 1.0 for row in rows)
+    finally:
+        serve.shutdown()
