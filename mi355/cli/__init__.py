@@ -42,8 +42,7 @@ def cmd_status(args):
     if not os.path.isdir(base):
         print(f"no results under {base}")
         return 0
-    for run in sorted(os.listdir(base)):
-        run_dir = os.path.join(base, run)
+    def show(run, run_dir, indent=""):
         metrics = os.path.join(run_dir, "metrics.jsonl")
         line = ""
         if os.path.exists(metrics):
@@ -51,8 +50,18 @@ def cmd_status(args):
                 rows = f.readlines()
             if rows:
                 line = rows[-1].strip()
-        ckpts = [d for d in os.listdir(run_dir) if d.startswith("checkpoint")]
-        print(f"{run}: {len(ckpts)} checkpoint(s) {line}")
+        entries = os.listdir(run_dir)
+        ckpts = [d for d in entries if d.startswith("checkpoint")]
+        print(f"{indent}{run}: {len(ckpts)} checkpoint(s) {line}")
+        for sub in sorted(entries):  # tune trials nest one level down
+            sub_dir = os.path.join(run_dir, sub)
+            if sub.startswith("trial_") and os.path.isdir(sub_dir):
+                show(sub, sub_dir, indent + "  ")
+
+    for run in sorted(os.listdir(base)):
+        run_dir = os.path.join(base, run)
+        if os.path.isdir(run_dir):
+            show(run, run_dir)
     return 0
 
 
