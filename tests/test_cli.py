@@ -31,3 +31,24 @@ def test_job_submit_propagates_failure(tmp_path):
                        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
     assert r.returncode == 3
     assert "FAILED" in r.stdout
+
+
+def test_status_lists_runs_and_nested_trials(tmp_path, capsys):
+    import json
+
+    from mi355.cli import cmd_status
+
+    run = tmp_path / "myrun"
+    (run / "checkpoint_000001").mkdir(parents=True)
+    (run / "metrics.jsonl").write_text(json.dumps({"epoch": 1, "loss": 0.5}) + "\n")
+    trial = tmp_path / "tune" / "trial_0000"
+    (trial / "checkpoint_000001").mkdir(parents=True)
+    (trial / "metrics.jsonl").write_text(json.dumps({"eval_loss": 0.1}) + "\n")
+
+    class A:
+        dir = str(tmp_path)
+
+    assert cmd_status(A()) == 0
+    out = capsys.readouterr().out
+    assert "myrun: 1 checkpoint(s)" in out
+    assert "trial_0000: 1 checkpoint(s)" in out and "eval_loss" in out
