@@ -130,6 +130,12 @@ def test_timeline_chrome_trace(tmp_path):
             return i + 1
 
         mi355.get([named_work.remote(i) for i in range(4)])
+        # events are appended by the reaper thread, which can trail get()
+        import time as _t
+
+        deadline = _t.monotonic() + 10
+        while len(mi355.timeline()) < 4 and _t.monotonic() < deadline:
+            _t.sleep(0.02)
         out = tmp_path / "trace.json"
         events = mi355.timeline(str(out))
         assert len(events) >= 4
