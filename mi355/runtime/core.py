@@ -472,6 +472,10 @@ def init(num_workers: int | None = None, num_gpus: int | None = None,
             return _runtime
         raise RuntimeError("mi355 runtime already initialized")
     _runtime = Runtime(num_workers=num_workers, num_gpus=num_gpus)
+    # the reference's env smoke test is "ray.init() prints a started line"
+    # (Install_locally.md) — mirror that signal
+    print(f"mi355 runtime started: {len(_runtime.workers)} workers, "
+          f"{_runtime.num_gpus} GPU(s), session {_session_prefix()}")
     return _runtime
 
 
