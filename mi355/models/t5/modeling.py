@@ -453,8 +453,10 @@ class T5ForConditionalGeneration(nn.Module):
                 and not self.training):
             try:
                 return self._generate_graphed(input_ids, attention_mask, max_new_tokens)
-            except Exception:
-                pass  # eager fallback below
+            except Exception as e:
+                if not getattr(self, "_gen_graph_warned", False):
+                    self._gen_graph_warned = True
+                    print(f"# decode-graph capture failed, eager fallback: {e!r}")
         return self._generate_eager(input_ids, attention_mask, max_new_tokens,
                                     min_new_tokens, eos_token_id)
 
