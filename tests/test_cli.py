@@ -4,6 +4,7 @@ import os
 import subprocess
 import sys
 
+import pytest
 import yaml
 
 
@@ -52,3 +53,22 @@ def test_status_lists_runs_and_nested_trials(tmp_path, capsys):
     out = capsys.readouterr().out
     assert "myrun: 1 checkpoint(s)" in out
     assert "trial_0000: 1 checkpoint(s)" in out and "eval_loss" in out
+
+
+@pytest.mark.timeout(600)
+@pytest.mark.parametrize("spec", ["examples/flan-t5-job.yml",
+                                  "examples/segformer-job.yml"])
+def test_committed_job_specs_run(spec, tmp_path, monkeypatch):
+    """The job ymls shipped in examples/ must submit and succeed (SMALL/CPU)."""
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, MI355_SMALL="1", MI355_NUM_WORKERS="2")
+    out = subprocess.run([sys.executable, "-m", "mi355.cli", "job", "submit",
+                          os.path.join(repo, spec)],
+                         capture_output=True, text=True, timeout=540,
+                         cwd=repo, env=env)
+    assert out.returncode == 0, out.stdout[-1500:] + out.stderr[-1500:]
+    assert "SUCCEEDED" in out.stdout
