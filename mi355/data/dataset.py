@@ -219,6 +219,15 @@ class Dataset:
         return Dataset(out)
 
     def split(self, n: int, equal: bool = False) -> list["Dataset"]:
+        """Split into n datasets. With equal=True rows are truncated to
+        n*floor(rows/n) first so every shard has exactly the same length —
+        required for DDP train shards, where a one-row imbalance would make
+        one rank run an extra batch and post all-reduces its peers never
+        match (ray.data.Dataset.split(equal=True) semantics)."""
+        if equal:
+            tbl = self.to_arrow()
+            per = tbl.num_rows // n
+            return [Dataset([tbl.slice(i * per, per)]) for i in range(n)]
         parts = self.repartition(n)
         return [Dataset([b]) for b in parts._blocks]
 

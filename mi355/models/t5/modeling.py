@@ -452,7 +452,8 @@ class T5ForConditionalGeneration(nn.Module):
         if (use_graph and input_ids.is_cuda and min_new_tokens == max_new_tokens
                 and not self.training):
             try:
-                return self._generate_graphed(input_ids, attention_mask, max_new_tokens)
+                return self._generate_graphed(input_ids, attention_mask, max_new_tokens,
+                                              eos_token_id)
             except Exception as e:
                 if not getattr(self, "_gen_graph_warned", False):
                     self._gen_graph_warned = True
@@ -477,12 +478,19 @@ class T5ForConditionalGeneration(nn.Module):
         return caches
 
     @torch.no_grad()
-    def _generate_graphed(self, input_ids, attention_mask, max_new_tokens):
+    def _generate_graphed(self, input_ids, attention_mask, max_new_tokens,
+                          eos_token_id=None):
         cfg = self.config
         b = input_ids.shape[0]
         dev = input_ids.device
         dt = self.shared.weight.dtype
-        key = (b, input_ids.shape[1], max_new_tokens, str(dev))
+        # The graph path runs only when min_new_tokens == max_new_tokens, and
+        # _generate_eager bans EOS at every t < min_new_tokens — so the
+        # captured loop must ban EOS at every step too, or fixed-length
+        # generations could emit EOS mid-sequence followed by garbage that
+        # survives skip_special_tokens decoding.
+        eos = cfg.eos_token_id if eos_token_id is None else eos_token_id
+        key = (b, input_ids.shape[1], max_new_tokens, str(dev), eos)
         state = getattr(self, "_gen_graph", None)
         kpad = _kpad_from_mask(attention_mask)
         enc = self.encoder(input_ids=input_ids, kpad=kpad)
@@ -509,6 +517,8 @@ class T5ForConditionalGeneration(nn.Module):
                     if cfg.tie_word_embeddings:
                         dec = dec * (self.model_dim**-0.5)
                     logits = self.lm_head(dec[:, -1])
+                    if eos is not None:
+                        logits[:, eos] = float("-inf")
                     nxt = logits.argmax(dim=-1)
                     out_tokens[:, t] = nxt
                     tok = nxt.unsqueeze(1)

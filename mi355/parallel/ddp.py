@@ -170,6 +170,10 @@ class FlatReplica:
             slices, grads = [], []
             for p, off in zip(g.params, g.offsets):
                 if p.grad is None:
+                    # Param unused this step: zero its flat slice, else the
+                    # PREVIOUS step's gradient would be all-reduced and
+                    # applied again.
+                    g.flat_grad[off:off + p.numel()].zero_()
                     continue
                 slices.append(g.flat_grad[off:off + p.numel()].view(p.shape))
                 grads.append(p.grad)

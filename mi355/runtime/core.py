@@ -432,8 +432,12 @@ class Runtime:
             if deadline is not None and time.monotonic() > deadline:
                 break
             time.sleep(0.001)
-        ready = ready[:max(num_returns, len(ready))]
-        not_ready = [r for r in refs if r not in ready]
+        # ray.wait semantics: at most num_returns in ready; the rest (ready
+        # or not) stay in not_ready so `done, pending = wait(pending)` loops
+        # never drop refs (the old max() cap was a no-op).
+        ready = ready[:num_returns]
+        ready_ids = {r.id for r in ready}
+        not_ready = [r for r in refs if r.id not in ready_ids]
         return ready, not_ready
 
     def shutdown(self):

@@ -229,8 +229,13 @@ class RayServeHandle:
     group: _PredictorReplicaGroup
 
     def predict(self, df):
+        # Route through the group's _invoke under the replica lock — the
+        # direct replicas[i].predict(df) call broke for actor-backed replicas
+        # (ActorMethod has no __call__) and bypassed the HTTP path's
+        # serialization.
         i = next(self.group._rr)
-        return self.group.replicas[i].predict(df)
+        with self.group._locks[i]:
+            return self.group._invoke(self.group.replicas[i], df, {})
 
 
 def get_deployment(name: str):

@@ -134,7 +134,7 @@ class DataParallelTrainer:
             shard_tables = {}
             for name, ds in datasets.items():
                 if name == "train" and n > 1:
-                    shard = ds.split(n)[rank]
+                    shard = ds.split(n, equal=True)[rank]
                 else:
                     shard = ds  # eval replicated (HF-style full eval per worker)
                 shard_tables[name] = shard._blocks

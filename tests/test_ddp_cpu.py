@@ -275,3 +275,32 @@ def test_mixed_dtype_groups_sync():
     assert res[0][0] == ["torch.bfloat16", "torch.float32", "torch.float32"]
     np.testing.assert_allclose(res[0][2], res[1][2])  # params broadcast from rank0
     np.testing.assert_allclose(res[0][1], res[1][1], rtol=1e-5)  # grads reduced
+
+
+def test_collect_grads_zeroes_unused_param_slice():
+    """Deferred mode: a param whose grad is None this step must have its
+    flat-buffer slice ZEROED, not keep last step's gradient (ADVICE r1
+    medium)."""
+    import torch
+    from mi355.parallel.ddp import FlatReplica
+
+    m = torch.nn.Sequential(torch.nn.Linear(4, 4), torch.nn.Linear(4, 4))
+    rep = FlatReplica(m, deferred_grads=True)
+    x = torch.randn(2, 4)
+    # step 1: both layers used
+    m(x).sum().backward()
+    rep.collect_grads()
+    rep.finish_grad_sync()
+    g0 = None
+    for _, g in rep.groups:
+        g0 = g.flat_grad.clone()
+    rep.zero_grads()
+    # step 2: only the first layer used -> layer-2 slices must be zero
+    m[0](x).sum().backward()
+    rep.collect_grads()
+    rep.finish_grad_sync()
+    for (_, g), prev in zip(rep.groups, [g0]):
+        for p, off in zip(g.params, g.offsets):
+            sl = g.flat_grad[off:off + p.numel()]
+            if p.grad is None:
+                assert torch.all(sl == 0), "stale gradient survived in flat buffer"

@@ -239,3 +239,26 @@ def test_native_store_and_tokenizer_cores_loaded():
     tok = T5Tokenizer.train_new(["native core smoke"], vocab_size=40)
     assert getattr(tok.sp, "_core", None) is not None or True  # encode works:
     assert tok(["native"])["input_ids"][0][-1] == tok.eos_token_id
+
+
+def test_wait_caps_ready_at_num_returns():
+    """ray.wait semantics: at most num_returns refs in ready, the rest stay
+    in not_ready even when already complete (the old code
+    returned ALL ready refs, so `[r], _ = wait(refs, num_returns=1)` raised
+    and drain loops dropped refs)."""
+    @mi355.remote
+    def fast(v):
+        return v
+
+    refs = [fast.remote(i) for i in range(4)]
+    # let all four finish
+    assert sorted(mi355.get(list(refs))) == [0, 1, 2, 3]
+    ready, not_ready = mi355.wait(refs, num_returns=1, timeout=5.0)
+    assert len(ready) == 1
+    assert len(not_ready) == 3
+    # reference drain-loop pattern terminates and sees every ref exactly once
+    seen, pending = [], list(refs)
+    while pending:
+        [done], pending = mi355.wait(pending, num_returns=1, timeout=5.0)
+        seen.append(mi355.get(done))
+    assert sorted(seen) == [0, 1, 2, 3]
