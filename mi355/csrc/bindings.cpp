@@ -45,6 +45,9 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
                                     double p_drop, c10::optional<torch::Tensor> seed,
                                     c10::optional<torch::Tensor> packed);
 torch::Tensor seg_postprocess(torch::Tensor logits, long Ho, long Wo);
+torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b, long splitk);
+torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b, long splitk);
+torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b, long splitk);
 std::vector<torch::Tensor> attn_fwd_v2(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                        c10::optional<torch::Tensor> bias,
                                        c10::optional<torch::Tensor> kpad, bool causal,
@@ -86,6 +89,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("kpad"), py::arg("causal"), py::arg("need_dbias"),
         py::arg("p_drop"), py::arg("seed"), py::arg("packed") = c10::nullopt);
   m.def("seg_postprocess", &seg_postprocess);
+  m.def("gemm_nt", &gemm_nt, py::arg("a"), py::arg("b"), py::arg("splitk") = 0);
+  m.def("gemm_nn", &gemm_nn, py::arg("a"), py::arg("b"), py::arg("splitk") = 0);
+  m.def("gemm_tn", &gemm_tn, py::arg("a"), py::arg("b"), py::arg("splitk") = 0);
   m.def("attn_fwd_v2", &attn_fwd_v2);
   m.def("attn_bwd_v2", &attn_bwd_v2);
   m.def("attn_bwd_v3", &attn_bwd_v3, py::arg("dout"), py::arg("q"), py::arg("k"),

@@ -1,0 +1,429 @@
+// Hand-written MFMA GEMM for CDNA4/gfx950 — the T5 hot-shape projection
+// GEMMs (replace hipBLASLt Cijk_* on the hot path).
+//
+// Three layouts, all C[M][N] = sum_k A'[m][k] * B'[k][n], fp32 accumulate:
+//   gemm_nt : A [M][K] k-contig,  B = W [N][K] k-contig   (fwd: y = x @ W^T)
+//   gemm_nn : A [M][Kc] k-contig, B = W [Kc][N] n-contig  (bwd: dx = dy @ W)
+//   gemm_tn : A [Kc][M] m-contig, B = X [Kc][N] n-contig  (bwd: dW = dy^T @ x)
+//
+// Design (cdna_hip_programming.md §5):
+//   * 128x128x64 tile, 4 waves (2x2), each wave 64x64 = 2x2 MFMA 32x32x16
+//     tiles; fp32 accumulators in AGPR-space (f32x16 per tile).
+//   * NT path: both operands staged HBM->LDS with global_load_lds_dwordx4
+//     (16 B, wave-uniform dest + lane*16), double-buffered, one
+//     vmcnt(0)+s_barrier per K-tile (the guide's 2-buffer overlap recipe);
+//     LDS slot-XOR swizzle (slot ^= row&7) applied on the per-lane SOURCE
+//     address and re-applied on the ds_read_b128 address (rule 21).
+//   * NN/TN path: register staging (load 8-elem vectors -> write tile t+1
+//     into LDS AFTER the consuming barrier), transposing in the write pass
+//     for contraction-major operands; +8-element row pad instead of swizzle.
+//   * Split-K over blockIdx.z for underfilled grids: fp32 partial slabs +
+//     a separate vectorized combine kernel (boundary ~1.5 us, simpler and
+//     placement-independent vs in-launch counters; revisit if profiles say).
+//   * XCD-aware bijective blockIdx remap, mt-fast so one XCD's chunk
+//     re-reads the same B panel from its L2 (T1).
+//
+// Requires M%128==0, N%128==0, K%64==0 — every T5-base/large hot shape
+// (M = B*S in {256,1024,8192}, N/K in {768,1536,2048,2304,4096,32128})
+// satisfies this; the Python dispatcher falls back to hipBLASLt otherwise.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "../common.h"
+
+namespace {
+
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+typedef __bf16 bf16x8v __attribute__((ext_vector_type(8)));
+typedef _Float16 f16x8v __attribute__((ext_vector_type(8)));
+
+template <typename T> struct GMF;
+template <> struct GMF<__hip_bfloat16> {
+  using frag = bf16x8v;
+  static DEVINL f32x16 mma(frag a, frag b, f32x16 c) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+  }
+};
+template <> struct GMF<__half> {
+  using frag = f16x8v;
+  static DEVINL f32x16 mma(frag a, frag b, f32x16 c) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_f16(a, b, c, 0, 0, 0);
+  }
+};
+
+constexpr int BM = 128, BN = 128, BK = 64;
+constexpr int PAD = 8;              // reg-staged image row pad (elements)
+constexpr int LDP = BK + PAD;       // reg-staged image row stride
+
+// Bijective XCD-aware remap (T1), mt-fast within each XCD chunk.
+DEVINL int xcd_remap(int b, int nwg) {
+  if (nwg < 16) return b;
+  const int q = nwg >> 3, r = nwg & 7;
+  const int xcd = b & 7, i = b >> 3;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+}
+
+// ---------------------------------------------------------------------- //
+// MFMA compute on one staged K-tile.                                     //
+// Image layout: SWZ=true -> linear [128][BK] with slot^=(row&7) swizzle  //
+// (glds path); SWZ=false -> padded [128][LDP] (register path).          //
+// ---------------------------------------------------------------------- //
+template <typename T, bool SWZ>
+DEVINL typename GMF<T>::frag frag_ld(const T* img, int row, int k8) {
+  // k8 = element offset of this lane's 8 contiguous k values (always 8-aligned)
+  if constexpr (SWZ) {
+    const int slot = (k8 >> 3) ^ (row & 7);
+    return *reinterpret_cast<const typename GMF<T>::frag*>(
+        img + row * BK + slot * 8);
+  } else {
+    return *reinterpret_cast<const typename GMF<T>::frag*>(
+        img + row * LDP + k8);
+  }
+}
+
+template <typename T, bool SWZ>
+DEVINL void mma_tile(const T* aimg, const T* bimg, int wr, int wc, int lane,
+                     f32x16 acc[2][2]) {
+  const int lq = lane & 31, half = lane >> 5;
+#pragma unroll
+  for (int ks = 0; ks < BK / 16; ++ks) {
+    const int k8 = ks * 16 + half * 8;
+    typename GMF<T>::frag a[2], b[2];
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi)
+      a[mi] = frag_ld<T, SWZ>(aimg, wr * 64 + mi * 32 + lq, k8);
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni)
+      b[ni] = frag_ld<T, SWZ>(bimg, wc * 64 + ni * 32 + lq, k8);
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+        acc[mi][ni] = GMF<T>::mma(a[mi], b[ni], acc[mi][ni]);
+  }
+}
+
+// Epilogue: C layout col = lane&31, row = (r&3) + 8*(r>>2) + 4*(lane>>5).
+template <typename T>
+DEVINL void epilogue(f32x16 acc[2][2], T* c, float* part, long N,
+                     int m0, int n0, int wr, int wc, int lane) {
+  const int lq = lane & 31, half = lane >> 5;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const int cb = n0 + wc * 64 + ni * 32 + lq;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int rb = m0 + wr * 64 + mi * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+        if (part != nullptr)
+          part[(long)rb * N + cb] = acc[mi][ni][r];
+        else
+          c[(long)rb * N + cb] = DT<T>::from_f32(acc[mi][ni][r]);
+      }
+    }
+}
+
+// ---------------------------------------------------------------------- //
+// NT kernel — glds double-buffered.                                      //
+// LDS: buf(32 KiB) x2 = {A 16 KiB, B 16 KiB}; one __shared__ object      //
+// (§5 trap 4a).  Each wave stages its 4 chunks of A and of B; chunk =    //
+// 8 rows x 64 k = 1 KiB = 64 lanes x 16 B.                               //
+// ---------------------------------------------------------------------- //
+template <typename T>
+__global__ __launch_bounds__(256, 2) void gemm_nt_kernel(
+    const T* __restrict__ a, const T* __restrict__ b, T* __restrict__ c,
+    float* __restrict__ part, int M, int N, int K, long lda, long ldb,
+    int MT, int NTt) {
+  __shared__ T lds[2 * 2 * BM * BK];  // [buf][A|B][128][64]
+
+  const int wg = xcd_remap(blockIdx.x, MT * NTt);
+  const int mt = wg % MT, nt = wg / MT;
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int wr = wid >> 1, wc = wid & 1;
+
+  // this z-slice's K range (whole tiles, balanced)
+  const int nkt = K / BK;
+  const int kt0 = (int)((long)blockIdx.z * nkt / gridDim.z);
+  const int kt1 = (int)((long)(blockIdx.z + 1) * nkt / gridDim.z);
+  if (kt1 == kt0) return;
+
+  const T* arow = a + (long)mt * BM * lda;
+  const T* brow = b + (long)nt * BN * ldb;
+
+  // per-lane staging geometry: chunk c' in 0..3 (wave-local), row/slot
+  const int srow = (lane >> 3);         // 0..7 within chunk
+  const int sslot = lane & 7;           // 16-B slot within row
+  auto stage = [&](int buf, int kt) {
+    const long k0 = (long)kt * BK;
+#pragma unroll
+    for (int cc = 0; cc < 4; ++cc) {
+      const int chunk = wid * 4 + cc;       // 0..15
+      const int row = chunk * 8 + srow;     // 0..127
+      const int slot = sslot ^ (row & 7);   // source pre-swizzle (rule 21)
+      const T* asrc = arow + (long)row * lda + k0 + slot * 8;
+      const T* bsrc = brow + (long)row * ldb + k0 + slot * 8;
+      T* adst = &lds[buf * 2 * BM * BK + chunk * (8 * BK)];
+      T* bdst = &lds[buf * 2 * BM * BK + BM * BK + chunk * (8 * BK)];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)asrc,
+          (__attribute__((address_space(3))) void*)adst, 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)bsrc,
+          (__attribute__((address_space(3))) void*)bdst, 16, 0, 0);
+    }
+  };
+
+  f32x16 acc[2][2];
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[mi][ni][r] = 0.f;
+
+  stage(0, kt0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  int buf = 0;
+  for (int kt = kt0; kt < kt1; ++kt) {
+    if (kt + 1 < kt1) stage(buf ^ 1, kt + 1);   // issue next-tile glds first
+    const T* aimg = &lds[buf * 2 * BM * BK];
+    const T* bimg = aimg + BM * BK;
+    __builtin_amdgcn_s_setprio(1);
+    mma_tile<T, true>(aimg, bimg, wr, wc, lane, acc);
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // next tile landed
+    __builtin_amdgcn_s_barrier();                     // + all reads of cur done
+    buf ^= 1;
+  }
+
+  float* pslab = (gridDim.z > 1)
+                     ? part + (long)blockIdx.z * M * (long)N
+                     : nullptr;
+  epilogue<T>(acc, c, pslab, N, mt * BM, nt * BN, wr, wc, lane);
+}
+
+// ---------------------------------------------------------------------- //
+// Register-staged kernel for contraction-major operands (NN: TB only;    //
+// TN: TA and TB).  TA/TB = operand stored [Kc][out] (transpose-stage).   //
+// Single LDS buffer, write-tile-after-barrier (T14 as §5's table says).  //
+// ---------------------------------------------------------------------- //
+template <typename T, bool TA, bool TB>
+__global__ __launch_bounds__(256, 2) void gemm_reg_kernel(
+    const T* __restrict__ a, const T* __restrict__ b, T* __restrict__ c,
+    float* __restrict__ part, int M, int N, int K, long lda, long ldb,
+    int MT, int NTt) {
+  __shared__ T lds[2 * BM * LDP];  // [A|B][128][72]
+
+  const int wg = xcd_remap(blockIdx.x, MT * NTt);
+  const int mt = wg % MT, nt = wg / MT;
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int wr = wid >> 1, wc = wid & 1;
+
+  const int nkt = K / BK;
+  const int kt0 = (int)((long)blockIdx.z * nkt / gridDim.z);
+  const int kt1 = (int)((long)(blockIdx.z + 1) * nkt / gridDim.z);
+  if (kt1 == kt0) return;
+
+  // ---- staging geometry ----
+  // row-staged operand ([out][k] k-contig): thread t -> out row t/8 (+32/pass),
+  //   16-B chunk t%8; 4 passes of 32 rows.
+  const int rr = threadIdx.x >> 3, rk = (threadIdx.x & 7) * 8;
+  // transpose-staged operand ([k][out] out-contig): thread t -> k row t/16
+  //   (+16/pass), 8-elem col chunk t%16; 4 passes of 16 k-rows.
+  const int tr = threadIdx.x >> 4, tc = (threadIdx.x & 15) * 8;
+
+  VecT<T, 8> sa[4], sb[4];
+  auto load_tile = [&](int kt) {
+    const long k0 = (long)kt * BK;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      if constexpr (!TA) {
+        sa[p] = vload<T, 8>(a + (long)(mt * BM + rr + 32 * p) * lda + k0 + rk);
+      } else {
+        sa[p] = vload<T, 8>(a + (k0 + tr + 16 * p) * lda + mt * BM + tc);
+      }
+      if constexpr (!TB) {
+        sb[p] = vload<T, 8>(b + (long)(nt * BN + rr + 32 * p) * ldb + k0 + rk);
+      } else {
+        sb[p] = vload<T, 8>(b + (k0 + tr + 16 * p) * ldb + nt * BN + tc);
+      }
+    }
+  };
+  auto write_tile = [&]() {
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      if constexpr (!TA) {
+        vstore<T, 8>(&lds[(rr + 32 * p) * LDP + rk], sa[p]);
+      } else {
+#pragma unroll
+        for (int u = 0; u < 8; ++u)
+          lds[(tc + u) * LDP + tr + 16 * p] = sa[p].data[u];
+      }
+      if constexpr (!TB) {
+        vstore<T, 8>(&lds[BM * LDP + (rr + 32 * p) * LDP + rk], sb[p]);
+      } else {
+#pragma unroll
+        for (int u = 0; u < 8; ++u)
+          lds[BM * LDP + (tc + u) * LDP + tr + 16 * p] = sb[p].data[u];
+      }
+    }
+  };
+
+  f32x16 acc[2][2];
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[mi][ni][r] = 0.f;
+
+  load_tile(kt0);
+  write_tile();
+  if (kt0 + 1 < kt1) load_tile(kt0 + 1);
+  __syncthreads();
+
+  for (int kt = kt0; kt < kt1; ++kt) {
+    __builtin_amdgcn_s_setprio(1);
+    mma_tile<T, false>(&lds[0], &lds[BM * LDP], wr, wc, lane, acc);
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();                     // everyone done reading tile kt
+    if (kt + 1 < kt1) {
+      write_tile();                      // tile kt+1 (T14: write after barrier)
+      if (kt + 2 < kt1) load_tile(kt + 2);
+      __syncthreads();
+    }
+  }
+
+  float* pslab = (gridDim.z > 1)
+                     ? part + (long)blockIdx.z * M * (long)N
+                     : nullptr;
+  epilogue<T>(acc, c, pslab, N, mt * BM, nt * BN, wr, wc, lane);
+}
+
+// Split-K combine: C = cast(sum_z part[z]); vectorized f32x4 / T x4.
+template <typename T>
+__global__ void gemm_splitk_combine(const float* __restrict__ part,
+                                    T* __restrict__ c, long MN, int Z) {
+  long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  if (i >= MN) return;
+  VecT<float, 4> s = vload<float, 4>(part + i);
+  for (int z = 1; z < Z; ++z) {
+    VecT<float, 4> p = vload<float, 4>(part + (long)z * MN + i);
+#pragma unroll
+    for (int u = 0; u < 4; ++u) s.data[u] += p.data[u];
+  }
+  VecT<T, 4> o;
+#pragma unroll
+  for (int u = 0; u < 4; ++u) o.data[u] = DT<T>::from_f32(s.data[u]);
+  vstore<T, 4>(c + i, o);
+}
+
+// ---------------------------------------------------------------------- //
+// Host launchers                                                         //
+// ---------------------------------------------------------------------- //
+enum class Layout { NT, NN, TN };
+
+int pick_splitk(long tiles, int nkt, int requested) {
+  if (requested > 0) return std::min(requested, nkt);
+  int z = 1;
+  // target >=512 blocks (2/CU) without making K-slices shorter than 3 tiles
+  while (tiles * z < 512 && (z * 2) * 3 <= nkt) z *= 2;
+  return z;
+}
+
+torch::Tensor gemm_run(Layout lay, torch::Tensor a, torch::Tensor b,
+                       long splitk) {
+  TORCH_CHECK(a.is_cuda() && b.is_cuda() && a.dim() == 2 && b.dim() == 2);
+  TORCH_CHECK(a.scalar_type() == b.scalar_type());
+  TORCH_CHECK(a.stride(1) == 1 && b.stride(1) == 1,
+              "gemm: operands must be row-contiguous");
+  long M, N, K, lda = a.stride(0), ldb = b.stride(0);
+  switch (lay) {
+    case Layout::NT:  // C[M,N] = A[M,K] @ B[N,K]^T
+      M = a.size(0); K = a.size(1); N = b.size(0);
+      TORCH_CHECK(b.size(1) == K, "gemm_nt: K mismatch");
+      break;
+    case Layout::NN:  // C[M,N] = A[M,Kc] @ B[Kc,N]
+      M = a.size(0); K = a.size(1); N = b.size(1);
+      TORCH_CHECK(b.size(0) == K, "gemm_nn: K mismatch");
+      break;
+    case Layout::TN:  // C[M,N] = A[Kc,M]^T @ B[Kc,N]
+      K = a.size(0); M = a.size(1); N = b.size(1);
+      TORCH_CHECK(b.size(0) == K, "gemm_tn: K mismatch");
+      break;
+  }
+  TORCH_CHECK(M % BM == 0 && N % BN == 0 && K % BK == 0,
+              "gemm: requires M%128==0, N%128==0, K%64==0 (got ", M, "x", N,
+              "x", K, ") — dispatcher must fall back");
+
+  auto c = torch::empty({M, N}, a.options());
+  const int MT = (int)(M / BM), NTt = (int)(N / BN), nkt = (int)(K / BK);
+  const int Z = pick_splitk((long)MT * NTt, nkt, (int)splitk);
+
+  torch::Tensor part;
+  float* part_p = nullptr;
+  if (Z > 1) {
+    part = torch::empty({Z, M, N}, a.options().dtype(torch::kFloat));
+    part_p = part.data_ptr<float>();
+  }
+  dim3 grid(MT * NTt, 1, Z);
+  auto stream = at::hip::getCurrentHIPStream().stream();
+
+#define GEMM_LAUNCH(T)                                                        \
+  do {                                                                        \
+    if (lay == Layout::NT)                                                    \
+      hipLaunchKernelGGL((gemm_nt_kernel<T>), grid, dim3(256), 0, stream,     \
+                         (const T*)a.data_ptr(), (const T*)b.data_ptr(),      \
+                         (T*)c.data_ptr(), part_p, (int)M, (int)N, (int)K,    \
+                         lda, ldb, MT, NTt);                                  \
+    else if (lay == Layout::NN)                                               \
+      hipLaunchKernelGGL((gemm_reg_kernel<T, false, true>), grid, dim3(256),  \
+                         0, stream, (const T*)a.data_ptr(),                   \
+                         (const T*)b.data_ptr(), (T*)c.data_ptr(), part_p,    \
+                         (int)M, (int)N, (int)K, lda, ldb, MT, NTt);          \
+    else                                                                      \
+      hipLaunchKernelGGL((gemm_reg_kernel<T, true, true>), grid, dim3(256),   \
+                         0, stream, (const T*)a.data_ptr(),                   \
+                         (const T*)b.data_ptr(), (T*)c.data_ptr(), part_p,    \
+                         (int)M, (int)N, (int)K, lda, ldb, MT, NTt);          \
+  } while (0)
+
+  if (a.scalar_type() == torch::kBFloat16) {
+    GEMM_LAUNCH(__hip_bfloat16);
+  } else if (a.scalar_type() == torch::kHalf) {
+    GEMM_LAUNCH(__half);
+  } else {
+    TORCH_CHECK(false, "gemm: bf16/f16 only");
+  }
+#undef GEMM_LAUNCH
+  HIP_CHECK_KERNEL();
+
+  if (Z > 1) {
+    const long MN = M * N;
+    const long thr = MN / 4;
+    auto launch_combine = [&](auto tptr) {
+      using T = std::remove_pointer_t<decltype(tptr)>;
+      hipLaunchKernelGGL((gemm_splitk_combine<T>), dim3((thr + 255) / 256),
+                         dim3(256), 0, stream, part_p, (T*)c.data_ptr(), MN, Z);
+    };
+    if (a.scalar_type() == torch::kBFloat16) launch_combine((__hip_bfloat16*)nullptr);
+    else launch_combine((__half*)nullptr);
+    HIP_CHECK_KERNEL();
+  }
+  return c;
+}
+
+}  // namespace
+
+torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b, long splitk) {
+  return gemm_run(Layout::NT, a, b, splitk);
+}
+torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b, long splitk) {
+  return gemm_run(Layout::NN, a, b, splitk);
+}
+torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b, long splitk) {
+  return gemm_run(Layout::TN, a, b, splitk);
+}

