@@ -68,6 +68,12 @@ DEVINL int xcd_remap(int b, int nwg) {
 // Image layout: SWZ=true -> linear [128][BK] with slot^=(row&7) swizzle  //
 // (glds path); SWZ=false -> padded [128][LDP] (register path).          //
 // ---------------------------------------------------------------------- //
+// Image addressing.  SWZ=1 (glds path): linear [128][BK], 16-B slot XOR'd
+// with row&7 (applied to the glds SOURCE address and re-applied here —
+// rule 21).  SWZ=0 (register path): padded [128][LDP] rows, slot XOR'd with
+// (row>>3)&7 — the transpose-stage writes walk a COLUMN (row varies by 8
+// per element), so the key must come from row's bit 3..5 or all 16 lanes
+// of a write group land on one bank (16-way, the r1 TN kernel's killer).
 template <typename T, bool SWZ>
 DEVINL typename GMF<T>::frag frag_ld(const T* img, int row, int k8) {
   // k8 = element offset of this lane's 8 contiguous k values (always 8-aligned)
@@ -76,8 +82,9 @@ DEVINL typename GMF<T>::frag frag_ld(const T* img, int row, int k8) {
     return *reinterpret_cast<const typename GMF<T>::frag*>(
         img + row * BK + slot * 8);
   } else {
+    const int slot = (k8 >> 3) ^ ((row >> 3) & 7);
     return *reinterpret_cast<const typename GMF<T>::frag*>(
-        img + row * LDP + k8);
+        img + row * LDP + slot * 8);
   }
 }
 
@@ -103,24 +110,51 @@ DEVINL void mma_tile(const T* aimg, const T* bimg, int wr, int wc, int lane,
   }
 }
 
-// Epilogue: C layout col = lane&31, row = (r&3) + 8*(r>>2) + 4*(lane>>5).
+// Epilogue.  The MFMA C layout (col = lane&31, row = (r&3)+8*(r>>2)+
+// 4*(lane>>5)) gives each lane one COLUMN — direct stores are 64 scalar
+// u16/u32 per thread, store-issue-bound (guide T21).  Repack each 32x32
+// tile through a per-wave 4 KiB LDS scratch (the staging buffers are dead
+// by now) and store row-major b128: 8x fewer store instructions.
 template <typename T>
 DEVINL void epilogue(f32x16 acc[2][2], T* c, float* part, long N,
-                     int m0, int n0, int wr, int wc, int lane) {
+                     int m0, int n0, int wr, int wc, int lane,
+                     float* scratch /* per-wave, 1024 floats */) {
   const int lq = lane & 31, half = lane >> 5;
 #pragma unroll
   for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
     for (int ni = 0; ni < 2; ++ni) {
-      const int cb = n0 + wc * 64 + ni * 32 + lq;
+      const int rt = m0 + wr * 64 + mi * 32;   // tile row base in C
+      const int ct = n0 + wc * 64 + ni * 32;   // tile col base in C
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int rb = m0 + wr * 64 + mi * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
-        if (part != nullptr)
-          part[(long)rb * N + cb] = acc[mi][ni][r];
-        else
-          c[(long)rb * N + cb] = DT<T>::from_f32(acc[mi][ni][r]);
+      for (int r = 0; r < 16; ++r)
+        scratch[((r & 3) + 8 * (r >> 2) + 4 * half) * 32 + lq] = acc[mi][ni][r];
+      // compiler inserts the lgkmcnt wait between same-wave write/read
+      if (part != nullptr) {
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const int unit = lane + 64 * u;          // 256 units: row, 4-col chunk
+          const int row = unit >> 3, c4 = (unit & 7) * 4;
+          VecT<float, 4> v = vload<float, 4>(scratch + row * 32 + c4);
+          vstore<float, 4>(part + (long)(rt + row) * N + ct + c4, v);
+        }
+      } else {
+#pragma unroll
+        for (int u = 0; u < 2; ++u) {
+          const int unit = lane + 64 * u;          // 128 units: row, 8-col chunk
+          const int row = unit >> 2, c8 = (unit & 3) * 8;
+          VecT<float, 4> lo = vload<float, 4>(scratch + row * 32 + c8);
+          VecT<float, 4> hi = vload<float, 4>(scratch + row * 32 + c8 + 4);
+          VecT<T, 8> o;
+#pragma unroll
+          for (int q = 0; q < 4; ++q) {
+            o.data[q] = DT<T>::from_f32(lo.data[q]);
+            o.data[4 + q] = DT<T>::from_f32(hi.data[q]);
+          }
+          vstore<T, 8>(c + (long)(rt + row) * N + ct + c8, o);
+        }
       }
+      // serialize scratch reuse across the 4 tiles (same wave: lgkm order)
     }
 }
 
@@ -202,7 +236,8 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_kernel(
   float* pslab = (gridDim.z > 1)
                      ? part + (long)blockIdx.z * M * (long)N
                      : nullptr;
-  epilogue<T>(acc, c, pslab, N, mt * BM, nt * BN, wr, wc, lane);
+  epilogue<T>(acc, c, pslab, N, mt * BM, nt * BN, wr, wc, lane,
+              reinterpret_cast<float*>(&lds[0]) + wid * 1024);
 }
 
 // ---------------------------------------------------------------------- //
@@ -252,22 +287,26 @@ __global__ __launch_bounds__(256, 2) void gemm_reg_kernel(
       }
     }
   };
+  // swizzled image address for element (row, k) — must match frag_ld<.,false>
+  auto swz = [](int row, int k) {
+    return row * LDP + ((((k >> 3) ^ ((row >> 3) & 7))) << 3) + (k & 7);
+  };
   auto write_tile = [&]() {
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
       if constexpr (!TA) {
-        vstore<T, 8>(&lds[(rr + 32 * p) * LDP + rk], sa[p]);
+        vstore<T, 8>(&lds[swz(rr + 32 * p, rk)], sa[p]);
       } else {
 #pragma unroll
         for (int u = 0; u < 8; ++u)
-          lds[(tc + u) * LDP + tr + 16 * p] = sa[p].data[u];
+          lds[swz(tc + u, tr + 16 * p)] = sa[p].data[u];
       }
       if constexpr (!TB) {
-        vstore<T, 8>(&lds[BM * LDP + (rr + 32 * p) * LDP + rk], sb[p]);
+        vstore<T, 8>(&lds[BM * LDP + swz(rr + 32 * p, rk)], sb[p]);
       } else {
 #pragma unroll
         for (int u = 0; u < 8; ++u)
-          lds[BM * LDP + (tc + u) * LDP + tr + 16 * p] = sb[p].data[u];
+          lds[BM * LDP + swz(tc + u, tr + 16 * p)] = sb[p].data[u];
       }
     }
   };
@@ -300,7 +339,8 @@ __global__ __launch_bounds__(256, 2) void gemm_reg_kernel(
   float* pslab = (gridDim.z > 1)
                      ? part + (long)blockIdx.z * M * (long)N
                      : nullptr;
-  epilogue<T>(acc, c, pslab, N, mt * BM, nt * BN, wr, wc, lane);
+  epilogue<T>(acc, c, pslab, N, mt * BM, nt * BN, wr, wc, lane,
+              reinterpret_cast<float*>(&lds[0]) + wid * 1024);
 }
 
 // Split-K combine: C = cast(sum_z part[z]); vectorized f32x4 / T x4.
