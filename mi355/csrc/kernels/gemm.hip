@@ -343,6 +343,148 @@ __global__ __launch_bounds__(256, 2) void gemm_reg_kernel(
               reinterpret_cast<float*>(&lds[0]) + wid * 1024);
 }
 
+// ---------------------------------------------------------------------- //
+// NT 256x256 8-phase kernel (guide §5 "256² 8-phase template", adapted    //
+// to 32x32x16 MFMA).  8 waves (2M x 4N), per-wave 128x64 = 4x2 tiles of   //
+// 32², BK=64 in 4 k-subtile phases.                                       //
+//                                                                         //
+// Staging: k-slice-major LDS images — A: [4 ks][256 rows][2 slots x 8],   //
+// B likewise (32 KiB each, x2 buffers = 128 KiB).  Phase p reads only     //
+// k-slice p, so slice p+1..3 of the CURRENT tile and slices 0..p-1 of     //
+// the NEXT stay in flight across every phase boundary: steady-state wait  //
+// is a counted s_waitcnt vmcnt(6), never 0, per the T3+T4 recipe.  Slot   //
+// parity is XOR'd with row bit 3 (source-side, rule 21) which makes the   //
+// b128 fragment reads bank-conflict-free.                                 //
+// Waves pair via two raw s_barriers per phase; the MFMA cluster sits in   //
+// s_setprio(1) (T5).  M and N may be non-multiples of the tile: source    //
+// rows clamp, stores bound-check (lm_head's N=32128 = 125.5 tiles).       //
+// ---------------------------------------------------------------------- //
+template <typename T>
+__global__ __launch_bounds__(512, 2) void gemm_nt256_kernel(
+    const T* __restrict__ a, const T* __restrict__ b, T* __restrict__ c,
+    int M, int N, int K, long lda, long ldb, int MT, int NTt) {
+  constexpr int SM = 256;              // tile M = tile N
+  constexpr int SLICE = SM * 16;       // elems per k-slice image (256 x 16)
+  __shared__ T lds[2 * 2 * SM * BK];   // [buf][A|B][4 ks][256][16]
+
+  const int wg = xcd_remap(blockIdx.x, MT * NTt);
+  const int mt = wg % MT, nt = wg / MT;
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int wr = wid >> 2, wc = wid & 3;   // 2 x 4 wave grid
+  const int lq = lane & 31, half = lane >> 5;
+
+  const int nkt = K / BK;
+  const int m0 = mt * SM, n0 = nt * SM;
+
+  // glds: slice s, wave w stages A rows 32w..32w+31 and B rows likewise.
+  // lane l -> row 32w + l/2, slot l&1; chunk dest is lane-linear.
+  const int srow = lane >> 1, sslot = lane & 1;
+  auto stage_slice = [&](int buf, int kt, int s) {
+    const long k0 = (long)kt * BK + s * 16;
+    const int arow = min(m0 + wid * 32 + srow, M - 1);
+    const int brow = min(n0 + wid * 32 + srow, N - 1);
+    const int aslot = sslot ^ ((srow >> 3) & 1);
+    const T* asrc = a + (long)arow * lda + k0 + aslot * 8;
+    const T* bsrc = b + (long)brow * ldb + k0 + aslot * 8;
+    T* adst = &lds[buf * 2 * SM * BK + s * SLICE + (wid * 32) * 16];
+    T* bdst = &lds[buf * 2 * SM * BK + SM * BK + s * SLICE + (wid * 32) * 16];
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)asrc,
+        (__attribute__((address_space(3))) void*)adst, 16, 0, 0);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)bsrc,
+        (__attribute__((address_space(3))) void*)bdst, 16, 0, 0);
+  };
+
+  f32x16 acc[4][2];
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[mi][ni][r] = 0.f;
+
+  // prologue: all 4 slices of tile 0
+#pragma unroll
+  for (int s = 0; s < 4; ++s) stage_slice(0, 0, s);
+
+  int buf = 0;
+  for (int kt = 0; kt < nkt; ++kt) {
+    const bool has_next = kt + 1 < nkt;
+    const T* aimg = &lds[buf * 2 * SM * BK];
+    const T* bimg = aimg + SM * BK;
+
+#define GEMM256_PHASE(P, VM_TAIL)                                             \
+    do {                                                                      \
+      if (has_next) {                                                         \
+        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");                      \
+      } else {                                                                \
+        asm volatile("s_waitcnt vmcnt(" #VM_TAIL ")" ::: "memory");           \
+      }                                                                       \
+      __builtin_amdgcn_s_barrier();                                           \
+      if (has_next) stage_slice(buf ^ 1, kt + 1, P);                          \
+      typename GMF<T>::frag af[4], bf[2];                                     \
+      _Pragma("unroll")                                                       \
+      for (int mi = 0; mi < 4; ++mi) {                                        \
+        const int row = wr * 128 + mi * 32 + lq;                              \
+        af[mi] = *reinterpret_cast<const typename GMF<T>::frag*>(             \
+            aimg + P * SLICE + row * 16 + (half ^ ((row >> 3) & 1)) * 8);     \
+      }                                                                       \
+      _Pragma("unroll")                                                       \
+      for (int ni = 0; ni < 2; ++ni) {                                        \
+        const int row = wc * 64 + ni * 32 + lq;                               \
+        bf[ni] = *reinterpret_cast<const typename GMF<T>::frag*>(             \
+            bimg + P * SLICE + row * 16 + (half ^ ((row >> 3) & 1)) * 8);     \
+      }                                                                       \
+      __builtin_amdgcn_s_setprio(1);                                          \
+      _Pragma("unroll")                                                       \
+      for (int mi = 0; mi < 4; ++mi)                                          \
+        _Pragma("unroll")                                                     \
+        for (int ni = 0; ni < 2; ++ni)                                        \
+          acc[mi][ni] = GMF<T>::mma(af[mi], bf[ni], acc[mi][ni]);             \
+      __builtin_amdgcn_s_setprio(0);                                          \
+      __builtin_amdgcn_s_barrier();                                           \
+    } while (0)
+
+    GEMM256_PHASE(0, 6);
+    GEMM256_PHASE(1, 4);
+    GEMM256_PHASE(2, 2);
+    GEMM256_PHASE(3, 0);
+#undef GEMM256_PHASE
+    buf ^= 1;
+  }
+
+  // epilogue: per-wave LDS repack (8 waves x 4 KiB = 32 KiB of the dead
+  // staging buffers), b128 stores, bounds-checked for partial edge tiles.
+  float* scratch = reinterpret_cast<float*>(&lds[0]) + wid * 1024;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const int rt = m0 + wr * 128 + mi * 32;
+      const int ct = n0 + wc * 64 + ni * 32;
+#pragma unroll
+      for (int r = 0; r < 16; ++r)
+        scratch[((r & 3) + 8 * (r >> 2) + 4 * half) * 32 + lq] = acc[mi][ni][r];
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        const int unit = lane + 64 * u;
+        const int row = unit >> 2, c8 = (unit & 3) * 8;
+        if (rt + row < M && ct + c8 < N) {
+          VecT<float, 4> lo = vload<float, 4>(scratch + row * 32 + c8);
+          VecT<float, 4> hi = vload<float, 4>(scratch + row * 32 + c8 + 4);
+          VecT<T, 8> o;
+#pragma unroll
+          for (int q = 0; q < 4; ++q) {
+            o.data[q] = DT<T>::from_f32(lo.data[q]);
+            o.data[4 + q] = DT<T>::from_f32(hi.data[q]);
+          }
+          vstore<T, 8>(c + (long)(rt + row) * N + ct + c8, o);
+        }
+      }
+    }
+}
+
 // Split-K combine: C = cast(sum_z part[z]); vectorized f32x4 / T x4.
 template <typename T>
 __global__ void gemm_splitk_combine(const float* __restrict__ part,
@@ -400,6 +542,35 @@ torch::Tensor gemm_run(Layout lay, torch::Tensor a, torch::Tensor b,
               "x", K, ") — dispatcher must fall back");
 
   auto c = torch::empty({M, N}, a.options());
+
+  // NT fill shapes: the 256x256 8-phase kernel (counted-vmcnt pipeline)
+  // when the 256-tile grid still fills the 256 CUs; override with
+  // MI355_GEMM_TILE=128|256 for A/B.
+  if (lay == Layout::NT) {
+    const int MT2 = (int)((M + 255) / 256), NT2 = (int)((N + 255) / 256);
+    const char* ft = std::getenv("MI355_GEMM_TILE");
+    const bool use256 = ft ? (atoi(ft) == 256)
+                           : ((long)MT2 * NT2 >= 224 && splitk <= 1);
+    if (use256) {
+      auto stream = at::hip::getCurrentHIPStream().stream();
+      dim3 grid(MT2 * NT2);
+      if (a.scalar_type() == torch::kBFloat16)
+        hipLaunchKernelGGL((gemm_nt256_kernel<__hip_bfloat16>), grid,
+                           dim3(512), 0, stream,
+                           (const __hip_bfloat16*)a.data_ptr(),
+                           (const __hip_bfloat16*)b.data_ptr(),
+                           (__hip_bfloat16*)c.data_ptr(), (int)M, (int)N,
+                           (int)K, lda, ldb, MT2, NT2);
+      else
+        hipLaunchKernelGGL((gemm_nt256_kernel<__half>), grid, dim3(512), 0,
+                           stream, (const __half*)a.data_ptr(),
+                           (const __half*)b.data_ptr(), (__half*)c.data_ptr(),
+                           (int)M, (int)N, (int)K, lda, ldb, MT2, NT2);
+      HIP_CHECK_KERNEL();
+      return c;
+    }
+  }
+
   const int MT = (int)(M / BM), NTt = (int)(N / BN), nkt = (int)(K / BK);
   const int Z = pick_splitk((long)MT * NTt, nkt, (int)splitk);
 

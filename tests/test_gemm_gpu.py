@@ -108,3 +108,22 @@ def test_mfma_linear_autograd_matches_torch():
     assert _relerr(y, y32) < 3e-2
     assert _relerr(got_dx, x32.grad) < 3e-2
     assert _relerr(got_dw, w32.grad) < 3e-2
+
+
+@gpu
+@pytest.mark.parametrize("shape", [
+    (256, 256, 64),        # single 256-tile
+    (1024, 2304, 768),     # qkv
+    (1024, 32128, 768),    # lm_head: partial N tile (32128 = 125.5 * 256)
+    (512, 1280, 128),      # M,N%256 != 0 -> clamped edge tiles
+])
+def test_gemm_nt256_matches_fp32_ref(shape, monkeypatch):
+    """The 256x256 8-phase kernel, forced via MI355_GEMM_TILE=256."""
+    monkeypatch.setenv("MI355_GEMM_TILE", "256")
+    M, N, K = shape
+    torch.manual_seed(5)
+    a = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
+    got = ops.gemm_nt(a, b)
+    ref = a.float() @ b.float().t()
+    assert _relerr(got, ref) < 3e-2
