@@ -431,7 +431,10 @@ class T5ForConditionalGeneration(nn.Module):
                            enc_hidden=encoder_outputs, enc_kpad=kpad)
         if self.config.tie_word_embeddings:
             dec = dec * (self.model_dim**-0.5)
-        logits = self.lm_head(dec)
+        # route through ops.linear so the per-shape MFMA GEMM dispatch applies
+        # (at B=2/S=512 the lm_head backward-dX runs on the hand-written
+        # kernel: 505 vs 291 TF — profiles/r02_gemm_ab.md)
+        logits = ops.linear(dec, self.lm_head.weight)
         loss = None
         if labels is not None:
             loss = ops.cross_entropy(logits, labels, ignore_index=-100)

@@ -18,10 +18,39 @@ from __future__ import annotations
 
 import os
 
-# (op, rows, cols, kc) entries where the HIP kernel won the A/B.
-# Populated from profiles/r02_gemm_ab.md; start empty — before the first
-# hardware A/B only MI355_GEMM=hip engages the kernel.
-_WINS: set[tuple[str, int, int, int]] = set()
+# (op, rows, cols, kc) entries where the HIP kernel won the A/B by >=2%
+# (profiles/r02_gemm_ab.md, measured on MI355X; tools/gemm_ab.py regenerates).
+# Pattern: the hand-written kernel wins every latency/underfill shape — the
+# B=2-per-GPU training step (M=1024) attention-out and wo projections, every
+# decode-step projection (M=256), and ALL dW GEMMs at decode plus the
+# N=768-out dW at train M — while hipBLASLt keeps the chip-filling shapes
+# (M=8192 forward panels, lm_head fwd) where its tuned 1+ PF kernels win.
+_WINS: set[tuple[str, int, int, int]] = {
+    ('nn', 256, 768, 768),
+    ('nn', 256, 768, 1536),
+    ('nn', 256, 768, 2304),
+    ('nn', 256, 768, 4096),
+    ('nn', 256, 768, 32128),
+    ('nn', 256, 2048, 768),
+    ('nn', 1024, 768, 768),
+    ('nn', 1024, 768, 32128),
+    ('nt', 256, 768, 768),
+    ('nt', 256, 768, 2048),
+    ('nt', 256, 1536, 768),
+    ('nt', 256, 2304, 768),
+    ('nt', 256, 4096, 768),
+    ('nt', 1024, 768, 768),
+    ('nt', 1024, 768, 2048),
+    ('tn', 768, 768, 256),
+    ('tn', 768, 768, 1024),
+    ('tn', 768, 768, 8192),
+    ('tn', 768, 2048, 256),
+    ('tn', 768, 2048, 8192),
+    ('tn', 1536, 768, 256),
+    ('tn', 1536, 768, 8192),
+    ('tn', 2304, 768, 256),
+    ('tn', 4096, 768, 256),
+}
 
 
 def _dims_ok(op: str, rows: int, cols: int, kc: int) -> bool:
