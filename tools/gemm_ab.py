@@ -51,18 +51,36 @@ def run_op(op: str, rows: int, cols: int, kc: int, dtype, hip: bool):
     return f
 
 
-def time_fn(f, iters: int) -> float:
+def time_fn(f, iters: int, graph: bool = True) -> float:
+    """Median-of-3 timing.  graph=True (default) captures `iters` calls in a
+    hipGraph and times the replay — the representative regime: the train/infer
+    steps run graph-captured, where launch overhead (hipBLASLt's ~18us eager
+    floor) vanishes and only kernel time + graph-internal boundaries remain.
+    The first eager A/B (profiles/, r02 ab1-ab3) mismeasured exactly this."""
     for _ in range(5):
         f()
     torch.cuda.synchronize()
-    s = torch.cuda.Event(enable_timing=True)
-    e = torch.cuda.Event(enable_timing=True)
-    s.record()
-    for _ in range(iters):
-        f()
-    e.record()
-    torch.cuda.synchronize()
-    return s.elapsed_time(e) / iters * 1e3  # us
+    if graph:
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            for _ in range(iters):
+                f()
+        run = g.replay
+    else:
+        def run():
+            for _ in range(iters):
+                f()
+    best = []
+    for _ in range(3):
+        torch.cuda.synchronize()
+        s = torch.cuda.Event(enable_timing=True)
+        e = torch.cuda.Event(enable_timing=True)
+        s.record()
+        run()
+        e.record()
+        torch.cuda.synchronize()
+        best.append(s.elapsed_time(e) / iters * 1e3)  # us
+    return sorted(best)[1]
 
 
 def main():
