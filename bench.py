@@ -235,6 +235,19 @@ def main():
     if args.mode == "train":
         elapsed, extra = run_train(args, device, dtype)
         metric = "train_samples_per_s"
+        # BASELINE's metric is "fine-tune samples/sec + batch-infer seq/sec":
+        # measure the inference half too (batch-256 fp16 generate, the
+        # reference's cell-64/67 config) so the seq/s number is produced by
+        # the same driver-run command.  GPU-only —
+        # a 256x128-token generate on CPU would take minutes.
+        if device.type == "cuda" and os.environ.get("MI355_BENCH_INFER", "1") != "0":
+            iargs = argparse.Namespace(**vars(args))
+            iargs.batch, iargs.steps, iargs.warmup = 256, 3, 1
+            ielapsed, _ = run_infer(iargs, device, torch.float16)
+            extra = dict(extra)
+            extra["_infer_seq_per_s"] = round(256 * n * iargs.steps / ielapsed, 1)
+            extra["_infer_config"] = {"batch": 256, "max_new_tokens": 128,
+                                      "dtype": "fp16", "model": args.model}
     else:
         elapsed, extra = run_infer(args, device, dtype)
         metric = "infer_seq_per_s"
@@ -242,7 +255,14 @@ def main():
     ms_per_step = elapsed / args.steps * 1000.0
     value = args.batch * n * args.steps / elapsed
 
+    infer_seq_per_s = extra.pop("_infer_seq_per_s", None) if isinstance(extra, dict) else None
+    infer_config = extra.pop("_infer_config", None) if isinstance(extra, dict) else None
+
     if rank == 0:
+        extra_top = {}
+        if infer_seq_per_s is not None:
+            extra_top = {"infer_seq_per_s": infer_seq_per_s,
+                         "infer_config": infer_config}
         print(json.dumps({
             "metric": metric,
             "value": round(value, 3),
@@ -263,6 +283,7 @@ def main():
                 "parallelism": f"dp{n}",
                 **extra,
             },
+            **extra_top,
         }))
 
 

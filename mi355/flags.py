@@ -43,6 +43,16 @@ FLAGS = [
     Flag("MI355_HYBRID_GRAPH", "0",
          "Force the two-graph hybrid step at world_size 1 (validation knob)."),
     # -- kernels -------------------------------------------------------------
+    Flag("MI355_GEMM", "auto",
+         "Projection-GEMM dispatch: auto = per-shape table from the "
+         "graph-captured A/B (profiles/r02_gemm_ab.md), hip = always the "
+         "hand-written MFMA kernels, blaslt = always the library."),
+    Flag("MI355_GEMM_TILE", "<auto>",
+         "Force the NT MFMA GEMM tile (128 or 256); auto picks 256 (8-phase "
+         "counted-vmcnt kernel) when the 256-tile grid still fills the CUs."),
+    Flag("MI355_BENCH_INFER", "1",
+         "bench.py train mode also measures batch-256 fp16 generate and "
+         "reports infer_seq_per_s in the same JSON line (GPU only)."),
     Flag("MI355_GEMM_2S", "0",
          "1 = two-stream linear backward (dW on a side stream). Off by "
          "default: measured slower at B=2 (event overhead)."),

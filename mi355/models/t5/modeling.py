@@ -105,8 +105,41 @@ def _register_fused_statedict_compat(parent: nn.Module, attr: str):
     parent._register_load_state_dict_pre_hook(load_pre_hook)
 
 
+def _register_cross_kv_compat(stack: nn.Module, n_layers: int, inner: int):
+    """HF key compatibility for the decoder-level fused cross-kv parameter:
+    on save, split ``cross_kv`` into the per-layer
+    ``block.{i}.layer.1.EncDecAttention.{k,v}.weight`` slices; on load, merge
+    those HF keys back into the one fused weight."""
+
+    def save_hook(module, state_dict, prefix, local_metadata):
+        key = f"{prefix}cross_kv"
+        if key in state_dict:
+            w = state_dict.pop(key)
+            for i in range(n_layers):
+                blk = w[i * 2 * inner:(i + 1) * 2 * inner]
+                base = f"{prefix}block.{i}.layer.1.EncDecAttention"
+                state_dict[f"{base}.k.weight"] = blk[:inner]
+                state_dict[f"{base}.v.weight"] = blk[inner:]
+        return state_dict
+
+    def load_pre_hook(state_dict, prefix, local_metadata, strict, missing_keys,
+                      unexpected_keys, error_msgs):
+        pairs = []
+        for i in range(n_layers):
+            base = f"{prefix}block.{i}.layer.1.EncDecAttention"
+            pairs.append((f"{base}.k.weight", f"{base}.v.weight"))
+        if all(k in state_dict and v in state_dict for k, v in pairs):
+            state_dict[f"{prefix}cross_kv"] = torch.cat(
+                [torch.cat([state_dict.pop(k), state_dict.pop(v)], dim=0)
+                 for k, v in pairs], dim=0)
+
+    stack._register_state_dict_hook(save_hook)
+    stack._register_load_state_dict_pre_hook(load_pre_hook)
+
+
 class T5Attention(nn.Module):
-    def __init__(self, config: T5Config, has_relative_attention_bias: bool, is_decoder: bool):
+    def __init__(self, config: T5Config, has_relative_attention_bias: bool, is_decoder: bool,
+                 kv_external: bool = False):
         super().__init__()
         self.is_decoder = is_decoder
         self.has_relative_attention_bias = has_relative_attention_bias
@@ -116,7 +149,15 @@ class T5Attention(nn.Module):
         self.d_kv = config.d_kv
         self.n_heads = config.num_heads
         self.inner_dim = self.n_heads * self.d_kv
-        self.qkv = _FusedLinears(["q", "k", "v"], self.d_model, self.inner_dim)
+        # kv_external (decoder cross-attention): the k/v weights of ALL
+        # layers live in ONE decoder-level parameter (T5Stack.cross_kv) so
+        # the 12 per-layer kv projections over the SAME encoder output become
+        # a single (B*S, 12*2*inner) GEMM per forward; this module keeps only
+        # its q projection (HF state-dict keys are preserved by the stack's
+        # save/load hooks).
+        self.kv_external = kv_external
+        names = ["q"] if kv_external else ["q", "k", "v"]
+        self.qkv = _FusedLinears(names, self.d_model, self.inner_dim)
         _register_fused_statedict_compat(self, "qkv")
         self.o = nn.Linear(self.inner_dim, self.d_model, bias=False)
         if has_relative_attention_bias:
@@ -157,10 +198,12 @@ class T5Attention(nn.Module):
         return x.view(b, -1, self.n_heads, self.d_kv).transpose(1, 2)
 
     def forward(self, hidden, kv_hidden=None, position_bias=None, kpad=None,
-                causal=False, kv_cache=None):
+                causal=False, kv_cache=None, kv5=None):
         """kv_cache: optional dict used by generate() —
         {"k","v": (B,H,Tmax,D) preallocated, "len": int} for self-attn append,
-        or {"k","v" precomputed static} for cross-attn."""
+        or {"k","v" precomputed static} for cross-attn.
+        kv5: (B, Sk, 2, H, D) precomputed packed kv (decoder cross-attention
+        with kv_external: one stack-level GEMM feeds every layer)."""
         b, sq, _ = hidden.shape
         static = kv_cache is not None and kv_cache.get("static", False)
         if kv_hidden is None and kv_cache is None:
@@ -192,12 +235,15 @@ class T5Attention(nn.Module):
             qs = self._shape(self.q(hidden), b)
             ks, vs = kv_cache["k"], kv_cache["v"]
         else:
-            # cross-attention (training): ONE fused kv GEMM over the last two
-            # thirds of the qkv weight, consumed packed (one dkv grad, and a
-            # single contiguous weight-slice grad instead of two)
+            # cross-attention (training): packed kv — either precomputed at
+            # the stack level (kv_external, one GEMM for all 12 layers) or a
+            # per-layer fused kv GEMM over the last two thirds of qkv
             qs = self._shape(self.q(hidden), b)
-            kv5 = ops.linear(kv_hidden, self.qkv.weight[self.inner_dim:]).view(
-                b, -1, 2, self.n_heads, self.d_kv)
+            if kv5 is None:
+                assert not self.kv_external, \
+                    "kv_external cross-attention needs the stack-level kv5"
+                kv5 = ops.linear(kv_hidden, self.qkv.weight[self.inner_dim:]).view(
+                    b, -1, 2, self.n_heads, self.d_kv)
             p_drop = self.dropout if self.training else 0.0
             out = ops.attention_kv_packed(qs, kv5, bias=position_bias,
                                           kpad=kpad, p_drop=p_drop)
@@ -235,13 +281,15 @@ class T5LayerSelfAttention(nn.Module):
 class T5LayerCrossAttention(nn.Module):
     def __init__(self, config):
         super().__init__()
-        self.EncDecAttention = T5Attention(config, False, is_decoder=True)
+        self.EncDecAttention = T5Attention(config, False, is_decoder=True,
+                                           kv_external=True)
         self.layer_norm = T5LayerNorm(config.d_model, config.layer_norm_epsilon)
         self.dropout = nn.Dropout(config.dropout_rate)
 
-    def forward(self, hidden, enc_hidden, kpad=None, kv_cache=None):
+    def forward(self, hidden, enc_hidden, kpad=None, kv_cache=None, kv5=None):
         normed = self.layer_norm(hidden)
-        att = self.EncDecAttention(normed, kv_hidden=enc_hidden, kpad=kpad, kv_cache=kv_cache)
+        att = self.EncDecAttention(normed, kv_hidden=enc_hidden, kpad=kpad,
+                                   kv_cache=kv_cache, kv5=kv5)
         return hidden + self.dropout(att)
 
 
@@ -306,6 +354,16 @@ class T5Stack(nn.Module):
         )
         self.final_layer_norm = T5LayerNorm(config.d_model, config.layer_norm_epsilon)
         self.dropout = nn.Dropout(config.dropout_rate)
+        if is_decoder:
+            # Cross-layer fused kv projection: every decoder layer's
+            # cross-attention projects the SAME encoder output, so the 12
+            # per-layer (M, 2*inner) GEMMs collapse into one
+            # (M, n*2*inner) GEMM per forward (round-1 ROADMAP item;
+            # SURVEY K5).  Layout: rows [layer][k|v][inner]; the save/load
+            # hooks below keep the HF per-layer k/v key names.
+            inner = config.num_heads * config.d_kv
+            self.cross_kv = nn.Parameter(torch.empty(n * 2 * inner, config.d_model))
+            _register_cross_kv_compat(self, n, inner)
 
     def forward(self, input_ids=None, inputs_embeds=None, kpad=None, enc_hidden=None,
                 enc_kpad=None, caches=None, past_len: int = 0):
@@ -333,6 +391,17 @@ class T5Stack(nn.Module):
             return ops.add_rms_norm(resid, delta, ln.weight, ln.variance_epsilon,
                                     p_drop=pdrop)
 
+        # one fused kv projection for every layer's cross-attention (the
+        # unbind keeps the backward a single stack instead of 12 per-layer
+        # zero+scatter slice grads)
+        kv_layers = None
+        if self.is_decoder and enc_hidden is not None and caches is None:
+            h, dkv = self.config.num_heads, self.config.d_kv
+            be, se, _ = enc_hidden.shape
+            kv_all = ops.linear(enc_hidden, self.cross_kv).view(
+                be, se, len(self.block), 2, h, dkv)
+            kv_layers = kv_all.unbind(2)
+
         for i, blk in enumerate(self.block):
             self_cache = caches[i]["self"] if caches is not None else None
             cross_cache = caches[i]["cross"] if (caches is not None and self.is_decoder) else None
@@ -344,7 +413,8 @@ class T5Stack(nn.Module):
                 ca = blk.layer[1]
                 resid, normed = add_norm(ca.layer_norm, resid, delta)
                 delta = ca.EncDecAttention(normed, kv_hidden=enc_hidden,
-                                           kpad=enc_kpad, kv_cache=cross_cache)
+                                           kpad=enc_kpad, kv_cache=cross_cache,
+                                           kv5=kv_layers[i] if kv_layers is not None else None)
             ff = blk.layer[-1]
             resid, normed = add_norm(ff.layer_norm, resid, delta)
             delta = ff.DenseReluDense(normed)
@@ -398,11 +468,14 @@ class T5ForConditionalGeneration(nn.Module):
                 m.weight.data.fill_(f * 1.0)
             elif isinstance(m, T5Attention):
                 m.qkv.part(0).data.normal_(0.0, f * (d_model * d_kv) ** -0.5)
-                m.qkv.part(1).data.normal_(0.0, f * d_model**-0.5)
-                m.qkv.part(2).data.normal_(0.0, f * d_model**-0.5)
+                if not m.kv_external:
+                    m.qkv.part(1).data.normal_(0.0, f * d_model**-0.5)
+                    m.qkv.part(2).data.normal_(0.0, f * d_model**-0.5)
                 m.o.weight.data.normal_(0.0, f * (n_heads * d_kv) ** -0.5)
                 if m.has_relative_attention_bias:
                     m.relative_attention_bias.weight.data.normal_(0.0, f * d_model**-0.5)
+            elif isinstance(m, T5Stack) and m.is_decoder:
+                m.cross_kv.data.normal_(0.0, f * d_model**-0.5)
             elif isinstance(m, T5DenseGatedActDense):
                 m.wi.weight.data.normal_(0.0, f * d_model**-0.5)
                 m.wo.weight.data.normal_(0.0, f * d_ff**-0.5)
@@ -464,14 +537,25 @@ class T5ForConditionalGeneration(nn.Module):
         return self._generate_eager(input_ids, attention_mask, max_new_tokens,
                                     min_new_tokens, eos_token_id)
 
+    def _cross_kv_all(self, enc, b):
+        """One fused (B*S, n*2*inner) GEMM for every decoder layer's static
+        cross-attention kv; returns (n, 2) nested views shaped (B,H,S,D)."""
+        cfg = self.config
+        h, dkv = cfg.num_heads, cfg.d_kv
+        n = len(self.decoder.block)
+        kv_all = ops.linear(enc, self.decoder.cross_kv).view(
+            b, enc.shape[1], n, 2, h, dkv)
+        return [(kv_all[:, :, i, 0].permute(0, 2, 1, 3).contiguous(),
+                 kv_all[:, :, i, 1].permute(0, 2, 1, 3).contiguous())
+                for i in range(n)]
+
     def _decode_caches(self, enc, b, max_new_tokens, dev, dt):
         cfg = self.config
         h, dkv = cfg.num_heads, cfg.d_kv
         caches = []
-        for blk in self.decoder.block:
-            att = blk.layer[1].EncDecAttention
-            ck = att._shape(att.k(enc), b)
-            cv = att._shape(att.v(enc), b)
+        kv_pairs = self._cross_kv_all(enc, b)
+        for i, blk in enumerate(self.decoder.block):
+            ck, cv = kv_pairs[i]
             caches.append({
                 "self": {"k": torch.zeros(b, h, max_new_tokens, dkv, device=dev, dtype=dt),
                          "v": torch.zeros(b, h, max_new_tokens, dkv, device=dev, dtype=dt),
@@ -532,10 +616,10 @@ class T5ForConditionalGeneration(nn.Module):
         state["enc"].copy_(enc)
         if kpad is not None:
             state["kpad"].copy_(kpad)
+        kv_pairs = self._cross_kv_all(enc, input_ids.shape[0])
         for i, blk in enumerate(self.decoder.block):
-            att = blk.layer[1].EncDecAttention
-            state["caches"][i]["cross"]["k"].copy_(att._shape(att.k(enc), input_ids.shape[0]))
-            state["caches"][i]["cross"]["v"].copy_(att._shape(att.v(enc), input_ids.shape[0]))
+            state["caches"][i]["cross"]["k"].copy_(kv_pairs[i][0])
+            state["caches"][i]["cross"]["v"].copy_(kv_pairs[i][1])
             state["caches"][i]["self"]["len"] = 0
         state["graph"].replay()
         return state["out"].clone()
