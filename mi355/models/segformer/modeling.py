@@ -124,8 +124,10 @@ class DWConv(nn.Module):
 
     def forward(self, x, h, w):
         b, n, c = x.shape
-        x = x.transpose(1, 2).view(b, c, h, w)
-        x = self.dwconv(x)
+        x = x.transpose(1, 2).view(b, c, h, w).contiguous()
+        # hand-written CDNA4 stencil (ops.dwconv3x3) — MIOpen's grouped-conv
+        # path for groups==C bf16 runs generic/naive kernels (K15)
+        x = ops.dwconv3x3(x, self.dwconv.weight, self.dwconv.bias)
         return x.flatten(2).transpose(1, 2)
 
 

@@ -252,6 +252,38 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return torch.nn.functional.linear(x, w)
 
 
+class _DWConv3x3(torch.autograd.Function):
+    """Depthwise 3x3 s1 p1 conv (SegFormer Mix-FFN) on the
+    hand-written stencil kernel; dX reuses the forward kernel with flipped
+    taps, dW/dbias a per-channel reduction kernel."""
+
+    @staticmethod
+    def forward(ctx, x, w, bias):
+        y = _load_ext().dwconv3x3_fwd(x, w, bias, False)
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = _load_ext().dwconv3x3_fwd(dy, w, None, True)
+        dw = db = None
+        if ctx.needs_input_grad[1] or (ctx.has_bias and ctx.needs_input_grad[2]):
+            dw32, db32 = _load_ext().dwconv3x3_wrw(dy, x)
+            dw = dw32.view_as(w).to(w.dtype)
+            db = db32.to(w.dtype) if ctx.has_bias else None
+        return dx, dw, db
+
+
+def dwconv3x3(x: torch.Tensor, w: torch.Tensor, bias=None) -> torch.Tensor:
+    """Depthwise 3x3 stride-1 pad-1 conv. ``w`` is the (C,1,3,3) conv weight."""
+    if _use_hip(x) and x.shape[-1] % 8 == 0:
+        return _DWConv3x3.apply(x, w, bias)
+    return torch.nn.functional.conv2d(x, w, bias, padding=1, groups=x.shape[1])
+
+
 def gemm_nt(a: torch.Tensor, b: torch.Tensor, splitk: int = 0) -> torch.Tensor:
     """C[M,N] = a[M,K] @ b[N,K]^T on the hand-written MFMA kernel."""
     return _load_ext().gemm_nt(a, b, splitk)

@@ -95,3 +95,38 @@ def test_seg_postprocess_kernel_vs_ref():
     ref = torch_ref.seg_postprocess(logits, (512, 512))
     agree = (out == ref).float().mean().item()
     assert agree > 0.999, agree  # ties may break differently
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [(8, 128, 128, 128), (8, 1024, 16, 16),
+                                   (2, 64, 32, 40)])
+def test_dwconv3x3_matches_fp32_ref(shape):
+    """Hand-written depthwise 3x3 stencil vs fp32 F.conv2d (K15)."""
+    import torch.nn.functional as F
+    from mi355 import ops
+
+    torch.manual_seed(0)
+    b, c, h, w = shape
+    x = torch.randn(b, c, h, w, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    wt = torch.randn(c, 1, 3, 3, device="cuda", dtype=torch.bfloat16,
+                     requires_grad=True)
+    bias = torch.randn(c, device="cuda", dtype=torch.bfloat16,
+                       requires_grad=True)
+    y = ops.dwconv3x3(x, wt, bias)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    x32 = x.detach().float().requires_grad_(True)
+    w32 = wt.detach().float().requires_grad_(True)
+    b32 = bias.detach().float().requires_grad_(True)
+    y32 = F.conv2d(x32, w32, b32, padding=1, groups=c)
+    y32.backward(gy.float())
+
+    def relerr(a, ref):
+        a, ref = a.float(), ref.float()
+        return ((a - ref).abs().max() / (ref.abs().max() + 1e-6)).item()
+
+    assert relerr(y, y32) < 2e-2
+    assert relerr(x.grad, x32.grad) < 2e-2
+    assert relerr(wt.grad, w32.grad) < 2e-2
+    assert relerr(bias.grad, b32.grad) < 2e-2

@@ -13,9 +13,17 @@ from mi355.models.segformer import (SegformerConfig,
 
 
 def main():
+    import argparse
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--train-only", action="store_true")
+    args = ap.parse_args()
     torch.manual_seed(0)
     m = SegformerForSemanticSegmentation(SegformerConfig()).to("cuda", torch.bfloat16)
     # inference: 10 batches x 16 images (Scaling_batch_inference cell-45/56)
+    if args.train_only:
+        _train(m := m.train(), args)
+        return
     m.eval()
     pix = torch.randn(16, 3, 512, 512, device="cuda", dtype=torch.bfloat16)
     with torch.no_grad():
@@ -35,8 +43,10 @@ def main():
     torch.cuda.synchronize()
     print(f"postprocess (fused upsample+argmax): {(time.perf_counter()-t0)*1000:.1f} ms")
 
-    # train step: batch 8 (Scaling_model_training cell-31/47)
-    m.train()
+    _train(m.train(), args)
+
+
+def _train(m, args):
     pix8 = torch.randn(8, 3, 512, 512, device="cuda", dtype=torch.bfloat16)
     labels = torch.randint(0, 150, (8, 512, 512), device="cuda")
     opt = torch.optim.AdamW(m.parameters(), lr=1e-4)
@@ -48,12 +58,13 @@ def main():
         step()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for _ in range(10):
+    for _ in range(args.steps):
         loss = step()
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
-    print(f"segformer-b0 train: batch 8 @512x512, {dt/10*1000:.1f} ms/step = "
-          f"{80/dt:.1f} img/s, loss={loss.item():.3f}")
+    n = args.steps
+    print(f"segformer-b0 train: batch 8 @512x512, {dt/n*1000:.1f} ms/step = "
+          f"{8*n/dt:.1f} img/s, loss={loss.item():.3f}")
 
 
 if __name__ == "__main__":
