@@ -49,6 +49,9 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b, long splitk);
 torch::Tensor dwconv3x3_fwd(torch::Tensor x, torch::Tensor w,
                             c10::optional<torch::Tensor> bias, bool flip);
 std::vector<torch::Tensor> dwconv3x3_wrw(torch::Tensor dout, torch::Tensor x);
+torch::Tensor patch_embed_s1_fwd(torch::Tensor x, torch::Tensor w,
+                                 c10::optional<torch::Tensor> bias);
+std::vector<torch::Tensor> patch_embed_s1_wrw(torch::Tensor dout, torch::Tensor x);
 torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b, long splitk);
 torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b, long splitk);
 std::vector<torch::Tensor> attn_fwd_v2(torch::Tensor q, torch::Tensor k, torch::Tensor v,
@@ -96,6 +99,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dwconv3x3_fwd", &dwconv3x3_fwd, py::arg("x"), py::arg("w"),
         py::arg("bias") = c10::nullopt, py::arg("flip") = false);
   m.def("dwconv3x3_wrw", &dwconv3x3_wrw);
+  m.def("patch_embed_s1_fwd", &patch_embed_s1_fwd, py::arg("x"), py::arg("w"),
+        py::arg("bias") = c10::nullopt);
+  m.def("patch_embed_s1_wrw", &patch_embed_s1_wrw);
   m.def("gemm_nn", &gemm_nn, py::arg("a"), py::arg("b"), py::arg("splitk") = 0);
   m.def("gemm_tn", &gemm_tn, py::arg("a"), py::arg("b"), py::arg("splitk") = 0);
   m.def("attn_fwd_v2", &attn_fwd_v2);

@@ -74,29 +74,32 @@ __global__ void dwconv3x3_kernel(const T* __restrict__ in,
 template <typename T>
 __global__ void dwconv3x3_wrw_kernel(const T* __restrict__ dout,
                                      const T* __restrict__ in,
-                                     float* __restrict__ dw10,  // (C, 10)
-                                     int B, int C, int H, int W) {
+                                     float* __restrict__ dw10,  // (C, 10), zeroed
+                                     int B, int C, int H, int W, int slabs) {
   const int c = blockIdx.x;
   const long HW = (long)H * W;
+  const long M = (long)B * HW;
+  const long m0 = blockIdx.y * M / slabs;
+  const long m1 = (blockIdx.y + 1) * M / slabs;
   float acc[9] = {0, 0, 0, 0, 0, 0, 0, 0, 0};
   float db = 0.f;
-  for (int b = 0; b < B; ++b) {
+  for (long m = m0 + threadIdx.x; m < m1; m += blockDim.x) {
+    const int b = (int)(m / HW);
+    const long i = m % HW;
+    const int y = (int)(i / W), x = (int)(i % W);
     const T* dp = dout + ((long)b * C + c) * HW;
     const T* ip = in + ((long)b * C + c) * HW;
-    for (long i = threadIdx.x; i < HW; i += blockDim.x) {
-      const int y = (int)(i / W), x = (int)(i % W);
-      const float g = DT<T>::to_f32(dp[i]);
-      db += g;
+    const float g = DT<T>::to_f32(dp[i]);
+    db += g;
 #pragma unroll
-      for (int dy = 0; dy < 3; ++dy) {
-        const int yy = y + dy - 1;
-        if (yy < 0 || yy >= H) continue;
+    for (int dy = 0; dy < 3; ++dy) {
+      const int yy = y + dy - 1;
+      if (yy < 0 || yy >= H) continue;
 #pragma unroll
-        for (int dx = 0; dx < 3; ++dx) {
-          const int xx = x + dx - 1;
-          if (xx < 0 || xx >= W) continue;
-          acc[dy * 3 + dx] += g * DT<T>::to_f32(ip[(long)yy * W + xx]);
-        }
+      for (int dx = 0; dx < 3; ++dx) {
+        const int xx = x + dx - 1;
+        if (xx < 0 || xx >= W) continue;
+        acc[dy * 3 + dx] += g * DT<T>::to_f32(ip[(long)yy * W + xx]);
       }
     }
   }
@@ -104,10 +107,10 @@ __global__ void dwconv3x3_wrw_kernel(const T* __restrict__ dout,
 #pragma unroll
   for (int t = 0; t < 9; ++t) {
     const float r = block_reduce_sum(acc[t], smem);
-    if (threadIdx.x == 0) dw10[c * 10 + t] = r;
+    if (threadIdx.x == 0) atomicAdd(&dw10[c * 10 + t], r);
   }
   const float r = block_reduce_sum(db, smem);
-  if (threadIdx.x == 0) dw10[c * 10 + 9] = r;
+  if (threadIdx.x == 0) atomicAdd(&dw10[c * 10 + 9], r);
 }
 
 }  // namespace
@@ -154,25 +157,28 @@ std::vector<torch::Tensor> dwconv3x3_wrw(torch::Tensor dout, torch::Tensor x) {
   const int B = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   auto dc = dout.contiguous();
   auto xc = x.contiguous();
-  auto dw10 = torch::empty({C, 10}, x.options().dtype(torch::kFloat));
+  auto dw10 = torch::zeros({C, 10}, x.options().dtype(torch::kFloat));
+  // fill ~2 blocks/CU regardless of channel count
+  const int slabs = std::max(1, (int)(512 / C) + 1);
+  dim3 grid(C, slabs);
   auto stream = at::hip::getCurrentHIPStream().stream();
   AT_DISPATCH_SWITCH(x.scalar_type(), "dwconv3x3_wrw", AT_DISPATCH_CASE(
       torch::kBFloat16, [&] {
-        hipLaunchKernelGGL((dwconv3x3_wrw_kernel<__hip_bfloat16>), dim3(C),
+        hipLaunchKernelGGL((dwconv3x3_wrw_kernel<__hip_bfloat16>), grid,
                            dim3(256), 0, stream,
                            (const __hip_bfloat16*)dc.data_ptr(),
                            (const __hip_bfloat16*)xc.data_ptr(),
-                           dw10.data_ptr<float>(), B, C, H, W);
+                           dw10.data_ptr<float>(), B, C, H, W, slabs);
       }) AT_DISPATCH_CASE(torch::kHalf, [&] {
-        hipLaunchKernelGGL((dwconv3x3_wrw_kernel<__half>), dim3(C),
+        hipLaunchKernelGGL((dwconv3x3_wrw_kernel<__half>), grid,
                            dim3(256), 0, stream, (const __half*)dc.data_ptr(),
                            (const __half*)xc.data_ptr(),
-                           dw10.data_ptr<float>(), B, C, H, W);
+                           dw10.data_ptr<float>(), B, C, H, W, slabs);
       }) AT_DISPATCH_CASE(torch::kFloat, [&] {
-        hipLaunchKernelGGL((dwconv3x3_wrw_kernel<float>), dim3(C),
+        hipLaunchKernelGGL((dwconv3x3_wrw_kernel<float>), grid,
                            dim3(256), 0, stream, (const float*)dc.data_ptr(),
                            (const float*)xc.data_ptr(),
-                           dw10.data_ptr<float>(), B, C, H, W);
+                           dw10.data_ptr<float>(), B, C, H, W, slabs);
       }));
   HIP_CHECK_KERNEL();
   auto dw = dw10.narrow(1, 0, 9).contiguous();

@@ -70,7 +70,12 @@ class OverlapPatchEmbed(nn.Module):
         self.layer_norm = nn.LayerNorm(out_ch, eps=eps)
 
     def forward(self, x):
-        x = self.proj(x)
+        if self.proj.kernel_size == (7, 7) and self.proj.in_channels == 3:
+            # hand-written stage-1 conv (K13; ops.patch_embed_s1 falls back
+            # to torch off-GPU or when x requires grad)
+            x = ops.patch_embed_s1(x, self.proj.weight, self.proj.bias)
+        else:
+            x = self.proj(x)
         _, _, h, w = x.shape
         x = x.flatten(2).transpose(1, 2)  # (B, N, C)
         return self.layer_norm(x), h, w
@@ -253,6 +258,9 @@ class SegformerForSemanticSegmentation(nn.Module):
         # solvers instead. Convs only exist on this model family.
         if torch.cuda.is_available():
             torch.backends.cudnn.benchmark = True
+            from .miopen_cache import seed_miopen_find_db
+
+            seed_miopen_find_db()
         self.config = config
         self.segformer = SegformerEncoder(config)
         self.decode_head = SegformerDecodeHead(config)

@@ -284,6 +284,37 @@ def dwconv3x3(x: torch.Tensor, w: torch.Tensor, bias=None) -> torch.Tensor:
     return torch.nn.functional.conv2d(x, w, bias, padding=1, groups=x.shape[1])
 
 
+class _PatchEmbedS1(torch.autograd.Function):
+    """SegFormer stage-1 overlapping patch-embed conv (7x7 s4 p3, 3->32) on
+    the hand-written tiled kernel (K13).  dX is not produced — stage 1
+    consumes the input image, which never requires grad."""
+
+    @staticmethod
+    def forward(ctx, x, w, bias):
+        y = _load_ext().patch_embed_s1_fwd(x, w, bias)
+        ctx.save_for_backward(x)
+        ctx.w_dtype = w.dtype
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        dw32, db32 = _load_ext().patch_embed_s1_wrw(dy.contiguous(), x)
+        return (None, dw32.to(ctx.w_dtype),
+                db32.to(ctx.w_dtype) if ctx.has_bias else None)
+
+
+def patch_embed_s1(x: torch.Tensor, w: torch.Tensor, bias=None) -> torch.Tensor:
+    """conv2d(x, w, bias, stride=4, padding=3) for the (32,3,7,7) stage-1
+    SegFormer patch embed; falls back to torch when the input needs grad or
+    the shape differs (MIOpen runs this bf16 Cin=3 shape on naive_conv_*)."""
+    if (_use_hip(x) and not x.requires_grad and tuple(w.shape) == (32, 3, 7, 7)
+            and x.shape[1] == 3):
+        return _PatchEmbedS1.apply(x, w, bias)
+    return torch.nn.functional.conv2d(x, w, bias, stride=4, padding=3)
+
+
 def gemm_nt(a: torch.Tensor, b: torch.Tensor, splitk: int = 0) -> torch.Tensor:
     """C[M,N] = a[M,K] @ b[N,K]^T on the hand-written MFMA kernel."""
     return _load_ext().gemm_nt(a, b, splitk)

@@ -130,3 +130,34 @@ def test_dwconv3x3_matches_fp32_ref(shape):
     assert relerr(x.grad, x32.grad) < 2e-2
     assert relerr(wt.grad, w32.grad) < 2e-2
     assert relerr(bias.grad, b32.grad) < 2e-2
+
+
+@pytest.mark.gpu
+def test_patch_embed_s1_matches_fp32_ref():
+    """Hand-written stage-1 patch-embed conv (7x7 s4 p3, 3->32) vs fp32
+    F.conv2d, forward + dW/dbias (K13)."""
+    import torch.nn.functional as F
+    from mi355 import ops
+
+    torch.manual_seed(1)
+    x = torch.randn(2, 3, 512, 512, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(32, 3, 7, 7, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    b = torch.randn(32, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = ops.patch_embed_s1(x, w, b)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    w32 = w.detach().float().requires_grad_(True)
+    b32 = b.detach().float().requires_grad_(True)
+    y32 = F.conv2d(x.float(), w32, b32, stride=4, padding=3)
+    y32.backward(gy.float())
+
+    def relerr(a, ref):
+        a, ref = a.float(), ref.float()
+        return ((a - ref).abs().max() / (ref.abs().max() + 1e-6)).item()
+
+    assert y.shape == y32.shape
+    assert relerr(y, y32) < 2e-2
+    assert relerr(w.grad, w32.grad) < 2e-2
+    assert relerr(b.grad, b32.grad) < 2e-2
