@@ -54,7 +54,12 @@ def _train(m, args):
         out = m(pixel_values=pix8, labels=labels)
         opt.zero_grad(); out.loss.backward(); opt.step()
         return out.loss
-    for _ in range(6):  # MIOpen find mode autotunes on first calls
+    t0 = time.perf_counter()
+    step()
+    torch.cuda.synchronize()
+    print(f"first train step (incl. any MIOpen find): "
+          f"{time.perf_counter()-t0:.2f} s")
+    for _ in range(5):
         step()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
