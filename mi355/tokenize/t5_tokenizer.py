@@ -65,23 +65,27 @@ class T5Tokenizer:
 
     # -- encoding -------------------------------------------------------- #
     def _encode_one(self, text: str, text_pair: str | None, max_length: int,
-                    truncation: bool):
-        ids = self.sp.encode_text(text) + [self.eos_token_id]
+                    truncation: bool, add_special_tokens: bool = True):
+        eos = [self.eos_token_id] if add_special_tokens else []
+        ids = self.sp.encode_text(text) + eos
         if text_pair is not None:
-            ids += self.sp.encode_text(text_pair) + [self.eos_token_id]
+            ids += self.sp.encode_text(text_pair) + eos
         if truncation and len(ids) > max_length:
             ids = ids[:max_length]
-            ids[-1] = self.eos_token_id
+            if add_special_tokens:
+                ids[-1] = self.eos_token_id
         return ids
 
     def __call__(self, text, text_pair=None, padding=False, truncation=False,
-                 max_length=None, return_tensors=None, **_):
+                 max_length=None, return_tensors=None,
+                 add_special_tokens=True, **_):
         max_length = max_length or self.model_max_length
         batched = isinstance(text, (list, tuple))
         texts = list(text) if batched else [text]
         pairs = (list(text_pair) if batched else [text_pair]) if text_pair is not None \
             else [None] * len(texts)
-        encoded = [self._encode_one(t, p, max_length, bool(truncation))
+        encoded = [self._encode_one(t, p, max_length, bool(truncation),
+                                    add_special_tokens)
                    for t, p in zip(texts, pairs)]
         if padding == "max_length":
             target = max_length

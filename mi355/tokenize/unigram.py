@@ -10,10 +10,13 @@ build a tokenizer without any downloaded model file.
 from __future__ import annotations
 
 import math
+import re
 import struct
+import unicodedata
 from dataclasses import dataclass, field
 
 SPACE = "▁"  # the sentencepiece meta-space
+_WS_RUN = re.compile(r"\s+")
 
 
 # --------------------------------------------------------------------------- #
@@ -150,8 +153,14 @@ class UnigramModel:
 
     # -- encoding ------------------------------------------------------- #
     def encode_text(self, text: str) -> list[int]:
-        """Normalize (spaces -> meta-symbol, leading space) then Viterbi."""
-        text = SPACE + text.strip().replace(" ", SPACE)
+        """Normalize then Viterbi.  Normalization follows sentencepiece's
+        default nmt_nfkc spec (what google/flan-t5's spiece.model declares):
+        NFKC, whitespace runs collapsed and stripped (remove_extra_
+        whitespaces=1), dummy-prefix space, spaces -> U+2581 (byte-parity
+        checked against the real spm runtime in tests/test_hf_interchange)."""
+        text = unicodedata.normalize("NFKC", text)
+        text = _WS_RUN.sub(" ", text).strip()
+        text = SPACE + text.replace(" ", SPACE)
         if self._core is not None:
             return list(self._core.encode(text.encode("utf-8")))
         n = len(text)
@@ -192,7 +201,10 @@ class UnigramModel:
                     parts.append(piece)
                 elif t == 2:
                     parts.append(" ⁇ ")  # sp's unk surface
-        return "".join(parts).replace(SPACE, " ").strip()
+        s = "".join(parts).replace(SPACE, " ")
+        # spm removes only the dummy-prefix leading space — a trailing space
+        # (e.g. from a final unk's " ⁇ " surface) is preserved
+        return s[1:] if s.startswith(" ") else s
 
     def id_to_piece(self, i: int) -> str:
         return self.pieces[i][0]
