@@ -476,10 +476,20 @@ def init(num_workers: int | None = None, num_gpus: int | None = None,
             return _runtime
         raise RuntimeError("mi355 runtime already initialized")
     _runtime = Runtime(num_workers=num_workers, num_gpus=num_gpus)
+    dash = ""
+    if os.environ.get("MI355_DASHBOARD", "1") != "0":
+        try:
+            from ..dashboard import start_dashboard
+
+            host, port = start_dashboard()
+            dash = f", dashboard at http://{host}:{port}"
+        except Exception:
+            pass
     # the reference's env smoke test is "ray.init() prints a started line"
-    # (Install_locally.md) — mirror that signal
+    # (Install_locally.md) — mirror that signal, incl. the dashboard URL
+    # (Install_locally.md:64-67)
     print(f"mi355 runtime started: {len(_runtime.workers)} workers, "
-          f"{_runtime.num_gpus} GPU(s), session {_session_prefix()}")
+          f"{_runtime.num_gpus} GPU(s), session {_session_prefix()}{dash}")
     return _runtime
 
 
