@@ -123,6 +123,12 @@ class FlatReplica:
 
         self._works = []
         self._flushed = set()
+        # collective-order log for the divergence assertion (VERDICT r1
+        # item 2): every rank must issue the same collectives in the same
+        # order or RCCL deadlocks/corrupts; checked on the first step (and
+        # every step under MI355_CHECK_COLLECTIVES=1)
+        self._order_log: list = []
+        self._order_checked = False
         self._hooks = []
         if is_distributed():
             if not self.deferred:
@@ -159,6 +165,7 @@ class FlatReplica:
         work = dist.all_reduce(view, op=dist.ReduceOp.SUM, group=self.pg, async_op=True)
         self._works.append(work)
         self._flushed.add((gi, bi))
+        self._order_log.append(("bucket", gi, bi, int(end - start)))
 
     def collect_grads(self):
         """Deferred mode: batch-copy the engine-stored per-param grads into
@@ -203,6 +210,7 @@ class FlatReplica:
                     self._works.append(work)
                     self._flushed.update((gi, bi)
                                          for bi in range(len(g.buckets)))
+                    self._order_log.append(("whole", gi, -1, g.flat_grad.numel()))
                 continue
             for bi, (bs, be, _last) in enumerate(g.buckets):
                 if (gi, bi) in self._flushed:
@@ -211,12 +219,36 @@ class FlatReplica:
                                        group=self.pg, async_op=True)
                 self._works.append(work)
                 self._flushed.add((gi, bi))
+                self._order_log.append(("bucket", gi, bi, int(be - bs)))
 
     # ------------------------------------------------------------------ #
+    def _check_collective_order(self):
+        """Assert every rank issued the SAME collectives in the SAME order
+        this step (a divergence means an RCCL hang or silent corruption on
+        hardware).  Runs on the first synced step, and on every step under
+        MI355_CHECK_COLLECTIVES=1.  Uses all_gather_object (control-plane,
+        works on gloo and nccl/RCCL alike)."""
+        world = get_world_size()
+        if world <= 1 or not self._order_log:
+            return
+        every = os.environ.get("MI355_CHECK_COLLECTIVES", "0") == "1"
+        if self._order_checked and not every:
+            return
+        self._order_checked = True
+        gathered: list = [None] * world
+        dist.all_gather_object(gathered, self._order_log, group=self.pg)
+        if any(g != gathered[0] for g in gathered[1:]):
+            detail = "\n".join(f"rank{r}: {g[:6]}... ({len(g)} colls)"
+                               for r, g in enumerate(gathered))
+            raise RuntimeError(
+                "collective-order divergence across ranks — every rank must "
+                f"flush the same buckets in the same order:\n{detail}")
+
     def finish_grad_sync(self):
         """Wait for in-flight bucket all-reduces and average."""
         world = get_world_size()
         if is_distributed():
+            self._check_collective_order()
             for w in self._works:
                 w.wait()
             # Safety net: reduce any bucket that never flushed (e.g. a
@@ -228,6 +260,7 @@ class FlatReplica:
                                         group=self.pg)
             self._works.clear()
             self._flushed.clear()
+            self._order_log = []
         if world > 1:
             for _, g in self.groups:
                 g.flat_grad.div_(world)

@@ -304,3 +304,59 @@ def test_collect_grads_zeroes_unused_param_slice():
             sl = g.flat_grad[off:off + p.numel()]
             if p.grad is None:
                 assert torch.all(sl == 0), "stale gradient survived in flat buffer"
+
+
+def _order_check_worker(rank, world, port, diverge):
+    import os
+
+    import torch
+    import torch.distributed as dist
+
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from mi355.parallel.ddp import FlatReplica
+
+    m = torch.nn.Linear(8, 8)
+    rep = FlatReplica(m)
+    m(torch.randn(2, 8)).sum().backward()
+    if diverge and rank == 1:
+        # simulate a rank issuing an extra collective
+        rep._order_log.append(("bucket", 9, 9, 1))
+    try:
+        rep.finish_grad_sync()
+        ok = True
+    except RuntimeError as e:
+        ok = "divergence" not in str(e)
+    dist.destroy_process_group()
+    if diverge and ok:
+        raise SystemExit(3)  # should have raised
+    if not diverge and not ok:
+        raise SystemExit(4)
+
+
+def _free_port():
+    import socket
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+@pytest.mark.parametrize("diverge", [False, True])
+def test_collective_order_divergence_assertion(diverge):
+    """every rank must issue the same collectives in the
+    same order; a divergence must raise instead of hanging RCCL."""
+    import multiprocessing as mp
+
+    port = _free_port()
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_order_check_worker, args=(r, 2, port, diverge))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(120)
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
