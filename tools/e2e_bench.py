@@ -64,8 +64,9 @@ def main():
         model = T5ForConditionalGeneration(cfg)
         if use_gpu:
             model = model.to(torch.bfloat16)
+        import tempfile
         targs = TrainingArguments(
-            "e2e-bench", learning_rate=config.get("learning_rate", 2e-5),
+            os.path.join(tempfile.gettempdir(), "e2e-bench"), learning_rate=config.get("learning_rate", 2e-5),
             num_train_epochs=args.epochs, weight_decay=0.01,
             per_device_train_batch_size=2, per_device_eval_batch_size=2,
             evaluation_strategy="epoch", save_strategy="epoch",
