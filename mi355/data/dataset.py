@@ -74,8 +74,42 @@ def _from_table(tbl: pa.Table, fmt: str):
 
 
 class Dataset:
-    def __init__(self, blocks: list[pa.Table]):
+    """Arrow-block dataset.  Blocks are either materialized
+    ``pa.Table``s or runtime ``ObjectRef``s pointing into the shared-memory
+    object store (map_batches through the task runtime
+    keeps results as refs, so chained transforms PIPELINE block-by-block
+    through shm and the driver's peak memory is one block, not the dataset).
+    Row-streaming consumers (count/schema/take/iter_batches/map_batches)
+    resolve refs lazily one block at a time; whole-table consumers
+    (to_arrow/to_pandas/sort/groupby/split) materialize, which the DDP
+    trainer does anyway when sharding."""
+
+    def __init__(self, blocks: list):
         self._blocks = [b for b in blocks]
+
+    # -- ref-block plumbing --------------------------------------------- #
+    @staticmethod
+    def _is_ref(b) -> bool:
+        return not isinstance(b, pa.Table)
+
+    def _resolve(self, b) -> pa.Table:
+        if isinstance(b, pa.Table):
+            return b
+        from .. import runtime
+
+        return runtime.get(b)
+
+    def _tables(self):
+        """Yield blocks as tables, resolving object-store refs lazily —
+        bounded driver memory (one block at a time)."""
+        for b in self._blocks:
+            yield self._resolve(b)
+
+    def materialize(self) -> "Dataset":
+        return Dataset(list(self._tables()))
+
+    def is_fully_materialized(self) -> bool:
+        return all(isinstance(b, pa.Table) for b in self._blocks)
 
     # ------------------------------------------------------------------ #
     # constructors (module-level wrappers below)                         #
@@ -144,23 +178,24 @@ class Dataset:
     # inspection                                                          #
     # ------------------------------------------------------------------ #
     def count(self) -> int:
-        return sum(b.num_rows for b in self._blocks)
+        return sum(b.num_rows for b in self._tables())
 
     def __len__(self):
         return self.count()
 
     def schema(self):
-        return self._blocks[0].schema if self._blocks else None
+        return self._resolve(self._blocks[0]).schema if self._blocks else None
 
     def columns(self):
-        return list(self._blocks[0].column_names) if self._blocks else []
+        return list(self._resolve(self._blocks[0]).column_names) \
+            if self._blocks else []
 
     def num_blocks(self) -> int:
         return len(self._blocks)
 
     def take(self, n: int = 20):
         out = []
-        for b in self._blocks:
+        for b in self._tables():
             for row in b.slice(0, builtins.min(n - len(out), b.num_rows)).to_pylist():
                 out.append(row)
             if len(out) >= n:
@@ -177,11 +212,12 @@ class Dataset:
     def to_pandas(self, limit: int | None = None) -> pd.DataFrame:
         if not self._blocks:
             return pd.DataFrame()
-        df = pa.concat_tables(self._blocks, promote_options="default").to_pandas()
+        df = pa.concat_tables(list(self._tables()),
+                              promote_options="default").to_pandas()
         return df.head(limit) if limit is not None else df
 
     def to_arrow(self) -> pa.Table:
-        return pa.concat_tables(self._blocks, promote_options="default")
+        return pa.concat_tables(list(self._tables()), promote_options="default")
 
     def stats(self) -> str:
         return (f"Dataset(num_blocks={self.num_blocks()}, num_rows={self.count()}, "
@@ -196,7 +232,7 @@ class Dataset:
     def limit(self, n: int) -> "Dataset":
         out = []
         left = n
-        for b in self._blocks:
+        for b in self._tables():
             if left <= 0:
                 break
             take = builtins.min(left, b.num_rows)
@@ -255,14 +291,14 @@ class Dataset:
         return Dataset([tbl.sort_by([(key, order)])])
 
     def drop_columns(self, cols) -> "Dataset":
-        return Dataset([b.drop_columns(cols) for b in self._blocks])
+        return Dataset([b.drop_columns(cols) for b in self._tables()])
 
     def select_columns(self, cols) -> "Dataset":
-        return Dataset([b.select(cols) for b in self._blocks])
+        return Dataset([b.select(cols) for b in self._tables()])
 
     def add_column(self, name: str, fn) -> "Dataset":
         out = []
-        for b in self._blocks:
+        for b in self._tables():
             df = b.to_pandas()
             df[name] = fn(df)
             out.append(pa.Table.from_pandas(df, preserve_index=False))
@@ -270,11 +306,11 @@ class Dataset:
 
     def rename_columns(self, mapping: dict) -> "Dataset":
         new_names = [mapping.get(c, c) for c in self.columns()]
-        return Dataset([b.rename_columns(new_names) for b in self._blocks])
+        return Dataset([b.rename_columns(new_names) for b in self._tables()])
 
     def filter(self, fn) -> "Dataset":
         out = []
-        for b in self._blocks:
+        for b in self._tables():
             df = b.to_pandas()
             out.append(pa.Table.from_pandas(df[df.apply(fn, axis=1)],
                                             preserve_index=False))
@@ -295,23 +331,37 @@ class Dataset:
         from .. import runtime
 
         fn_kwargs = fn_constructor_kwargs or {}
-        batches = list(self._iter_block_batches(batch_size))
 
         if isinstance(compute, ActorPoolStrategy):
+            batches = list(self._iter_block_batches(batch_size))
             return self._map_actor_pool(fn, batches, batch_format, compute,
                                         fn_constructor_args, fn_kwargs, num_gpus)
 
-        if runtime.is_initialized() and len(batches) > 1 and not isinstance(fn, type):
+        if runtime.is_initialized() and not isinstance(fn, type) \
+                and (self.num_blocks() > 1
+                     or any(self._is_ref(b) for b in self._blocks)):
+            # STREAMING task path: one task per block.
+            # Ref blocks are handed to the worker AS refs (it fetches from
+            # shm directly — the driver never touches the bytes) and results
+            # stay in the object store as refs, so a chain of map_batches
+            # pipelines block-by-block without driver materialization.
             import cloudpickle
 
+            fnb = cloudpickle.dumps(fn)
             task = runtime.remote(_apply_batch_fn)
-            refs = [task.remote(cloudpickle.dumps(fn), tbl, batch_format)
-                    for tbl in batches]
-            return Dataset(runtime.get(refs))
+            refs = []
+            for blk in self._blocks:
+                if self._is_ref(blk):
+                    refs.append(task.remote(fnb, blk, batch_format, batch_size))
+                else:
+                    for tbl in _slice_batches(blk, batch_size):
+                        refs.append(task.remote(fnb, tbl, batch_format, None))
+            return Dataset(refs)
 
         if isinstance(fn, type):
             fn = fn(*fn_constructor_args, **fn_kwargs)
-        out = [_to_table(fn(_from_table(tbl, batch_format))) for tbl in batches]
+        out = [_to_table(fn(_from_table(tbl, batch_format)))
+               for tbl in self._iter_block_batches(batch_size)]
         return Dataset(out)
 
     def _map_actor_pool(self, fn_cls, batches, batch_format, strategy,
@@ -345,18 +395,13 @@ class Dataset:
         return Dataset(out)
 
     def _iter_block_batches(self, batch_size):
-        for b in self._blocks:
-            if batch_size is None or b.num_rows <= batch_size:
-                if b.num_rows:
-                    yield b
-            else:
-                for off in range(0, b.num_rows, batch_size):
-                    yield b.slice(off, builtins.min(batch_size, b.num_rows - off))
+        for b in self._tables():
+            yield from _slice_batches(b, batch_size)
 
     def iter_batches(self, *, batch_size: int = 256, batch_format: str = "default",
                      drop_last: bool = False):
         carry = None
-        for b in self._blocks:
+        for b in self._tables():
             tbl = b if carry is None else pa.concat_tables([carry, b])
             carry = None
             off = 0
@@ -460,8 +505,25 @@ class GroupedData:
         return self._agg("min", cols)
 
 
-def _apply_batch_fn(fn_bytes, tbl, batch_format):
+def _slice_batches(b: pa.Table, batch_size):
+    if batch_size is None or b.num_rows <= batch_size:
+        if b.num_rows:
+            yield b
+        return
+    for off in range(0, b.num_rows, batch_size):
+        yield b.slice(off, builtins.min(batch_size, b.num_rows - off))
+
+
+def _apply_batch_fn(fn_bytes, tbl, batch_format, batch_size=None):
+    """Worker-side map_batches kernel.  When the driver hands over a whole
+    (possibly ref-resolved) block with a batch_size, the block is
+    re-batched HERE and the outputs concatenated — the sub-batching happens
+    next to the data instead of in the driver."""
     import cloudpickle
 
     fn = cloudpickle.loads(fn_bytes)
+    if batch_size is not None and tbl.num_rows > batch_size:
+        parts = [_to_table(fn(_from_table(t, batch_format)))
+                 for t in _slice_batches(tbl, batch_size)]
+        return pa.concat_tables(parts, promote_options="default")
     return _to_table(fn(_from_table(tbl, batch_format)))

@@ -137,7 +137,10 @@ class DataParallelTrainer:
                     shard = ds.split(n, equal=True)[rank]
                 else:
                     shard = ds  # eval replicated (HF-style full eval per worker)
-                shard_tables[name] = shard._blocks
+                # resolve object-store ref blocks in the driver: trainer
+                # worker processes are not runtime workers, and per-rank
+                # shards are materialized for epoch iteration anyway
+                shard_tables[name] = list(shard._tables())
             shards_per_rank.append(shard_tables)
         return shards_per_rank
 

@@ -232,3 +232,59 @@ def test_split_equal_truncates_to_even_shards():
     # equal=False keeps every row
     shards = ds.split(3)
     assert sum(s.count() for s in shards) == 10
+
+
+def test_map_batches_streams_blocks_through_object_store():
+    """with the runtime up, chained map_batches keeps
+    blocks as object-store refs (driver never materializes the whole
+    dataset); streaming consumers resolve lazily and whole-table consumers
+    still work."""
+    import pandas as pd
+
+    import mi355
+    from mi355.data import Dataset
+
+    mi355.init(num_workers=2)
+    ds = Dataset.from_items([{"x": i} for i in range(100)], parallelism=5)
+    assert ds.num_blocks() == 5
+
+    a = ds.map_batches(lambda df: pd.DataFrame({"x": df["x"] * 2}),
+                       batch_format="pandas", batch_size=None)
+    b = a.map_batches(lambda df: pd.DataFrame({"x": df["x"] + 1}),
+                      batch_format="pandas", batch_size=None)
+    # both stages returned REF blocks: nothing materialized in the driver
+    assert not a.is_fully_materialized()
+    assert not b.is_fully_materialized()
+    # streaming consumers
+    assert b.count() == 100
+    assert b.take(3) == [{"x": 1}, {"x": 3}, {"x": 5}]
+    got = []
+    for batch in b.iter_batches(batch_size=16, batch_format="pandas"):
+        assert len(batch) <= 16
+        got.extend(batch["x"].tolist())
+    assert sorted(got) == sorted(i * 2 + 1 for i in range(100))
+    # whole-table consumer materializes correctly
+    assert b.to_pandas()["x"].sum() == sum(i * 2 + 1 for i in range(100))
+
+
+def test_map_batches_ref_blocks_rebatch_in_worker():
+    """A ref block larger than batch_size is re-batched next to the data
+    (worker side), not in the driver."""
+    import pandas as pd
+
+    import mi355
+    from mi355.data import Dataset
+
+    mi355.init(num_workers=2)
+    ds = Dataset.from_items([{"x": i} for i in range(64)], parallelism=2)
+    a = ds.map_batches(lambda df: df, batch_format="pandas", batch_size=None)
+    assert not a.is_fully_materialized()
+    seen_sizes = []
+
+    def probe(df):
+        seen_sizes.append(len(df))
+        return df
+
+    b = a.map_batches(probe, batch_format="pandas", batch_size=8)
+    assert b.count() == 64  # probe ran worker-side; sizes list stays empty here
+    assert seen_sizes == []  # the closure ran in ANOTHER process
