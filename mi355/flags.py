@@ -43,6 +43,15 @@ FLAGS = [
     Flag("MI355_HYBRID_GRAPH", "0",
          "Force the two-graph hybrid step at world_size 1 (validation knob)."),
     # -- kernels -------------------------------------------------------------
+    Flag("MI355_CHECK_COLLECTIVES", "0",
+         "1 = assert identical cross-rank collective order every step "
+         "(default: first synced step only)."),
+    Flag("MI355_DASHBOARD", "1",
+         "Serve the web dashboard from init() (0 disables)."),
+    Flag("MI355_DASHBOARD_PORT", "8265",
+         "Dashboard port (falls back to an ephemeral port when taken)."),
+    Flag("MI355_RESULTS_DIR", "~/mi355_results",
+         "Run/trial results tree shown by `mi355 status` and the dashboard."),
     Flag("MI355_GEMM", "auto",
          "Projection-GEMM dispatch: auto = per-shape table from the "
          "graph-captured A/B (profiles/r02_gemm_ab.md), hip = always the "
