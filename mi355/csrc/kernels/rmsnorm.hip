@@ -177,6 +177,76 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy, const T* __restrict
 }
 
 // column sum of the <=256 per-block partials, emitted directly in the weight
+// Single-pass bwd for D <= threads*VEC (T5-base D=768): each thread owns ONE
+// vector chunk of the row, so x/dy/w stay in REGISTERS across the dot
+// reduction instead of being re-read from HBM — the generic kernel's second
+// load pass was the latency bottleneck at B=2 (19 us/call, ROADMAP r1 #5).
+template <typename T, int VEC>
+__global__ void rmsnorm_bwd_1p_kernel(const T* __restrict__ dy, const T* __restrict__ x,
+                                      const T* __restrict__ w, const float* __restrict__ inv_rms,
+                                      const T* __restrict__ dextra,
+                                      T* __restrict__ dx, float* __restrict__ dw_part,
+                                      int rows, int D,
+                                      T* __restrict__ ddelta,
+                                      float p_drop, const long* __restrict__ seed_ptr,
+                                      long salt) {
+  __shared__ float red[16];
+  const long long seed = (seed_ptr ? *seed_ptr : 0) ^ salt;
+  const float keep_inv = p_drop > 0.f ? 1.f / (1.f - p_drop) : 1.f;
+  const int i = threadIdx.x * VEC;
+  const bool act = i < D;
+  float dwacc[VEC];
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) dwacc[j] = 0.f;
+  VecT<T, VEC> wv;
+  if (act) wv = vload<T, VEC>(w + i);
+
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const float inv = inv_rms[row];
+    VecT<T, VEC> xv, dyv;
+    float dot = 0.f;
+    if (act) {
+      xv = vload<T, VEC>(x + (long)row * D + i);
+      dyv = vload<T, VEC>(dy + (long)row * D + i);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        dot += DT<T>::to_f32(dyv.data[j]) * DT<T>::to_f32(wv.data[j]) *
+               DT<T>::to_f32(xv.data[j]);
+    }
+    dot = block_reduce_sum(dot, red);
+    if (!act) continue;
+    const float c = inv * inv * inv / D * dot;
+    VecT<T, VEC> o, od, ev;
+    if (dextra) ev = vload<T, VEC>(dextra + (long)row * D + i);
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      const float xf = DT<T>::to_f32(xv.data[j]);
+      const float dyf = DT<T>::to_f32(dyv.data[j]);
+      const float wf = DT<T>::to_f32(wv.data[j]);
+      const float extra = dextra ? DT<T>::to_f32(ev.data[j]) : 0.f;
+      const float dxf = inv * wf * dyf - xf * c + extra;
+      o.data[j] = DT<T>::from_f32(dxf);
+      if (ddelta) {
+        float mf = 1.f;
+        if (p_drop > 0.f) {
+          const long long idx = (long long)row * D + i + j;
+          mf = (hash_u01(seed, idx) >= p_drop) ? keep_inv : 0.f;
+        }
+        od.data[j] = DT<T>::from_f32(dxf * mf);
+      }
+      dwacc[j] += dyf * xf * inv;
+    }
+    vstore<T, VEC>(dx + (long)row * D + i, o);
+    if (ddelta) vstore<T, VEC>(ddelta + (long)row * D + i, od);
+  }
+  if (act) {
+    VecT<float, VEC> o;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) o.data[j] = dwacc[j];
+    vstore<float, VEC>(dw_part + (long)blockIdx.x * D + i, o);
+  }
+}
+
 // dtype (no separate fp32->bf16 cast launch, no zero-fill of dw).
 // block = 16 row-lanes x 16 columns so the row loop is 16-way parallel —
 // a 1-thread-per-column version was latency-bound at 59 us/call.
@@ -225,17 +295,31 @@ void launch_bwd(const torch::Tensor& dy, const torch::Tensor& x, const torch::Te
   constexpr int VEC = sizeof(T) == 4 ? 4 : 8;
   const int threads = (D / VEC >= 256) ? 256 : 128;
   TORCH_CHECK(D <= threads * VEC * 4, "rmsnorm bwd: D too large for dw accumulator");
+  const bool onepass = D <= 256 * VEC;
   // big-batch rows want more blocks; small batches keep the partial slab
-  // (and its colsum) tiny
-  const int grid = rows <= 2048 ? std::min(rows, 256) : 1024;
+  // (and its colsum) tiny.  The one-pass kernel is latency-bound at small
+  // batch, so give it 2 blocks/CU.
+  const int grid = onepass ? std::min(rows, 512)
+                           : (rows <= 2048 ? std::min(rows, 256) : 1024);
   auto part = torch::empty({grid, D}, dw.options().dtype(torch::kFloat));
-  hipLaunchKernelGGL((rmsnorm_bwd_kernel<T, VEC, 4>), dim3(grid), dim3(threads), 0, stream,
-                     (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (const T*)w.data_ptr(),
-                     inv.data_ptr<float>(),
-                     dextra ? (const T*)dextra->data_ptr() : nullptr,
-                     (T*)dx.data_ptr(), part.data_ptr<float>(), rows, D,
-                     ddelta ? (T*)ddelta->data_ptr() : nullptr, p_drop,
-                     seed ? seed->data_ptr<long>() : nullptr, salt);
+  if (onepass) {
+    const int t1 = std::min(256, ((D + VEC - 1) / VEC + 63) / 64 * 64);
+    hipLaunchKernelGGL((rmsnorm_bwd_1p_kernel<T, VEC>), dim3(grid), dim3(t1), 0, stream,
+                       (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (const T*)w.data_ptr(),
+                       inv.data_ptr<float>(),
+                       dextra ? (const T*)dextra->data_ptr() : nullptr,
+                       (T*)dx.data_ptr(), part.data_ptr<float>(), rows, D,
+                       ddelta ? (T*)ddelta->data_ptr() : nullptr, p_drop,
+                       seed ? seed->data_ptr<long>() : nullptr, salt);
+  } else {
+    hipLaunchKernelGGL((rmsnorm_bwd_kernel<T, VEC, 4>), dim3(grid), dim3(threads), 0, stream,
+                       (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (const T*)w.data_ptr(),
+                       inv.data_ptr<float>(),
+                       dextra ? (const T*)dextra->data_ptr() : nullptr,
+                       (T*)dx.data_ptr(), part.data_ptr<float>(), rows, D,
+                       ddelta ? (T*)ddelta->data_ptr() : nullptr, p_drop,
+                       seed ? seed->data_ptr<long>() : nullptr, salt);
+  }
   HIP_CHECK_KERNEL();
   if (grid > 256)
     hipLaunchKernelGGL((colsum_kernel<T, 64>), dim3((D + 3) / 4), dim3(256), 0,

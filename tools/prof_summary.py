@@ -37,4 +37,5 @@ def summarize(db_path, out_path=None, top=40):
 
 if __name__ == "__main__":
     out = sys.argv[2] if len(sys.argv) > 2 else None
-    print(summarize(sys.argv[1], out))
+    top = int(sys.argv[3]) if len(sys.argv) > 3 else 40
+    print(summarize(sys.argv[1], out, top))
