@@ -161,3 +161,32 @@ def test_patch_embed_s1_matches_fp32_ref():
     assert relerr(y, y32) < 2e-2
     assert relerr(w.grad, w32.grad) < 2e-2
     assert relerr(b.grad, b32.grad) < 2e-2
+
+
+def test_visualization_helpers():
+    """Reference display-glue surface (Semantic_segmentation/utils.py):
+    ade_palette / prepare_pixels_with_segmentation / visualize_predictions /
+    display_example_images / convert_image_to_rgb."""
+    from PIL import Image
+
+    from mi355.models.segformer.visualization import (
+        ade_palette, convert_image_to_rgb, display_example_images,
+        get_image_indices, prepare_pixels_with_segmentation,
+        visualize_predictions)
+
+    pal = ade_palette()
+    assert len(pal) == 151 and pal[0] == [0, 0, 0]
+    assert len({tuple(c) for c in pal}) == 151  # all distinct
+    img = Image.fromarray(
+        (np.random.default_rng(0).random((64, 64, 3)) * 255).astype("uint8"),
+        "RGB").convert("L")  # non-RGB mode exercises convert_image_to_rgb
+    assert convert_image_to_rgb(img).mode == "RGB"
+    seg = np.random.default_rng(1).integers(0, 150, (64, 64))
+    blended = prepare_pixels_with_segmentation(img, seg)
+    assert blended.shape == (64, 64, 3) and blended.dtype == np.uint8
+    ax = visualize_predictions(img, seg)
+    assert ax is not None
+    idx = get_image_indices(10, 4, seed=0)
+    assert len(idx) == 4 and len(set(idx)) == 4
+    fig = display_example_images([img] * 4, n=2)
+    assert fig is not None
