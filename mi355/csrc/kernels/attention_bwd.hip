@@ -44,6 +44,18 @@ constexpr int BKT = 64;  // keys per block
 constexpr int BQT = 64;  // queries per iteration
 constexpr float NEGB = -1e30f;
 
+// Transposed-image ([D][64], no pad) XOR swizzle: element (drow, col) lives
+// at drow*64 + (col ^ key). key mixes drow's low AND mid bits so BOTH access
+// patterns are spread: the staging writes walk a COLUMN with lanes 8 rows
+// apart (key's drow>>3 part varies per lane -> 2-way instead of ~16-way on
+// one bank) and the b128 fragment reads walk 16 consecutive rows (key's
+// drow&7 part varies per lane -> conflict-free). Replaces the +8 row pads
+// (12.5% LDS) whose write pattern was the round-1 PMC bank-conflict hotspot
+// (ROADMAP r1: SQ_LDS_BANK_CONFLICT ~12%% of bwd wave cycles).
+DEVINL int timg(int drow, int col) {
+  return drow * BKT + (col ^ ((((drow & 7) ^ ((drow >> 3) & 7))) << 3));
+}
+
 struct BStr { long sb, sh, ss; };
 
 // delta[b,h,q] = sum_d dO * O (both strided BSHD views)
@@ -78,17 +90,17 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
     float p_drop, const long* __restrict__ seed_ptr) {
   using FR = typename MFMA16B<T>::frag;
   constexpr int LD = D + 8;        // [64][D] images
-  constexpr int LDT = BKT + 8;     // [D][64] transposed images + slabs
+  constexpr int LDT = BKT + 8;     // ds_img + slab row stride
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   T* kt_img = (T*)smem;                         // [BKT][LD]
-  T* ktT_img = kt_img + BKT * LD;               // [D][LDT]
-  T* v_img = ktT_img + D * LDT;                 // [BKT][LD]
+  T* ktT_img = kt_img + BKT * LD;               // [D][64] XOR-swizzled (timg)
+  T* v_img = ktT_img + D * BKT;                 // [BKT][LD]
   T* q_img = v_img + BKT * LD;                  // [BQT][LD]
-  T* qT_img = q_img + BQT * LD;                 // [D][LDT]
-  T* do_img = qT_img + D * LDT;                 // [BQT][LD]
-  T* doT_img = do_img + BQT * LD;               // [D][LDT]
-  T* slab = doT_img + D * LDT;                  // [4][16][LDT] per-wave P^T / dS^T
+  T* qT_img = q_img + BQT * LD;                 // [D][64] XOR-swizzled
+  T* do_img = qT_img + D * BKT;                 // [BQT][LD]
+  T* doT_img = do_img + BQT * LD;               // [D][64] XOR-swizzled
+  T* slab = doT_img + D * BKT;                  // [4][16][LDT] per-wave P^T / dS^T
   // ds_img ALIASES q_img when the strides match (D==64): every q_img read
   // (the S^T B-fragments) completes before the P-slab barrier, and ds_img is
   // only written after it — saves 9 KB LDS -> 2 blocks/CU instead of 1.
@@ -137,7 +149,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
       vstore<T, 8>(&kt_img[jj * LD + d0], kv);
       vstore<T, 8>(&v_img[jj * LD + d0], vv);
 #pragma unroll
-      for (int u = 0; u < 8; ++u) ktT_img[(d0 + u) * LDT + jj] = kv.data[u];
+      for (int u = 0; u < 8; ++u) ktT_img[timg(d0 + u, jj)] = kv.data[u];
     }
   }
 
@@ -205,8 +217,8 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
         vstore<T, 8>(&do_img[jj * LD + d0], dov);
 #pragma unroll
         for (int u = 0; u < 8; ++u) {
-          qT_img[(d0 + u) * LDT + jj] = qv.data[u];
-          doT_img[(d0 + u) * LDT + jj] = dov.data[u];
+          qT_img[timg(d0 + u, jj)] = qv.data[u];
+          doT_img[timg(d0 + u, jj)] = dov.data[u];
         }
       }
     }
@@ -279,7 +291,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
 #pragma unroll
       for (int s = 0; s < BQT / 32; ++s) {
         FR a = *reinterpret_cast<const FR*>(&my_slab[lr * LDT + s * 32 + lg * 8]);
-        FR bfr = *reinterpret_cast<const FR*>(&doT_img[(t * 16 + lr) * LDT + s * 32 + lg * 8]);
+        FR bfr = *reinterpret_cast<const FR*>(&doT_img[timg(t * 16 + lr, s * 32 + lg * 8)]);
         dv_acc[t] = MFMA16B<T>::mma(a, bfr, dv_acc[t]);
       }
     }
@@ -299,7 +311,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
 #pragma unroll
       for (int s = 0; s < BQT / 32; ++s) {
         FR a = *reinterpret_cast<const FR*>(&my_slab[lr * LDT + s * 32 + lg * 8]);
-        FR bfr = *reinterpret_cast<const FR*>(&qT_img[(t * 16 + lr) * LDT + s * 32 + lg * 8]);
+        FR bfr = *reinterpret_cast<const FR*>(&qT_img[timg(t * 16 + lr, s * 32 + lg * 8)]);
         dk_acc[t] = MFMA16B<T>::mma(a, bfr, dk_acc[t]);
       }
     }
@@ -329,7 +341,7 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
 #pragma unroll
       for (int s = 0; s < BKT / 32; ++s) {
         FR a = *reinterpret_cast<const FR*>(&ds_img[(wid * 16 + lr) * LDT + s * 32 + lg * 8]);
-        FR bfr = *reinterpret_cast<const FR*>(&ktT_img[(t * 16 + lr) * LDT + s * 32 + lg * 8]);
+        FR bfr = *reinterpret_cast<const FR*>(&ktT_img[timg(t * 16 + lr, s * 32 + lg * 8)]);
         dq_acc[t] = MFMA16B<T>::mma(a, bfr, dq_acc[t]);
       }
     }
@@ -411,7 +423,7 @@ void launch_bwd(const torch::Tensor& dout, const torch::Tensor& q, const torch::
   }
   constexpr int LD = D + 8, LDT = BKT + 8;
   const int ds_extra = (LD == LDT) ? 0 : BQT * LDT;  // ds_img aliases q_img at D==64
-  const int smem = sizeof(T) * (BKT * LD * 2 + BQT * LD * 2 + D * LDT * 3 +
+  const int smem = sizeof(T) * (BKT * LD * 2 + BQT * LD * 2 + D * BKT * 3 +
                                 ds_extra + 4 * 16 * LDT);
   dim3 grid((Sk + BKT - 1) / BKT, B * H, nqc);
 #define LB(C, HB, HK, DB)                                                             \
