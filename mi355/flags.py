@@ -52,6 +52,11 @@ FLAGS = [
          "Dashboard port (falls back to an ephemeral port when taken)."),
     Flag("MI355_RESULTS_DIR", "~/mi355_results",
          "Run/trial results tree shown by `mi355 status` and the dashboard."),
+    Flag("MI355_SPILL_DIR", "<tmpdir>/mi355_<session>_spill",
+         "Directory for object-store spill files."),
+    Flag("MI355_SPILL_THRESHOLD_MB", "<unset>",
+         "Objects larger than this spill straight to disk; unset = only on "
+         "a full /dev/shm."),
     Flag("MI355_GEMM", "auto",
          "Projection-GEMM dispatch: auto = per-shape table from the "
          "graph-captured A/B (profiles/r02_gemm_ab.md), hip = always the "

@@ -76,7 +76,8 @@ class ActorMethod:
 
 
 class ActorHandle:
-    def __init__(self, cls, init_args, init_kwargs, num_gpus=0, max_restarts=0):
+    def __init__(self, cls, init_args, init_kwargs, num_gpus=0, max_restarts=0,
+                 replay_on_restart=False, max_replay_log=1000):
         rt = _rt()
         self._gpu_id = None
         if num_gpus and rt.num_gpus:
@@ -85,6 +86,15 @@ class ActorHandle:
         self._init_bytes = cloudpickle.dumps((list(init_args), dict(init_kwargs)))
         self._restarts_left = max_restarts
         self._pending: set[str] = set()
+        # opt-in call replay (r1 ROADMAP): record every completed method
+        # call; on a crash-restart re-issue the log (in order, results
+        # discarded) so state-mutating call sequences are reconstructed.
+        # Off by default — the reference's max_restarts restarts with fresh
+        # state — and bounded by max_replay_log (older entries dropped, so
+        # replay is exact only while the log covers the actor's lifetime).
+        self._replay = bool(replay_on_restart)
+        self._max_replay_log = int(max_replay_log)
+        self._call_log: list[tuple] = []
         self._spawn()
         rt.actors.append(self)
         self._cls_name = getattr(cls, "__name__", "Actor")
@@ -126,6 +136,12 @@ class ActorHandle:
         err = ActorDiedError(
             f"actor {getattr(self, '_cls_name', 'Actor')} died "
             f"(exitcode {self._proc.exitcode})")
+        if self._replay:
+            # replay only calls that COMPLETED before the crash — the call
+            # that killed the actor (and anything queued behind it) is
+            # still in _pending and must not be re-issued
+            self._call_log = [e for e in self._call_log
+                              if e[0] not in self._pending]
         for ref_id in self._pending:
             try:
                 store.put(_RemoteError(err, str(err)), ObjectRef(ref_id))
@@ -138,6 +154,18 @@ class ActorHandle:
 
             _prom.ACTOR_RESTARTS.labels(actor=self._cls_name).inc()
             self._spawn()
+            if self._replay and self._call_log:
+                # re-issue the completed calls to rebuild state; replay
+                # results go to throwaway refs (log entries keep their new
+                # refs so a repeat crash prunes correctly)
+                relog = []
+                for _old_ref, method, args, kwargs in list(self._call_log):
+                    rref = ObjectRef(uuid.uuid4().hex)
+                    self._pending.add(rref.id)
+                    relog.append((rref.id, method, args, kwargs))
+                    self._task_q.put((rref.id, method, list(args),
+                                      dict(kwargs)))
+                self._call_log = relog
         else:
             raise err
 
@@ -145,6 +173,10 @@ class ActorHandle:
         self._check_alive()
         ref = ObjectRef(uuid.uuid4().hex)
         self._pending.add(ref.id)
+        if self._replay:
+            self._call_log.append((ref.id, method, list(args), dict(kwargs)))
+            if len(self._call_log) > self._max_replay_log:
+                del self._call_log[0]
         self._task_q.put((ref.id, method, list(args), dict(kwargs)))
         return ref
 
@@ -166,18 +198,23 @@ class ActorHandle:
 
 
 class ActorClass:
-    def __init__(self, cls, num_gpus=0, num_cpus=1, max_restarts=0, **_):
+    def __init__(self, cls, num_gpus=0, num_cpus=1, max_restarts=0,
+                 replay_on_restart=False, **_):
         self._cls = cls
         self._num_gpus = num_gpus
         self._max_restarts = max_restarts
+        self._replay = replay_on_restart
 
     def remote(self, *args, **kwargs) -> ActorHandle:
         return ActorHandle(self._cls, args, kwargs, num_gpus=self._num_gpus,
-                           max_restarts=self._max_restarts)
+                           max_restarts=self._max_restarts,
+                           replay_on_restart=self._replay)
 
     def options(self, **opts):
         return ActorClass(self._cls, num_gpus=opts.get("num_gpus", self._num_gpus),
-                          max_restarts=opts.get("max_restarts", self._max_restarts))
+                          max_restarts=opts.get("max_restarts", self._max_restarts),
+                          replay_on_restart=opts.get("replay_on_restart",
+                                                     self._replay))
 
     def __call__(self, *a, **k):
         raise TypeError("actor classes must be instantiated with .remote()")

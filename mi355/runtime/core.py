@@ -24,6 +24,7 @@ import os
 import pickle
 import queue
 import struct
+import tempfile
 import threading
 import time
 import uuid
@@ -111,6 +112,36 @@ class ObjectStore:
         self._owned: list[str] = []
         self._attached: dict[str, shared_memory.SharedMemory] = {}
         self._native = _native_store
+        # spill-to-disk (SURVEY §2.2 C4 / r1 ROADMAP): objects larger than
+        # MI355_SPILL_THRESHOLD_MB, or any put that fails with a full
+        # /dev/shm, land as regular files under MI355_SPILL_DIR; readers
+        # check the spill path after a shm miss.  Write-then-rename keeps
+        # the sealed-then-immutable discipline (§5.2) on disk too.
+        self._spill_dir = os.environ.get(
+            "MI355_SPILL_DIR", os.path.join(tempfile.gettempdir(),
+                                            f"{_session_prefix()}_spill"))
+        thr = os.environ.get("MI355_SPILL_THRESHOLD_MB")
+        self._spill_threshold = int(float(thr) * 1024 * 1024) if thr else None
+        self._spilled: list[str] = []
+
+    # -- disk spill ------------------------------------------------------ #
+    def _spill_path(self, name: str) -> str:
+        return os.path.join(self._spill_dir, name)
+
+    def _spill_put(self, data: bytes, name: str):
+        os.makedirs(self._spill_dir, exist_ok=True)
+        tmp = self._spill_path(name) + ".tmp"
+        with open(tmp, "wb") as f:
+            f.write(data)
+        os.rename(tmp, self._spill_path(name))  # atomic seal
+        self._spilled.append(name)
+
+    def _spill_get(self, name: str):
+        try:
+            with open(self._spill_path(name), "rb") as f:
+                return f.read()
+        except FileNotFoundError:
+            return None
 
     # -- native-core fast paths ----------------------------------------- #
     def _nput(self, data: bytes, name: str):
@@ -126,8 +157,15 @@ class ObjectStore:
         ref = ref or ObjectRef(uuid.uuid4().hex)
         data = _serialize(obj)
         name = _shm_name(ref)
+        if self._spill_threshold is not None and len(data) > self._spill_threshold:
+            self._spill_put(data, name)
+            return ref
         if self._native is not None:
-            self._nput(data, name)
+            try:
+                self._nput(data, name)
+            except (OSError, MemoryError, RuntimeError):
+                # /dev/shm full: spill to disk instead of failing the put
+                self._spill_put(data, name)
             return ref
         seg = shared_memory.SharedMemory(name=name, create=True, size=len(data) + 8)
         _untrack(name)
@@ -155,6 +193,9 @@ class ObjectStore:
                 buf = self._nget(name)
                 if buf is not None:
                     return _deserialize(memoryview(buf))
+                data = self._spill_get(name)
+                if data is not None:
+                    return _deserialize(memoryview(data))
             else:
                 try:
                     seg = self._attach(name)
@@ -171,13 +212,14 @@ class ObjectStore:
     def ready(self, ref: ObjectRef) -> bool:
         name = _shm_name(ref)
         if self._native is not None:
-            return self._native.ready(name)
+            return self._native.ready(name) or \
+                os.path.exists(self._spill_path(name))
         try:
             seg = self._attach(name)
             (n,) = struct.unpack_from("<Q", seg.buf, 0)
             return n > 0
         except FileNotFoundError:
-            return False
+            return os.path.exists(self._spill_path(name))
 
     def cleanup(self):
         import glob
@@ -210,6 +252,11 @@ class ObjectStore:
         created results) at shutdown."""
         import glob
 
+        for path in glob.glob(os.path.join(self._spill_dir, "*")):
+            try:
+                os.unlink(path)
+            except OSError:
+                pass
         for path in glob.glob(f"/dev/shm/{_session_prefix()}_*"):
             try:
                 os.unlink(path)

@@ -83,3 +83,43 @@ def test_task_worker_crash_detected_and_respawned():
             [0, 2, 4, 6, 8, 10]
     finally:
         mi355.shutdown()
+
+
+def test_actor_replay_on_restart_rebuilds_state():
+    """Opt-in call replay (replay_on_restart=True): after a crash-restart
+    the recorded method calls are re-issued so mutable state is rebuilt;
+    the default stays fresh-state (reference max_restarts semantics)."""
+    import mi355
+
+    mi355.init(num_workers=2)
+
+    @mi355.remote
+    class Counter:
+        def __init__(self):
+            self.n = 0
+
+        def incr(self, by):
+            self.n += by
+            return self.n
+
+        def get(self):
+            return self.n
+
+        def crash(self):
+            import os
+
+            os._exit(1)
+
+    a = Counter.options(max_restarts=1, replay_on_restart=True).remote()
+    assert mi355.get(a.incr.remote(5)) == 5
+    assert mi355.get(a.incr.remote(7)) == 12
+    a.crash.remote()
+    import time
+
+    time.sleep(0.5)
+    # next call detects the death, restarts, replays incr(5), incr(7),
+    # crash... crash is also in the log! -> replay must not re-execute it
+    # blindly; the log records only COMPLETED queries before the crash is
+    # detected. Accept >= the pre-crash value via the rebuild:
+    val = mi355.get(a.get.remote())
+    assert val == 12, f"replayed state should be 12, got {val}"

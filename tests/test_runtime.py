@@ -262,3 +262,26 @@ def test_wait_caps_ready_at_num_returns():
         [done], pending = mi355.wait(pending, num_returns=1, timeout=5.0)
         seen.append(mi355.get(done))
     assert sorted(seen) == [0, 1, 2, 3]
+
+
+def test_object_store_spills_to_disk(monkeypatch, tmp_path):
+    """C4 spill: objects above MI355_SPILL_THRESHOLD_MB land as sealed disk
+    files instead of /dev/shm and resolve transparently through get/ready."""
+    import numpy as np
+
+    monkeypatch.setenv("MI355_SPILL_THRESHOLD_MB", "0.5")
+    monkeypatch.setenv("MI355_SPILL_DIR", str(tmp_path / "spill"))
+    from mi355.runtime.core import ObjectStore
+
+    store = ObjectStore()
+    big = np.arange(300_000, dtype=np.float64)  # 2.4 MB > 0.5 MB threshold
+    small = {"k": 1}
+    rbig = store.put(big)
+    rsmall = store.put(small)
+    spilled = list((tmp_path / "spill").glob("*"))
+    assert len(spilled) == 1, "big object should be on disk, small in shm"
+    assert store.ready(rbig) and store.ready(rsmall)
+    assert np.array_equal(store.get(rbig), big)
+    assert store.get(rsmall) == small
+    store.sweep_session()
+    assert not list((tmp_path / "spill").glob("*"))
