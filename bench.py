@@ -225,7 +225,13 @@ def main():
     if args.device:
         device = torch.device(args.device)
     elif torch.cuda.is_available():
-        device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", "0")))
+        idx = int(os.environ.get("LOCAL_RANK", "0"))
+        if os.environ.get("MI355_SHARE_GPU") == "1":
+            # de-risk smoke: world>1 ranks share one
+            # physical GPU so the RCCL multi-rank path can be exercised
+            # inside a 1-GPU lease; NOT a performance configuration
+            idx = idx % torch.cuda.device_count()
+        device = torch.device("cuda", idx)
         torch.cuda.set_device(device)
     else:
         device = torch.device("cpu")
