@@ -29,7 +29,11 @@ def init_distributed(backend: str | None = None, timeout_s: int = 300) -> int:
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29500")
     if backend == "nccl":
-        torch.cuda.set_device(local_rank())
+        idx = local_rank()
+        if os.environ.get("MI355_SHARE_GPU") == "1":
+            # multi-rank-per-GPU smoke inside a 1-GPU lease (de-risk only)
+            idx = idx % torch.cuda.device_count()
+        torch.cuda.set_device(idx)
     dist.init_process_group(backend=backend,
                             timeout=datetime.timedelta(seconds=timeout_s))
     return dist.get_rank()
