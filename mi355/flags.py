@@ -52,6 +52,10 @@ FLAGS = [
          "Dashboard port (falls back to an ephemeral port when taken)."),
     Flag("MI355_RESULTS_DIR", "~/mi355_results",
          "Run/trial results tree shown by `mi355 status` and the dashboard."),
+    Flag("MI355_SHARE_GPU", "0",
+         "Map world>1 ranks onto one GPU for smoke tests (RCCL rejects the "
+         "collectives — one rank per device — but init/device plumbing is "
+         "exercised; see profiles/r02_summary.md)."),
     Flag("MI355_SPILL_DIR", "<tmpdir>/mi355_<session>_spill",
          "Directory for object-store spill files."),
     Flag("MI355_SPILL_THRESHOLD_MB", "<unset>",
