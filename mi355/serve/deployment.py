@@ -30,15 +30,21 @@ class Deployment:
     num_replicas: int = 1
     route_prefix: str = "/"
     ray_actor_options: dict = field(default_factory=dict)
+    # ray-style autoscaling_config: {"min_replicas", "max_replicas",
+    # "target_num_ongoing_requests_per_replica", "interval_s"} — a controller
+    # thread resizes the replica pool toward
+    # ceil(ongoing / target) clamped to [min, max]
+    autoscaling_config: dict | None = None
 
     def options(self, name=None, num_replicas=None, route_prefix=None,
-                ray_actor_options=None, **_):
+                ray_actor_options=None, autoscaling_config=None, **_):
         return replace(
             self,
             name=name or self.name,
             num_replicas=num_replicas or self.num_replicas,
             route_prefix=route_prefix or self.route_prefix,
             ray_actor_options=ray_actor_options or self.ray_actor_options,
+            autoscaling_config=autoscaling_config or self.autoscaling_config,
         )
 
     def bind(self, *args, **kwargs) -> "Application":
@@ -89,8 +95,10 @@ class _PredictorReplicaGroup:
 
     def __init__(self, app: Application):
         d = app.deployment
+        self._dep = d
         self.http_adapter = app.kwargs.get("http_adapter", pandas_read_json)
         rest = {k: v for k, v in app.kwargs.items() if k != "http_adapter"}
+        self._is_actor = False
         if app.args and hasattr(app.args[0], "from_checkpoint"):
             # PredictorDeployment.bind(PredictorCls, checkpoint, ...)
             predictor_cls, checkpoint = app.args[0], app.args[1]
@@ -101,30 +109,90 @@ class _PredictorReplicaGroup:
 
                 num_gpus = d.ray_actor_options.get("num_gpus", 0)
                 cls = ActorClass(_ActorReplica, num_gpus=num_gpus)
-                self.replicas = [cls.remote(predictor_cls, checkpoint, rest)
-                                 for _ in range(d.num_replicas)]
+                self._make_replica = lambda: cls.remote(predictor_cls,
+                                                        checkpoint, rest)
                 self._invoke = _invoke_actor
+                self._is_actor = True
             else:
-                self.replicas = [predictor_cls.from_checkpoint(checkpoint, **rest)
-                                 for _ in range(d.num_replicas)]
+                self._make_replica = lambda: predictor_cls.from_checkpoint(
+                    checkpoint, **rest)
                 self._invoke = lambda rep, data, kw: rep.predict(data, **kw)
         elif isinstance(d.func_or_class, type):
             # generic @serve.deployment class: instance is the handler
-            self.replicas = [d.func_or_class(*app.args, **rest)
-                             for _ in range(d.num_replicas)]
+            self._make_replica = lambda: d.func_or_class(*app.args, **rest)
             self._invoke = lambda rep, data, kw: rep(data, **kw)
         else:
             # @serve.deployment function
-            self.replicas = [d.func_or_class] * max(d.num_replicas, 1)
+            self._make_replica = lambda: d.func_or_class
             self._invoke = lambda rep, data, kw: rep(data, **kw)
-        self._rr = itertools.cycle(range(len(self.replicas)))
+        self.replicas = [self._make_replica()
+                         for _ in range(max(d.num_replicas, 1))]
+        self._rr = itertools.count()
         self._locks = [threading.Lock() for _ in self.replicas]
+        self._pool_lock = threading.Lock()
+        self._ongoing = 0
+        self._stop = threading.Event()
+        if d.autoscaling_config:
+            threading.Thread(target=self._autoscale_loop, daemon=True).start()
+
+    # -- autoscaling ----------------------------------------------------- #
+    def _autoscale_loop(self):
+        cfg = self._dep.autoscaling_config
+        lo = int(cfg.get("min_replicas", 1))
+        hi = int(cfg.get("max_replicas", max(len(self.replicas), lo)))
+        target = float(cfg.get("target_num_ongoing_requests_per_replica", 1))
+        interval = float(cfg.get("interval_s", 2.0))
+        import math
+
+        while not self._stop.wait(interval):
+            desired = max(lo, min(hi, math.ceil(self._ongoing / max(target, 1e-9))))
+            with self._pool_lock:
+                cur = len(self.replicas)
+                if desired > cur:
+                    for _ in range(desired - cur):
+                        self.replicas.append(self._make_replica())
+                        self._locks.append(threading.Lock())
+                elif desired < cur:
+                    dead = self.replicas[desired:]
+                    del self.replicas[desired:]
+                    del self._locks[desired:]
+                    for rep in dead:
+                        self._kill_replica(rep)
+
+    def _kill_replica(self, rep):
+        if self._is_actor:
+            try:
+                rep._kill()
+            except Exception:
+                pass
+
+    def num_replicas(self) -> int:
+        return len(self.replicas)
+
+    def shutdown(self):
+        """Drain in-flight requests (acquire every replica lock) and stop."""
+        self._stop.set()
+        with self._pool_lock:
+            for lk, rep in zip(list(self._locks), list(self.replicas)):
+                with lk:
+                    self._kill_replica(rep)
+            self.replicas = []
+            self._locks = []
 
     def handle(self, body: bytes, query_kwargs: dict):
         data = self.http_adapter(body)
-        i = next(self._rr)
-        with self._locks[i]:
-            out = self._invoke(self.replicas[i], data, query_kwargs)
+        self._ongoing += 1
+        try:
+            with self._pool_lock:
+                n = len(self.replicas)
+                if n == 0:
+                    raise RuntimeError("deployment shut down")
+                i = next(self._rr) % n
+                lock, rep = self._locks[i], self.replicas[i]
+            with lock:
+                out = self._invoke(rep, data, query_kwargs)
+        finally:
+            self._ongoing -= 1
         if isinstance(out, pd.DataFrame):
             return out.to_dict(orient="records")
         import numpy as np
@@ -210,12 +278,19 @@ class _Handler(BaseHTTPRequestHandler):
 
 def run(app: Application, host: str = "127.0.0.1", port: int = 8000,
         _blocking: bool = False, **_):
-    """Deploy the application and (idempotently) start the HTTP proxy."""
+    """Deploy the application and (idempotently) start the HTTP proxy.
+
+    Graceful update: re-running a route builds the NEW replica group first,
+    atomically swaps it into the router, then drains and kills the old
+    group's replicas — in-flight requests finish on the old replicas."""
     d = app.deployment
     group = _PredictorReplicaGroup(app)
     route = d.route_prefix.rstrip("/") or "/"
+    old = _state.routes.get(route)
     _state.routes[route] = group
     _state.apps[route] = app
+    if old is not None:
+        threading.Thread(target=old.shutdown, daemon=True).start()
     if _state.server is None:
         _state.server = ThreadingHTTPServer((host, port), _Handler)
         _state.thread = threading.Thread(target=_state.server.serve_forever,
@@ -233,9 +308,12 @@ class RayServeHandle:
         # direct replicas[i].predict(df) call broke for actor-backed replicas
         # (ActorMethod has no __call__) and bypassed the HTTP path's
         # serialization.
-        i = next(self.group._rr)
-        with self.group._locks[i]:
-            return self.group._invoke(self.group.replicas[i], df, {})
+        g = self.group
+        with g._pool_lock:
+            i = next(g._rr) % len(g.replicas)
+            lock, rep = g._locks[i], g.replicas[i]
+        with lock:
+            return g._invoke(rep, df, {})
 
 
 def get_deployment(name: str):
