@@ -127,3 +127,22 @@ def test_gemm_nt256_matches_fp32_ref(shape, monkeypatch):
     got = ops.gemm_nt(a, b)
     ref = a.float() @ b.float().t()
     assert _relerr(got, ref) < 3e-2
+
+
+@gpu
+@pytest.mark.parametrize("shape", [
+    (1024, 768, 768),      # attn_o fwd @ B=2
+    (1024, 2304, 768),     # qkv fwd
+    (512, 64, 64),         # single skinny tile
+    (256, 128, 128),       # skinny edge: M=256 (one tile row)
+])
+def test_gemm_nt_skinny_matches_fp32_ref(shape, monkeypatch):
+    """The 256x64 skinny NT kernel, forced via MI355_GEMM_TILE=skinny."""
+    monkeypatch.setenv("MI355_GEMM_TILE", "skinny")
+    M, N, K = shape
+    torch.manual_seed(6)
+    a = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
+    got = ops.gemm_nt(a, b)
+    ref = a.float() @ b.float().t()
+    assert _relerr(got, ref) < 3e-2
