@@ -34,14 +34,9 @@ __global__ void decode_attn_kernel(const T* __restrict__ q, const T* __restrict_
   const float* brow = bias ? bias + (long)(bias_b_stride ? b : 0) * H * L + (long)h * L : nullptr;
 
   // q into registers: every lane holds the full row (D<=64 floats, L1 broadcast)
-  constexpr int KV = sizeof(T) == 4 ? 4 : 8;
   float qreg[D];
 #pragma unroll
-  for (int d = 0; d < D; d += KV) {
-    VecT<T, KV> qv = vload<T, KV>(qrow + d);
-#pragma unroll
-    for (int u = 0; u < KV; ++u) qreg[d + u] = DT<T>::to_f32(qv.data[u]);
-  }
+  for (int d = 0; d < D; ++d) qreg[d] = DT<T>::to_f32(qrow[d]);
 
   __shared__ float p_lds[4][WAVE];  // up to 4 waves per block
   const int wid = threadIdx.x / WAVE;
@@ -51,36 +46,19 @@ __global__ void decode_attn_kernel(const T* __restrict__ q, const T* __restrict_
 #pragma unroll
   for (int j = 0; j < 8; ++j) oacc[j] = 0.f;
 
-  // Software-pipelined K tiles: the NEXT tile's K rows are issued before
-  // this tile's V pass, so the two HBM phases overlap instead of
-  // serializing (each ~900-cycle round trip; the kernel measured ~72% of
-  // achievable bandwidth without this).
-  constexpr int NKV = D / KV;
-  VecT<T, KV> kcur[NKV], knxt[NKV];
-  {
-    const int j = lane;
-    const T* krow = kbase + (long)min(j, L - 1) * kst.ss;
-#pragma unroll
-    for (int d = 0; d < NKV; ++d) kcur[d] = vload<T, KV>(krow + d * KV);
-  }
-
   for (int j0 = 0; j0 < L; j0 += WAVE) {
     const int j = j0 + lane;
-    // issue next tile's K loads FIRST (consumed one iteration later)
-    if (j0 + WAVE < L) {
-      const int jn = min(j0 + WAVE + lane, L - 1);
-      const T* krow = kbase + (long)jn * kst.ss;
-#pragma unroll
-      for (int d = 0; d < NKV; ++d) knxt[d] = vload<T, KV>(krow + d * KV);
-    }
     float s = -INFINITY;
     if (j < L) {
+      const T* krow = kbase + (long)j * kst.ss;
       float acc = 0.f;
+      constexpr int KV = sizeof(T) == 4 ? 4 : 8;
 #pragma unroll
-      for (int d = 0; d < NKV; ++d)
+      for (int d = 0; d < D; d += KV) {
+        VecT<T, KV> kv = vload<T, KV>(krow + d);
 #pragma unroll
-        for (int u = 0; u < KV; ++u)
-          acc += qreg[d * KV + u] * DT<T>::to_f32(kcur[d].data[u]);
+        for (int u = 0; u < KV; ++u) acc += qreg[d + u] * DT<T>::to_f32(kv.data[u]);
+      }
       s = acc + (brow ? brow[j] : 0.f) + (kpad ? kpad[(long)b * L + j] : 0.f);
     }
     const float tile_max = wave_reduce_max(s);
@@ -109,8 +87,6 @@ __global__ void decode_attn_kernel(const T* __restrict__ q, const T* __restrict_
       }
     }
     m = mn;
-#pragma unroll
-    for (int d = 0; d < NKV; ++d) kcur[d] = knxt[d];
   }
   // fold the 8 key-groups: lanes l and l^8,l^16,l^32 share the same dim chunk
 #pragma unroll
