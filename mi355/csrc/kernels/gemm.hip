@@ -445,6 +445,10 @@ __global__ __launch_bounds__(512, 2) void gemm_nt256_kernel(
       __builtin_amdgcn_s_setprio(0);                                          \
       __builtin_amdgcn_s_barrier();                                           \
     } while (0)
+    // (A one-barrier-per-phase variant measured 632 vs 682 TF on
+    // ff_wi@8192 — the WAIT_ANY time is glds ARRIVAL, not barrier
+    // overhead; the paired barriers help the wave-pairing as the guide's
+    // template says.  PMC: profiles/r02_gemm256_pmc.md.)
 
     GEMM256_PHASE(0, 6);
     GEMM256_PHASE(1, 4);
