@@ -29,11 +29,12 @@ def init_distributed(backend: str | None = None, timeout_s: int = 300) -> int:
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29500")
     if backend == "nccl":
-        idx = local_rank()
-        if os.environ.get("MI355_SHARE_GPU") == "1":
-            # multi-rank-per-GPU smoke inside a 1-GPU lease (de-risk only)
-            idx = idx % torch.cuda.device_count()
-        torch.cuda.set_device(idx)
+        # LOCAL_RANK modulo the VISIBLE device count: trainer workers are
+        # pinned via HIP_VISIBLE_DEVICES (each sees ONE device, index 0,
+        # while LOCAL_RANK stays the global rank), torchrun ranks are
+        # unpinned (count == world). Without the modulo every pinned rank>0
+        # died on an invalid ordinal before rendezvous.
+        torch.cuda.set_device(local_rank() % max(1, torch.cuda.device_count()))
     dist.init_process_group(backend=backend,
                             timeout=datetime.timedelta(seconds=timeout_s))
     return dist.get_rank()

@@ -72,6 +72,9 @@ def _worker_entry(rank, world, port, use_gpu, loop_bytes, loop_config,
         from ..parallel import init_distributed
 
         if world > 1:
+            print(f"rendezvous: world={world} port={port} "
+                  f"visible_gpus={os.environ.get('HIP_VISIBLE_DEVICES', '<all>')}",
+                  flush=True)
             init_distributed()
         shards = {name: Dataset(tbls) for name, tbls in (shard_tables or {}).items()}
         ctx = _session_mod.TrainContext(world_rank=rank, world_size=world,
@@ -128,7 +131,7 @@ class DataParallelTrainer:
                 else:
                     ds = self.preprocessor.transform(ds)
                 datasets[name] = ds
-        n = self.scaling_config.num_workers
+        n = getattr(self, "_effective_world", None) or self.scaling_config.num_workers
         shards_per_rank = []
         for rank in range(n):
             shard_tables = {}
@@ -163,6 +166,23 @@ class DataParallelTrainer:
     def _fit_once(self, resume_dir=None) -> Result:
         world = max(self.scaling_config.num_workers, 1)
         use_gpu = self.scaling_config.use_gpu
+        if use_gpu:
+            import torch
+
+            n_gpus = torch.cuda.device_count()
+            if n_gpus == 0:
+                raise RuntimeError(
+                    "ScalingConfig(use_gpu=True) but no GPU is visible")
+            if world > n_gpus:
+                # RCCL enforces one rank per device (a 2-GPU-worker job on a
+                # 1-GPU box would hang at rendezvous); clamp like a
+                # resource-limited scheduler rather than waiting forever on
+                # resources that cannot appear on this node
+                print(f"[mi355 trainer] requested {world} GPU workers but "
+                      f"only {n_gpus} GPU(s) are visible — running "
+                      f"{n_gpus}-worker data parallel")
+                world = n_gpus
+        self._effective_world = world
         run_dir = self.run_config.run_dir()
         os.makedirs(run_dir, exist_ok=True)
 
