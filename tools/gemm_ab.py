@@ -23,12 +23,20 @@ from mi355 import ops  # noqa: E402
 # fwd NT: y[M,N] = x[M,K] @ W[N,K]^T
 # dX  NN: dx[M,K] = dy[M,N] @ W[N,K->K,N... stored [N,K]] — contraction N
 # dW  TN: dw[N,K] = dy[M,N]^T @ x[M,K] — contraction M
-def shape_set(M: int) -> list[tuple[str, str, int, int, int]]:
+SHAPES = {
+    # flan-t5-base: d_model 768, inner 768, d_ff 2048
+    "base": [("qkv", 2304, 768), ("attn_o", 768, 768), ("cross_kv", 1536, 768),
+             ("ff_wi", 4096, 768), ("ff_wo", 768, 2048), ("lm_head", 32128, 768)],
+    # flan-t5-large: d_model 1024, inner 1024 (16 heads x 64), d_ff 2816
+    "large": [("qkv", 3072, 1024), ("attn_o", 1024, 1024),
+              ("cross_kv", 2048, 1024), ("ff_wi", 5632, 1024),
+              ("ff_wo", 1024, 2816), ("lm_head", 32128, 1024)],
+}
+
+
+def shape_set(M: int, model: str = "base") -> list[tuple[str, str, int, int, int]]:
     out = []
-    for name, N, K in [
-        ("qkv", 2304, 768), ("attn_o", 768, 768), ("cross_kv", 1536, 768),
-        ("ff_wi", 4096, 768), ("ff_wo", 768, 2048), ("lm_head", 32128, 768),
-    ]:
+    for name, N, K in SHAPES[model]:
         out.append((f"{name}.fwd", "nt", M, N, K))
         out.append((f"{name}.dx", "nn", M, K, N))
         out.append((f"{name}.dw", "tn", N, K, M))
@@ -88,6 +96,7 @@ def main():
     ap.add_argument("--iters", type=int, default=50)
     ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp16"])
     ap.add_argument("--ms", type=int, nargs="*", default=[1024, 8192, 256])
+    ap.add_argument("--model", default="base", choices=["base", "large"])
     ap.add_argument("--json", default=None)
     args = ap.parse_args()
     dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float16
@@ -99,7 +108,7 @@ def main():
           "hip TF | blaslt TF | winner |")
     print("|---|---|---|---|---|---|---|---|---|---|")
     for M in args.ms:
-        for tag, op, rows, cols, kc in shape_set(M):
+        for tag, op, rows, cols, kc in shape_set(M, args.model):
             flop = 2.0 * rows * cols * kc
             t_hip = time_fn(run_op(op, rows, cols, kc, dtype, True), args.iters)
             t_lib = time_fn(run_op(op, rows, cols, kc, dtype, False), args.iters)
