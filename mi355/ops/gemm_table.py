@@ -33,6 +33,12 @@ _WINS: set[tuple[str, int, int, int]] = {
     ('tn', 1536, 768, 8192),    # cross_kv.dw @ B=16      (57 vs 62 us)
     ('tn', 768, 2048, 8192),    # ff_wo.dw   @ B=16       (61 vs 63 us)
     ('tn', 4096, 768, 256),     # ff_wi.dw   @ decode     (tie 8.7 us)
+    # flan-t5-large (d_model 1024, d_ff 2816) — gpurun_out/gemm_ab_large.md
+    ('nn', 1024, 1024, 32128),  # lm_head.dx @ train B=2  (117 vs 178 us)
+    ('nn', 256, 1024, 32128),   # lm_head.dx @ M=256      (45 vs 150 us)
+    ('nn', 1024, 1024, 5632),   # ff_wi.dx   @ train B=2  (32.9 vs 33.7 us)
+    ('nn', 256, 1024, 5632),    # ff_wi.dx   @ M=256      (18.7 vs 21.9 us)
+    ('tn', 1024, 2816, 256),    # ff_wo.dw   @ M=256      (tie 8.7 us)
 }
 
 
