@@ -75,3 +75,21 @@ def test_bench_hybrid_two_graph_path():
     d = json.loads(line)
     assert d["config"]["graphs"] is True, out.stdout + out.stderr
     assert d["value"] > 0
+
+
+@pytest.mark.gpu
+def test_gpu_workers_clamp_to_device_count(tmp_path):
+    """ScalingConfig asking for more GPU workers than visible devices must
+    clamp (RCCL enforces one rank per device; pre-fix this hung at
+    rendezvous for 301 s) and still complete training."""
+    import torch
+
+    n = torch.cuda.device_count()
+    trainer = HuggingFaceTrainer(
+        trainer_init_per_worker=gpu_trainer_init,
+        scaling_config=ScalingConfig(num_workers=n + 1, use_gpu=True),
+        datasets={"train": synthetic(16), "evaluation": synthetic(8)},
+        run_config=RunConfig(name="clamp-test", local_dir=str(tmp_path)),
+    )
+    result = trainer.fit()
+    assert result.error is None, result.error
