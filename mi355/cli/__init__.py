@@ -75,6 +75,21 @@ def cmd_flags(_args):
     return 0
 
 
+def cmd_dashboard(args):
+    """Start the web dashboard standalone (C25; init() also starts it)."""
+    from ..dashboard import start_dashboard
+
+    host, port = start_dashboard(port=args.port)
+    print(f"mi355 dashboard at http://{host}:{port} (ctrl-c to stop)")
+    import time
+
+    try:
+        while True:
+            time.sleep(3600)
+    except KeyboardInterrupt:
+        return 0
+
+
 def cmd_gpus(_args):
     try:
         out = subprocess.run(["rocm-smi", "--showuse", "--showmemuse"],
@@ -95,6 +110,9 @@ def main(argv=None):
     s = jobsub.add_parser("submit")
     s.add_argument("spec")
     s.set_defaults(fn=cmd_submit)
+    dash = sub.add_parser("dashboard")
+    dash.add_argument("--port", type=int, default=None)
+    dash.set_defaults(fn=cmd_dashboard)
     st = sub.add_parser("status")
     st.add_argument("--dir", default=None)
     st.set_defaults(fn=cmd_status)
