@@ -52,6 +52,11 @@ std::vector<torch::Tensor> dwconv3x3_wrw(torch::Tensor dout, torch::Tensor x);
 torch::Tensor patch_embed_s1_fwd(torch::Tensor x, torch::Tensor w,
                                  c10::optional<torch::Tensor> bias);
 std::vector<torch::Tensor> patch_embed_s1_wrw(torch::Tensor dout, torch::Tensor x);
+std::vector<torch::Tensor> seg_ce_fwd(torch::Tensor logits, torch::Tensor labels,
+                                      long Ho, long Wo, long ignore_index);
+torch::Tensor seg_ce_bwd(torch::Tensor logits, torch::Tensor labels,
+                         torch::Tensor lse, torch::Tensor gout,
+                         torch::Tensor nvalid, long ignore_index);
 torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b, long splitk);
 torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b, long splitk);
 std::vector<torch::Tensor> attn_fwd_v2(torch::Tensor q, torch::Tensor k, torch::Tensor v,
@@ -102,6 +107,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("patch_embed_s1_fwd", &patch_embed_s1_fwd, py::arg("x"), py::arg("w"),
         py::arg("bias") = c10::nullopt);
   m.def("patch_embed_s1_wrw", &patch_embed_s1_wrw);
+  m.def("seg_ce_fwd", &seg_ce_fwd);
+  m.def("seg_ce_bwd", &seg_ce_bwd);
   m.def("gemm_nn", &gemm_nn, py::arg("a"), py::arg("b"), py::arg("splitk") = 0);
   m.def("gemm_tn", &gemm_tn, py::arg("a"), py::arg("b"), py::arg("splitk") = 0);
   m.def("attn_fwd_v2", &attn_fwd_v2);

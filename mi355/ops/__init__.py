@@ -315,6 +315,45 @@ def patch_embed_s1(x: torch.Tensor, w: torch.Tensor, bias=None) -> torch.Tensor:
     return torch.nn.functional.conv2d(x, w, bias, stride=4, padding=3)
 
 
+class _SegCE(torch.autograd.Function):
+    """Fused upsampled pixelwise cross-entropy: CE of
+    bilinear-upsampled low-res logits against full-res labels WITHOUT
+    materializing the (B,C,H,W) upsampled tensor (the torch path's
+    interpolate+softmax+nll was 29% of the SegFormer train step)."""
+
+    @staticmethod
+    def forward(ctx, logits, labels, Ho, Wo, ignore_index):
+        loss_sum, lse, nvalid = _load_ext().seg_ce_fwd(
+            logits, labels, Ho, Wo, ignore_index)
+        ctx.save_for_backward(logits, labels, lse, nvalid)
+        ctx.ignore_index = ignore_index
+        return loss_sum / nvalid.clamp(min=1.0)
+
+    @staticmethod
+    def backward(ctx, gout):
+        logits, labels, lse, nvalid = ctx.saved_tensors
+        dlow = _load_ext().seg_ce_bwd(logits, labels, lse,
+                                      gout.contiguous().float(), nvalid,
+                                      ctx.ignore_index)
+        return dlow.to(logits.dtype), None, None, None, None
+
+
+def seg_cross_entropy_upsampled(logits: torch.Tensor, labels: torch.Tensor,
+                                ignore_index: int = 255) -> torch.Tensor:
+    """cross_entropy(interpolate(logits, labels.shape[-2:], bilinear),
+    labels, ignore_index) without the upsampled materialization."""
+    Ho, Wo = labels.shape[-2], labels.shape[-1]
+    hi, wi = logits.shape[-2], logits.shape[-1]
+    # the bwd LDS patch caps the upsample ratio; fall back otherwise
+    if _use_hip(logits) and Ho >= 3 * hi and Wo >= 3 * wi             and logits.shape[1] <= 256:
+        return _SegCE.apply(logits, labels.long().contiguous(), Ho, Wo,
+                            ignore_index)
+    up = torch.nn.functional.interpolate(
+        logits.float(), size=(Ho, Wo), mode="bilinear", align_corners=False)
+    return torch.nn.functional.cross_entropy(up, labels.long(),
+                                             ignore_index=ignore_index)
+
+
 def gemm_nt(a: torch.Tensor, b: torch.Tensor, splitk: int = 0) -> torch.Tensor:
     """C[M,N] = a[M,K] @ b[N,K]^T on the hand-written MFMA kernel."""
     return _load_ext().gemm_nt(a, b, splitk)

@@ -190,3 +190,33 @@ def test_visualization_helpers():
     assert len(idx) == 4 and len(set(idx)) == 4
     fig = display_example_images([img] * 4, n=2)
     assert fig is not None
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [(2, 150, 128, 128, 512),
+                                   (1, 19, 64, 64, 256)])
+def test_seg_ce_fused_matches_torch(shape):
+    """Fused upsample+CE (K18) vs the materializing torch path: loss and
+    dLogits, incl. ignore_index pixels."""
+    import torch.nn.functional as F
+    from mi355 import ops
+
+    b, c, h, w, H = shape
+    torch.manual_seed(3)
+    logits = torch.randn(b, c, h, w, device="cuda", dtype=torch.bfloat16,
+                         requires_grad=True)
+    labels = torch.randint(0, c, (b, H, H), device="cuda")
+    labels[:, :7] = 255  # ignore region
+    loss = ops.seg_cross_entropy_upsampled(logits, labels, ignore_index=255)
+    loss.backward()
+    got_grad = logits.grad.clone()
+
+    l32 = logits.detach().float().requires_grad_(True)
+    up = F.interpolate(l32, size=(H, H), mode="bilinear", align_corners=False)
+    ref = F.cross_entropy(up, labels.long(), ignore_index=255)
+    ref.backward()
+
+    assert abs(loss.item() - ref.item()) / abs(ref.item()) < 2e-3
+    gerr = ((got_grad.float() - l32.grad).abs().max() /
+            (l32.grad.abs().max() + 1e-9)).item()
+    assert gerr < 2e-2, gerr
