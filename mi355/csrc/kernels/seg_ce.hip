@@ -97,7 +97,7 @@ __global__ void seg_ce_fwd_kernel(const T* __restrict__ logits,
 
 // bwd: grid (tiles_x, tiles_y, B); 256 threads each own a 2x2 sub-block of
 // the 32x32 output tile.  dlow32 must be zero-initialized.
-template <typename T, int OT /*32*/>
+template <typename T, int OT /*16*/>
 __global__ void seg_ce_bwd_kernel(const T* __restrict__ logits,
                                   const long* __restrict__ labels,
                                   const float* __restrict__ lse,
@@ -106,7 +106,7 @@ __global__ void seg_ce_bwd_kernel(const T* __restrict__ logits,
                                   float* __restrict__ dlow32,
                                   int B, int C, int Hi, int Wi, int Ho, int Wo,
                                   long ignore_index) {
-  extern __shared__ float acc[];  // [C][ph][pw]
+  extern __shared__ float acc[];  // [C][ph][pw] fp32 accum + bf16 patch
   const int b = blockIdx.z;
   const int oy0 = blockIdx.y * OT, ox0 = blockIdx.x * OT;
   // low-res patch bounds covering this tile's taps
@@ -116,24 +116,22 @@ __global__ void seg_ce_bwd_kernel(const T* __restrict__ logits,
   const int py0 = t00.y0, px0 = t00.x0;
   const int ph = t11.y1 - py0 + 1, pw = t11.x1 - px0 + 1;
   const int pelems = C * ph * pw;
+  const long plane = (long)Hi * Wi;
+  const T* base = logits + (long)b * C * plane;
   for (int i = threadIdx.x; i < pelems; i += blockDim.x) acc[i] = 0.f;
   __syncthreads();
 
   const float gs = gout[0] / fmaxf(nvalid[0], 1.f);
-  const long plane = (long)Hi * Wi;
-  const T* base = logits + (long)b * C * plane;
 
-  // thread t -> 2x2 output pixels at (oy0 + 2*(t/16) .. +1, ox0 + 2*(t%16)..)
-  const int ty = (threadIdx.x >> 4) * 2, tx = (threadIdx.x & 15) * 2;
-#pragma unroll
-  for (int dy = 0; dy < 2; ++dy)
-#pragma unroll
-    for (int dx = 0; dx < 2; ++dx) {
-      const int y = oy0 + ty + dy, x = ox0 + tx + dx;
-      if (y >= Ho || x >= Wo) continue;
-      const long pidx = ((long)b * Ho + y) * Wo + x;
-      const long lab = labels[pidx];
-      if (lab == ignore_index || lab < 0 || lab >= C) continue;
+  // one output pixel per thread (OT=16: 256 threads = the whole tile);
+  // small tiles keep the LDS patch ~6 KiB so 4+ blocks/CU hide the
+  // read->exp->ds_add dependency chains that made a 32-tile version
+  // (1 block/CU) latency-bound
+  {
+    const int y = oy0 + (threadIdx.x >> 4), x = ox0 + (threadIdx.x & 15);
+    const long pidx = ((long)b * Ho + y) * Wo + x;
+    const long lab = (y < Ho && x < Wo) ? labels[pidx] : ignore_index;
+    if (y < Ho && x < Wo && lab != ignore_index && lab >= 0 && lab < C) {
       const Tap t = tap_at(y, x, Hi, Wi, Ho, Wo);
       const long o00 = (long)t.y0 * Wi + t.x0, o01 = (long)t.y0 * Wi + t.x1;
       const long o10 = (long)t.y1 * Wi + t.x0, o11 = (long)t.y1 * Wi + t.x1;
@@ -142,7 +140,12 @@ __global__ void seg_ce_bwd_kernel(const T* __restrict__ logits,
       const int a10 = (t.y1 - py0) * pw + (t.x0 - px0);
       const int a11 = (t.y1 - py0) * pw + (t.x1 - px0);
       const float l = lse[pidx];
-      for (int c = 0; c < C; ++c) {
+      // channel start staggered per lane: in lockstep-c every lane hammers
+      // the SAME channel plane (LDS same-address atomic serialize);
+      // staggered, lanes touch different planes
+      const int c0 = threadIdx.x % C;
+      for (int cc = 0; cc < C; ++cc) {
+        const int c = (cc + c0 >= C) ? cc + c0 - C : cc + c0;
         const T* p = base + (long)c * plane;
         const float z = t.w00 * DT<T>::to_f32(p[o00]) +
                         t.w01 * DT<T>::to_f32(p[o01]) +
@@ -150,12 +153,15 @@ __global__ void seg_ce_bwd_kernel(const T* __restrict__ logits,
                         t.w11 * DT<T>::to_f32(p[o11]);
         const float g = (__expf(z - l) - (c == (int)lab ? 1.f : 0.f)) * gs;
         float* ac = acc + (long)c * ph * pw;
-        atomicAdd(&ac[a00], t.w00 * g);
-        atomicAdd(&ac[a01], t.w01 * g);
-        atomicAdd(&ac[a10], t.w10 * g);
-        atomicAdd(&ac[a11], t.w11 * g);
+        // unsafeAtomicAdd = HW ds_add/global_add_f32 (plain atomicAdd on
+        // fp32 compiles to a CAS loop on ROCm)
+        unsafeAtomicAdd(&ac[a00], t.w00 * g);
+        unsafeAtomicAdd(&ac[a01], t.w01 * g);
+        unsafeAtomicAdd(&ac[a10], t.w10 * g);
+        unsafeAtomicAdd(&ac[a11], t.w11 * g);
       }
     }
+  }
   __syncthreads();
   // patch -> global fp32 dlow (borders shared with neighbor tiles: atomics)
   float* dbase = dlow32 + (long)b * C * plane;
@@ -164,7 +170,7 @@ __global__ void seg_ce_bwd_kernel(const T* __restrict__ logits,
     if (v != 0.f) {
       const int c = i / (ph * pw);
       const int r = (i / pw) % ph, q = i % pw;
-      atomicAdd(&dbase[(long)c * plane + (long)(py0 + r) * Wi + (px0 + q)], v);
+      unsafeAtomicAdd(&dbase[(long)c * plane + (long)(py0 + r) * Wi + (px0 + q)], v);
     }
   }
 }
@@ -206,7 +212,7 @@ torch::Tensor seg_ce_bwd(torch::Tensor logits, torch::Tensor labels,
   auto lb = labels.contiguous();
   const int B = lc.size(0), C = lc.size(1), Hi = lc.size(2), Wi = lc.size(3);
   const int Ho = lse.size(1), Wo = lse.size(2);
-  constexpr int OT = 32;
+  constexpr int OT = 16;
   // patch bound check: the 32x32 tile's taps must fit the dynamic-LDS cap
   const int ph_max = (int)((OT + 0.5f) * Hi / Ho + 2.f);
   const int pw_max = (int)((OT + 0.5f) * Wi / Wo + 2.f);

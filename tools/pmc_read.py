@@ -29,9 +29,11 @@ def main(path):
         JOIN rocpd_kernel_dispatch_{sfx} kd ON pe.{join_col} = kd.id
         JOIN rocpd_info_kernel_symbol_{sfx} ks ON kd.kernel_id = ks.id
         GROUP BY ks.display_name, pi.name ORDER BY 1, 2""").fetchall()
+    import re
     cur = None
     for name, ctr, val, n in rows:
-        short = name.split("(")[0][:70]
+        short = re.sub(r"\(anonymous namespace\)::", "", name)
+        short = re.sub(r"^void ", "", short).split("(")[0][:70]
         if short != cur:
             print(f"\n{short}  ({n} dispatches)")
             cur = short
