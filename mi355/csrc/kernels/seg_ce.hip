@@ -95,9 +95,17 @@ __global__ void seg_ce_fwd_kernel(const T* __restrict__ logits,
   }
 }
 
-// bwd: grid (tiles_x, tiles_y, B); 256 threads each own a 2x2 sub-block of
-// the 32x32 output tile.  dlow32 must be zero-initialized.
-template <typename T, int OT /*16*/>
+// bwd: each thread owns a 4x4 output-pixel QUAD.  At upsample ratio >= 3
+// a quad's 16 pixels' bilinear taps all fall in one 3x3 low-res cell
+// window, so per channel the thread loads 9 cells once, accumulates all 16
+// pixels' weighted gradients in a 3x3 REGISTER array (tap-cell choice done
+// with selects — no runtime register indexing, guide rule 20), and issues
+// only 9 ds_adds.  The first version did 4 LDS atomics per (pixel,
+// channel): PMC measured 94% SQ_WAIT_INST_ANY — LDS atomics are per-lane
+// serialized RMW, and those 1.2G of them were the whole 6.2 ms kernel.
+// Block = 256 threads = 64x64 output tile; channels processed in chunks so
+// the [CC][ph][pw] fp32 patch fits LDS.  dlow32 must be zero-initialized.
+template <typename T, int OT /*64*/>
 __global__ void seg_ce_bwd_kernel(const T* __restrict__ logits,
                                   const long* __restrict__ labels,
                                   const float* __restrict__ lse,
@@ -105,73 +113,120 @@ __global__ void seg_ce_bwd_kernel(const T* __restrict__ logits,
                                   const float* __restrict__ nvalid,
                                   float* __restrict__ dlow32,
                                   int B, int C, int Hi, int Wi, int Ho, int Wo,
-                                  long ignore_index) {
-  extern __shared__ float acc[];  // [C][ph][pw] fp32 accum + bf16 patch
+                                  long ignore_index, int CC) {
+  extern __shared__ float acc[];  // [CC][ph][pw]
   const int b = blockIdx.z;
   const int oy0 = blockIdx.y * OT, ox0 = blockIdx.x * OT;
-  // low-res patch bounds covering this tile's taps
-  const Tap t00 = tap_at(oy0, ox0, Hi, Wi, Ho, Wo);
-  const Tap t11 = tap_at(min(oy0 + OT - 1, Ho - 1), min(ox0 + OT - 1, Wo - 1),
-                         Hi, Wi, Ho, Wo);
-  const int py0 = t00.y0, px0 = t00.x0;
-  const int ph = t11.y1 - py0 + 1, pw = t11.x1 - px0 + 1;
-  const int pelems = C * ph * pw;
+  const Tap tc00 = tap_at(oy0, ox0, Hi, Wi, Ho, Wo);
+  const Tap tc11 = tap_at(min(oy0 + OT - 1, Ho - 1), min(ox0 + OT - 1, Wo - 1),
+                          Hi, Wi, Ho, Wo);
+  const int py0 = tc00.y0, px0 = tc00.x0;
+  const int ph = tc11.y1 - py0 + 1, pw = tc11.x1 - px0 + 1;
   const long plane = (long)Hi * Wi;
   const T* base = logits + (long)b * C * plane;
-  for (int i = threadIdx.x; i < pelems; i += blockDim.x) acc[i] = 0.f;
-  __syncthreads();
-
   const float gs = gout[0] / fmaxf(nvalid[0], 1.f);
 
-  // one output pixel per thread (OT=16: 256 threads = the whole tile);
-  // small tiles keep the LDS patch ~6 KiB so 4+ blocks/CU hide the
-  // read->exp->ds_add dependency chains that made a 32-tile version
-  // (1 block/CU) latency-bound
-  {
-    const int y = oy0 + (threadIdx.x >> 4), x = ox0 + (threadIdx.x & 15);
+  // quad base: 16x16 quads of 4x4 pixels
+  const int qy0 = oy0 + (threadIdx.x >> 4) * 4;
+  const int qx0 = ox0 + (threadIdx.x & 15) * 4;
+  const Tap tq = tap_at(qy0, qx0, Hi, Wi, Ho, Wo);
+  const int cby = tq.y0, cbx = tq.x0;  // 3x3 cell window base
+
+  // per-pixel constants (all arrays indexed by the UNROLLED px loop only)
+  float w00[16], w01[16], w10[16], w11[16], lse16[16];
+  int lab16[16], ry0_16[16], rx0_16[16], ry1_16[16], rx1_16[16];
+  bool ok16[16];
+#pragma unroll
+  for (int px = 0; px < 16; ++px) {
+    const int y = qy0 + (px >> 2), x = qx0 + (px & 3);
     const long pidx = ((long)b * Ho + y) * Wo + x;
-    const long lab = (y < Ho && x < Wo) ? labels[pidx] : ignore_index;
-    if (y < Ho && x < Wo && lab != ignore_index && lab >= 0 && lab < C) {
-      const Tap t = tap_at(y, x, Hi, Wi, Ho, Wo);
-      const long o00 = (long)t.y0 * Wi + t.x0, o01 = (long)t.y0 * Wi + t.x1;
-      const long o10 = (long)t.y1 * Wi + t.x0, o11 = (long)t.y1 * Wi + t.x1;
-      const int a00 = (t.y0 - py0) * pw + (t.x0 - px0);
-      const int a01 = (t.y0 - py0) * pw + (t.x1 - px0);
-      const int a10 = (t.y1 - py0) * pw + (t.x0 - px0);
-      const int a11 = (t.y1 - py0) * pw + (t.x1 - px0);
-      const float l = lse[pidx];
-      // channel start staggered per lane: in lockstep-c every lane hammers
-      // the SAME channel plane (LDS same-address atomic serialize);
-      // staggered, lanes touch different planes
-      const int c0 = threadIdx.x % C;
-      for (int cc = 0; cc < C; ++cc) {
-        const int c = (cc + c0 >= C) ? cc + c0 - C : cc + c0;
-        const T* p = base + (long)c * plane;
-        const float z = t.w00 * DT<T>::to_f32(p[o00]) +
-                        t.w01 * DT<T>::to_f32(p[o01]) +
-                        t.w10 * DT<T>::to_f32(p[o10]) +
-                        t.w11 * DT<T>::to_f32(p[o11]);
-        const float g = (__expf(z - l) - (c == (int)lab ? 1.f : 0.f)) * gs;
-        float* ac = acc + (long)c * ph * pw;
-        // unsafeAtomicAdd = HW ds_add/global_add_f32 (plain atomicAdd on
-        // fp32 compiles to a CAS loop on ROCm)
-        unsafeAtomicAdd(&ac[a00], t.w00 * g);
-        unsafeAtomicAdd(&ac[a01], t.w01 * g);
-        unsafeAtomicAdd(&ac[a10], t.w10 * g);
-        unsafeAtomicAdd(&ac[a11], t.w11 * g);
+    const bool inb = y < Ho && x < Wo;
+    const long lab = inb ? labels[pidx] : ignore_index;
+    ok16[px] = inb && lab != ignore_index && lab >= 0 && lab < C;
+    lab16[px] = (int)lab;
+    lse16[px] = ok16[px] ? lse[pidx] : 0.f;
+    const Tap t = tap_at(inb ? y : 0, inb ? x : 0, Hi, Wi, Ho, Wo);
+    w00[px] = t.w00; w01[px] = t.w01; w10[px] = t.w10; w11[px] = t.w11;
+    // window-relative tap cells; y1/x1 may EQUAL y0/x0 at the image edge
+    // (clamped taps) — masks below then route both weights to one cell
+    ry0_16[px] = min(max(t.y0 - cby, 0), 2);
+    rx0_16[px] = min(max(t.x0 - cbx, 0), 2);
+    ry1_16[px] = min(max(t.y1 - cby, 0), 2);
+    rx1_16[px] = min(max(t.x1 - cbx, 0), 2);
+  }
+  const int pb = (cby - py0) * pw + (cbx - px0);  // window base in the patch
+
+  for (int c0 = 0; c0 < C; c0 += CC) {
+    const int cn = min(CC, C - c0);
+    for (int i = threadIdx.x; i < cn * ph * pw; i += blockDim.x) acc[i] = 0.f;
+    __syncthreads();
+
+    for (int c = c0; c < c0 + cn; ++c) {
+      const T* p = base + (long)c * plane;
+      float c9[9];
+#pragma unroll
+      for (int dy = 0; dy < 3; ++dy)
+#pragma unroll
+        for (int dx = 0; dx < 3; ++dx)
+          c9[dy * 3 + dx] = DT<T>::to_f32(
+              p[(long)min(cby + dy, Hi - 1) * Wi + min(cbx + dx, Wi - 1)]);
+      float r9[9];
+#pragma unroll
+      for (int k = 0; k < 9; ++k) r9[k] = 0.f;
+#pragma unroll
+      for (int px = 0; px < 16; ++px) {
+        if (!ok16[px]) continue;
+        const int ry0 = ry0_16[px], rx0 = rx0_16[px];
+        const int ry1 = ry1_16[px], rx1 = rx1_16[px];
+        // 2x2 tap VALUES via cheap nested ternaries on the top-left cell
+        // (no runtime register indexing, guide rule 20).  This is exact even
+        // for edge-clamped taps: the 3x3 window is loaded with clamped input
+        // indices, so the neighbouring cell replicates the edge value.
+        const int by = min(ry0, 1), bx = min(rx0, 1);
+        const float v00 = by ? (bx ? c9[4] : c9[3]) : (bx ? c9[1] : c9[0]);
+        const float v01 = by ? (bx ? c9[5] : c9[4]) : (bx ? c9[2] : c9[1]);
+        const float v10 = by ? (bx ? c9[7] : c9[6]) : (bx ? c9[4] : c9[3]);
+        const float v11 = by ? (bx ? c9[8] : c9[7]) : (bx ? c9[5] : c9[4]);
+        const float z = w00[px] * v00 + w01[px] * v01 +
+                        w10[px] * v10 + w11[px] * v11;
+        const float g = (__expf(z - lse16[px]) -
+                         (c == lab16[px] ? 1.f : 0.f)) * gs;
+        const float g00 = w00[px] * g, g01 = w01[px] * g;
+        const float g10 = w10[px] * g, g11 = w11[px] * g;
+#pragma unroll
+        for (int k = 0; k < 9; ++k) {
+          const int dy = k / 3, dx = k % 3;
+          r9[k] += ((dy == ry0 && dx == rx0) ? g00 : 0.f) +
+                   ((dy == ry0 && dx == rx1) ? g01 : 0.f) +
+                   ((dy == ry1 && dx == rx0) ? g10 : 0.f) +
+                   ((dy == ry1 && dx == rx1) ? g11 : 0.f);
+        }
+      }
+      float* ac = acc + (long)(c - c0) * ph * pw + pb;
+#pragma unroll
+      for (int dy = 0; dy < 3; ++dy)
+#pragma unroll
+        for (int dx = 0; dx < 3; ++dx) {
+          const float v = r9[dy * 3 + dx];
+          // clamped taps collapse onto interior cells; window rows/cols
+          // beyond the patch carry zeros — skip them
+          if (v != 0.f && cby + dy - py0 < ph && cbx + dx - px0 < pw)
+            unsafeAtomicAdd(&ac[dy * pw + dx], v);
+        }
+    }
+    __syncthreads();
+    // chunk patch -> global fp32 dlow (tile borders overlap: atomics)
+    float* dbase = dlow32 + ((long)b * C + c0) * plane;
+    for (int i = threadIdx.x; i < cn * ph * pw; i += blockDim.x) {
+      const float v = acc[i];
+      if (v != 0.f) {
+        const int c = i / (ph * pw);
+        const int r = (i / pw) % ph, q = i % pw;
+        unsafeAtomicAdd(&dbase[(long)c * plane + (long)(py0 + r) * Wi +
+                               (px0 + q)], v);
       }
     }
-  }
-  __syncthreads();
-  // patch -> global fp32 dlow (borders shared with neighbor tiles: atomics)
-  float* dbase = dlow32 + (long)b * C * plane;
-  for (int i = threadIdx.x; i < pelems; i += blockDim.x) {
-    const float v = acc[i];
-    if (v != 0.f) {
-      const int c = i / (ph * pw);
-      const int r = (i / pw) % ph, q = i % pw;
-      unsafeAtomicAdd(&dbase[(long)c * plane + (long)(py0 + r) * Wi + (px0 + q)], v);
-    }
+    __syncthreads();
   }
 }
 
@@ -212,14 +267,14 @@ torch::Tensor seg_ce_bwd(torch::Tensor logits, torch::Tensor labels,
   auto lb = labels.contiguous();
   const int B = lc.size(0), C = lc.size(1), Hi = lc.size(2), Wi = lc.size(3);
   const int Ho = lse.size(1), Wo = lse.size(2);
-  constexpr int OT = 16;
-  // patch bound check: the 32x32 tile's taps must fit the dynamic-LDS cap
-  const int ph_max = (int)((OT + 0.5f) * Hi / Ho + 2.f);
-  const int pw_max = (int)((OT + 0.5f) * Wi / Wo + 2.f);
-  const long smem = (long)C * ph_max * pw_max * 4;
-  TORCH_CHECK(smem <= 160 * 1024,
-              "seg_ce_bwd: LDS patch too large (C=", C, " ph=", ph_max,
-              ") — dispatcher must fall back");
+  constexpr int OT = 64;
+  const int ph_max = (int)((OT + 0.5f) * Hi / Ho + 3.f);
+  const int pw_max = (int)((OT + 0.5f) * Wi / Wo + 3.f);
+  // channel chunk so the [CC][ph][pw] fp32 patch fits comfortably in LDS
+  const int CC = std::max(1, std::min(C, (int)(48 * 1024 /
+                                              (ph_max * pw_max * 4))));
+  const long smem = (long)CC * ph_max * pw_max * 4;
+  TORCH_CHECK(smem <= 160 * 1024, "seg_ce_bwd: LDS patch too large");
   auto dlow = torch::zeros_like(lc, lc.options().dtype(torch::kFloat));
   auto stream = at::hip::getCurrentHIPStream().stream();
   dim3 grid((Wo + OT - 1) / OT, (Ho + OT - 1) / OT, B);
@@ -228,7 +283,7 @@ torch::Tensor seg_ce_bwd(torch::Tensor logits, torch::Tensor labels,
                      stream, (const T*)lc.data_ptr(), lb.data_ptr<long>(),     \
                      lse.data_ptr<float>(), gout.data_ptr<float>(),            \
                      nvalid.data_ptr<float>(), dlow.data_ptr<float>(),         \
-                     B, C, Hi, Wi, Ho, Wo, ignore_index)
+                     B, C, Hi, Wi, Ho, Wo, ignore_index, CC)
   if (lc.scalar_type() == torch::kBFloat16) CASE(__hip_bfloat16);
   else if (lc.scalar_type() == torch::kHalf) CASE(__half);
   else if (lc.scalar_type() == torch::kFloat) CASE(float);
