@@ -3,7 +3,7 @@
 import sys
 import time
 
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
 import torch  # noqa: E402
 
 from mi355 import _C  # noqa: E402

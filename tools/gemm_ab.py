@@ -16,7 +16,7 @@ import sys
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
 from mi355 import ops  # noqa: E402
 
 # T5-base hot GEMM set: (tag, op, rows, cols, kc)

@@ -1,7 +1,7 @@
 """ASHA tuner smoke on real hardware (falls back to CPU): 2 trials,
 tiny T5, epoch-wise eval_loss reports driving the lockstep early stop."""
 import sys
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
 import torch
 import ray
 from ray import tune
