@@ -100,6 +100,91 @@ __global__ void add_rmsnorm_fwd_kernel(const T* __restrict__ x, const T* __restr
   }
 }
 
+// Single-pass add+RMSNorm for D <= threads*VEC: each thread owns ONE vector
+// chunk, s = x + drop(r) stays in REGISTERS across the sum-of-squares
+// reduction (the generic kernel re-reads s from global), and rows are
+// processed in pairs to overlap the two HBM load chains (same rationale as
+// rmsnorm_bwd_1p_kernel above).
+template <typename T, int VEC>
+__global__ void add_rmsnorm_fwd_1p_kernel(const T* __restrict__ x, const T* __restrict__ r,
+                                          T* __restrict__ s_out, T* __restrict__ y,
+                                          const T* __restrict__ w, float* __restrict__ inv_rms,
+                                          int rows, int D, float eps,
+                                          float p_drop, const long* __restrict__ seed_ptr,
+                                          long salt) {
+  __shared__ float red[16];
+  const long long seed = (seed_ptr ? *seed_ptr : 0) ^ salt;
+  const float keep_inv = p_drop > 0.f ? 1.f / (1.f - p_drop) : 1.f;
+  const int i = threadIdx.x * VEC;
+  const bool act = i < D;
+  VecT<T, VEC> wv;
+  if (act) wv = vload<T, VEC>(w + i);
+  const int half = (rows + 1) >> 1;
+  for (int p = blockIdx.x; p < half; p += gridDim.x) {
+    const int r0 = p, r1 = p + half;
+    const bool has1 = r1 < rows;
+    VecT<T, VEC> sv0, sv1;
+    float ss0 = 0.f, ss1 = 0.f;
+    if (act) {
+      VecT<T, VEC> xv0 = vload<T, VEC>(x + (long)r0 * D + i);
+      VecT<T, VEC> rv0 = vload<T, VEC>(r + (long)r0 * D + i);
+      VecT<T, VEC> xv1, rv1;
+      if (has1) {
+        xv1 = vload<T, VEC>(x + (long)r1 * D + i);
+        rv1 = vload<T, VEC>(r + (long)r1 * D + i);
+      }
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float rf = DT<T>::to_f32(rv0.data[j]);
+        if (p_drop > 0.f) {
+          const long long idx = (long long)r0 * D + i + j;
+          rf = (hash_u01(seed, idx) >= p_drop) ? rf * keep_inv : 0.f;
+        }
+        const float f = DT<T>::to_f32(xv0.data[j]) + rf;
+        sv0.data[j] = DT<T>::from_f32(f);
+        ss0 += f * f;
+      }
+      vstore<T, VEC>(s_out + (long)r0 * D + i, sv0);
+      if (has1) {
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+          float rf = DT<T>::to_f32(rv1.data[j]);
+          if (p_drop > 0.f) {
+            const long long idx = (long long)r1 * D + i + j;
+            rf = (hash_u01(seed, idx) >= p_drop) ? rf * keep_inv : 0.f;
+          }
+          const float f = DT<T>::to_f32(xv1.data[j]) + rf;
+          sv1.data[j] = DT<T>::from_f32(f);
+          ss1 += f * f;
+        }
+        vstore<T, VEC>(s_out + (long)r1 * D + i, sv1);
+      }
+    }
+    ss0 = block_reduce_sum(ss0, red);
+    ss1 = block_reduce_sum(ss1, red);
+    if (!act) continue;
+    const float inv0 = rsqrtf(ss0 / D + eps);
+    const float inv1 = rsqrtf(ss1 / D + eps);
+    if (threadIdx.x == 0) {
+      inv_rms[r0] = inv0;
+      if (has1) inv_rms[r1] = inv1;
+    }
+#pragma unroll
+    for (int half_row = 0; half_row < 2; ++half_row) {
+      if (half_row == 1 && !has1) break;
+      const int row = half_row ? r1 : r0;
+      const float inv = half_row ? inv1 : inv0;
+      const VecT<T, VEC>& sv = half_row ? sv1 : sv0;
+      VecT<T, VEC> o;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        o.data[j] = DT<T>::from_f32(DT<T>::to_f32(sv.data[j]) * inv *
+                                    DT<T>::to_f32(wv.data[j]));
+      vstore<T, VEC>(y + (long)row * D + i, o);
+    }
+  }
+}
+
 // dx = inv * w * dy - x * inv^3/D * sum(dy*w*x) ; dw_part[blk] = sum_rows dy*x*inv
 template <typename T, int VEC, int MAXSLOT>
 __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
@@ -181,6 +266,10 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy, const T* __restrict
 // vector chunk of the row, so x/dy/w stay in REGISTERS across the dot
 // reduction instead of being re-read from HBM — the generic kernel's second
 // load pass was the latency bottleneck at B=2 (19 us/call, ROADMAP r1 #5).
+// Rows are processed in PAIRS (r, r+rows/2): at the headline shape the grid
+// is ~1 wave/SIMD with nothing to hide the load->reduce->store chain, so
+// issuing both rows' loads before the first block reduce overlaps the two
+// HBM round trips inside one block.
 template <typename T, int VEC>
 __global__ void rmsnorm_bwd_1p_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                                       const T* __restrict__ w, const float* __restrict__ inv_rms,
@@ -201,43 +290,67 @@ __global__ void rmsnorm_bwd_1p_kernel(const T* __restrict__ dy, const T* __restr
   VecT<T, VEC> wv;
   if (act) wv = vload<T, VEC>(w + i);
 
-  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
-    const float inv = inv_rms[row];
-    VecT<T, VEC> xv, dyv;
-    float dot = 0.f;
+  const int half = (rows + 1) >> 1;
+  for (int p = blockIdx.x; p < half; p += gridDim.x) {
+    const int r0 = p, r1 = p + half;
+    const bool has1 = r1 < rows;
+    const float inv0 = inv_rms[r0];
+    const float inv1 = has1 ? inv_rms[r1] : 0.f;
+    VecT<T, VEC> xv0, dyv0, xv1, dyv1;
+    float dot0 = 0.f, dot1 = 0.f;
     if (act) {
-      xv = vload<T, VEC>(x + (long)row * D + i);
-      dyv = vload<T, VEC>(dy + (long)row * D + i);
+      xv0 = vload<T, VEC>(x + (long)r0 * D + i);
+      dyv0 = vload<T, VEC>(dy + (long)r0 * D + i);
+      if (has1) {
+        xv1 = vload<T, VEC>(x + (long)r1 * D + i);
+        dyv1 = vload<T, VEC>(dy + (long)r1 * D + i);
+      }
 #pragma unroll
       for (int j = 0; j < VEC; ++j)
-        dot += DT<T>::to_f32(dyv.data[j]) * DT<T>::to_f32(wv.data[j]) *
-               DT<T>::to_f32(xv.data[j]);
-    }
-    dot = block_reduce_sum(dot, red);
-    if (!act) continue;
-    const float c = inv * inv * inv / D * dot;
-    VecT<T, VEC> o, od, ev;
-    if (dextra) ev = vload<T, VEC>(dextra + (long)row * D + i);
+        dot0 += DT<T>::to_f32(dyv0.data[j]) * DT<T>::to_f32(wv.data[j]) *
+                DT<T>::to_f32(xv0.data[j]);
+      if (has1) {
 #pragma unroll
-    for (int j = 0; j < VEC; ++j) {
-      const float xf = DT<T>::to_f32(xv.data[j]);
-      const float dyf = DT<T>::to_f32(dyv.data[j]);
-      const float wf = DT<T>::to_f32(wv.data[j]);
-      const float extra = dextra ? DT<T>::to_f32(ev.data[j]) : 0.f;
-      const float dxf = inv * wf * dyf - xf * c + extra;
-      o.data[j] = DT<T>::from_f32(dxf);
-      if (ddelta) {
-        float mf = 1.f;
-        if (p_drop > 0.f) {
-          const long long idx = (long long)row * D + i + j;
-          mf = (hash_u01(seed, idx) >= p_drop) ? keep_inv : 0.f;
-        }
-        od.data[j] = DT<T>::from_f32(dxf * mf);
+        for (int j = 0; j < VEC; ++j)
+          dot1 += DT<T>::to_f32(dyv1.data[j]) * DT<T>::to_f32(wv.data[j]) *
+                  DT<T>::to_f32(xv1.data[j]);
       }
-      dwacc[j] += dyf * xf * inv;
     }
-    vstore<T, VEC>(dx + (long)row * D + i, o);
-    if (ddelta) vstore<T, VEC>(ddelta + (long)row * D + i, od);
+    dot0 = block_reduce_sum(dot0, red);
+    dot1 = block_reduce_sum(dot1, red);
+    if (!act) continue;
+#pragma unroll
+    for (int half_row = 0; half_row < 2; ++half_row) {
+      if (half_row == 1 && !has1) break;
+      const int row = half_row ? r1 : r0;
+      const float inv = half_row ? inv1 : inv0;
+      const float dot = half_row ? dot1 : dot0;
+      const VecT<T, VEC>& xv = half_row ? xv1 : xv0;
+      const VecT<T, VEC>& dyv = half_row ? dyv1 : dyv0;
+      const float c = inv * inv * inv / D * dot;
+      VecT<T, VEC> o, od, ev;
+      if (dextra) ev = vload<T, VEC>(dextra + (long)row * D + i);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        const float xf = DT<T>::to_f32(xv.data[j]);
+        const float dyf = DT<T>::to_f32(dyv.data[j]);
+        const float wf = DT<T>::to_f32(wv.data[j]);
+        const float extra = dextra ? DT<T>::to_f32(ev.data[j]) : 0.f;
+        const float dxf = inv * wf * dyf - xf * c + extra;
+        o.data[j] = DT<T>::from_f32(dxf);
+        if (ddelta) {
+          float mf = 1.f;
+          if (p_drop > 0.f) {
+            const long long idx = (long long)row * D + i + j;
+            mf = (hash_u01(seed, idx) >= p_drop) ? keep_inv : 0.f;
+          }
+          od.data[j] = DT<T>::from_f32(dxf * mf);
+        }
+        dwacc[j] += dyf * xf * inv;
+      }
+      vstore<T, VEC>(dx + (long)row * D + i, o);
+      if (ddelta) vstore<T, VEC>(ddelta + (long)row * D + i, od);
+    }
   }
   if (act) {
     VecT<float, VEC> o;
@@ -299,7 +412,7 @@ void launch_bwd(const torch::Tensor& dy, const torch::Tensor& x, const torch::Te
   // big-batch rows want more blocks; small batches keep the partial slab
   // (and its colsum) tiny.  The one-pass kernel is latency-bound at small
   // batch, so give it 2 blocks/CU.
-  const int grid = onepass ? std::min(rows, 512)
+  const int grid = onepass ? std::min((rows + 1) / 2, 512)
                            : (rows <= 2048 ? std::min(rows, 256) : 1024);
   auto part = torch::empty({grid, D}, dw.options().dtype(torch::kFloat));
   if (onepass) {
@@ -388,13 +501,26 @@ std::vector<torch::Tensor> add_rmsnorm_fwd(torch::Tensor x, torch::Tensor r,
   auto stream = at::hip::getCurrentHIPStream().stream();
   const int rows = x.size(0), D = x.size(1);
 #define ACASE(T, VEC)                                                                  \
-  hipLaunchKernelGGL((add_rmsnorm_fwd_kernel<T, VEC>),                                 \
-                     dim3(std::min(rows, 2048)), dim3((D / VEC >= 256) ? 256 : 128),   \
-                     0, stream, (const T*)x.data_ptr(), (const T*)r.data_ptr(),        \
-                     (T*)s_out.data_ptr(), (T*)y.data_ptr(), (const T*)w.data_ptr(),   \
-                     inv.data_ptr<float>(), rows, D, (float)eps, (float)p_drop,        \
-                     (seed.has_value() && seed->defined())                             \
-                         ? seed->data_ptr<long>() : nullptr, salt)
+  do {                                                                                 \
+    const long* sp = (seed.has_value() && seed->defined())                             \
+                         ? seed->data_ptr<long>() : nullptr;                           \
+    if (D <= 256 * VEC) {                                                              \
+      const int t1 = std::min(256, ((D + VEC - 1) / VEC + 63) / 64 * 64);              \
+      hipLaunchKernelGGL((add_rmsnorm_fwd_1p_kernel<T, VEC>),                          \
+                         dim3(std::min((rows + 1) / 2, 1024)), dim3(t1), 0, stream,    \
+                         (const T*)x.data_ptr(), (const T*)r.data_ptr(),               \
+                         (T*)s_out.data_ptr(), (T*)y.data_ptr(),                       \
+                         (const T*)w.data_ptr(), inv.data_ptr<float>(), rows, D,       \
+                         (float)eps, (float)p_drop, sp, salt);                         \
+    } else {                                                                           \
+      hipLaunchKernelGGL((add_rmsnorm_fwd_kernel<T, VEC>),                             \
+                         dim3(std::min(rows, 2048)), dim3((D / VEC >= 256) ? 256 : 128),\
+                         0, stream, (const T*)x.data_ptr(), (const T*)r.data_ptr(),    \
+                         (T*)s_out.data_ptr(), (T*)y.data_ptr(), (const T*)w.data_ptr(),\
+                         inv.data_ptr<float>(), rows, D, (float)eps, (float)p_drop,    \
+                         sp, salt);                                                    \
+    }                                                                                  \
+  } while (0)
   if (x.scalar_type() == torch::kBFloat16) ACASE(__hip_bfloat16, 8);
   else if (x.scalar_type() == torch::kHalf) ACASE(__half, 8);
   else if (x.scalar_type() == torch::kFloat) ACASE(float, 4);
