@@ -140,7 +140,12 @@ __global__ void seg_ce_bwd_kernel(const T* __restrict__ logits,
   // slower (3.1 ms vs 2.6 ms) than this packed form.  (A channel-pair
   // variant with 2x the loads in flight was also slower — 3.2 ms — for the
   // same register-pressure reason.)
-  float w00[16], w01[16], w10[16], w11[16], lse16[16];
+  // bilinear weights PACKED as half2 pairs: they are [0,1] fractions, so
+  // fp16 rounding (~5e-4) is far below the bf16 logits' own quantization —
+  // and it frees 32 VGPRs on a kernel that is occupancy-limited (see the
+  // seg_ce entries in ROADMAP.md negative results).
+  __half2 wA[16], wB[16];
+  float lse16[16];
   int pk16[16];
 #pragma unroll
   for (int px = 0; px < 16; ++px) {
@@ -151,7 +156,8 @@ __global__ void seg_ce_bwd_kernel(const T* __restrict__ logits,
     const bool ok = inb && lab != ignore_index && lab >= 0 && lab < C;
     lse16[px] = ok ? lse[pidx] : 0.f;
     const Tap t = tap_at(inb ? y : 0, inb ? x : 0, Hi, Wi, Ho, Wo);
-    w00[px] = t.w00; w01[px] = t.w01; w10[px] = t.w10; w11[px] = t.w11;
+    wA[px] = __floats2half2_rn(t.w00, t.w01);
+    wB[px] = __floats2half2_rn(t.w10, t.w11);
     // window-relative tap cells; y1/x1 may EQUAL y0/x0 at the image edge
     // (clamped taps) — the packed slots below then route both weights to
     // one cell
@@ -198,12 +204,13 @@ __global__ void seg_ce_bwd_kernel(const T* __restrict__ logits,
         const float v01 = by ? (bx ? c9[5] : c9[4]) : (bx ? c9[2] : c9[1]);
         const float v10 = by ? (bx ? c9[7] : c9[6]) : (bx ? c9[4] : c9[3]);
         const float v11 = by ? (bx ? c9[8] : c9[7]) : (bx ? c9[5] : c9[4]);
-        const float z = w00[px] * v00 + w01[px] * v01 +
-                        w10[px] * v10 + w11[px] * v11;
+        const float2 wa = __half22float2(wA[px]);
+        const float2 wb = __half22float2(wB[px]);
+        const float z = wa.x * v00 + wa.y * v01 + wb.x * v10 + wb.y * v11;
         const float g = (__expf(z - lse16[px]) -
                          (c == ((pk >> 20) & 255) ? 1.f : 0.f)) * gs;
-        const float g00 = w00[px] * g, g01 = w01[px] * g;
-        const float g10 = w10[px] * g, g11 = w11[px] * g;
+        const float g00 = wa.x * g, g01 = wa.y * g;
+        const float g10 = wb.x * g, g11 = wb.y * g;
         if ((pk >> 19) & 1) {
           // interior (no edge-clamped taps, the common case): by/bx have a
           // PROVABLE [0,1] range from the &1, so comparisons against the
