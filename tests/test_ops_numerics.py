@@ -50,7 +50,7 @@ def test_cross_entropy_ref_matches_torch():
 # --------------------------------------------------------------------------- #
 @gpu
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16, torch.float16])
-@pytest.mark.parametrize("shape", [(128, 768), (1024, 768), (33, 512), (64, 2048)])
+@pytest.mark.parametrize("shape", [(128, 768), (1024, 768), (33, 512), (64, 2048), (64, 4096)])
 def test_rmsnorm_fwd_bwd(dtype, shape):
     torch.manual_seed(0)
     tol = 1e-5 if dtype == torch.float32 else 2e-2
@@ -70,12 +70,13 @@ def test_rmsnorm_fwd_bwd(dtype, shape):
 
 @gpu
 @pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
-def test_add_rms_norm_fwd_bwd(dtype):
+@pytest.mark.parametrize("D", [768, 4096])  # 4096 covers the generic (non-1p) kernels
+def test_add_rms_norm_fwd_bwd(dtype, D):
     torch.manual_seed(7)
     tol = 1e-5 if dtype == torch.float32 else 2e-2
-    x = torch.randn(256, 768, device="cuda", dtype=dtype, requires_grad=True)
-    r = torch.randn(256, 768, device="cuda", dtype=dtype, requires_grad=True)
-    w = torch.randn(768, device="cuda", dtype=dtype, requires_grad=True)
+    x = torch.randn(256, D, device="cuda", dtype=dtype, requires_grad=True)
+    r = torch.randn(256, D, device="cuda", dtype=dtype, requires_grad=True)
+    w = torch.randn(D, device="cuda", dtype=dtype, requires_grad=True)
     s, y = ops.add_rms_norm(x, r, w, 1e-6)
     xr = x.detach().float().requires_grad_(True)
     rr = r.detach().float().requires_grad_(True)
