@@ -45,6 +45,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q, torch::
                                     double p_drop, c10::optional<torch::Tensor> seed,
                                     c10::optional<torch::Tensor> packed);
 torch::Tensor seg_postprocess(torch::Tensor logits, long Ho, long Wo);
+torch::Tensor upsample_bilinear_fwd(torch::Tensor x, long Ho, long Wo);
+torch::Tensor upsample_bilinear_bwd(torch::Tensor dout, long Hi, long Wi);
 torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b, long splitk);
 torch::Tensor dwconv3x3_fwd(torch::Tensor x, torch::Tensor w,
                             c10::optional<torch::Tensor> bias, bool flip);
@@ -100,6 +102,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("kpad"), py::arg("causal"), py::arg("need_dbias"),
         py::arg("p_drop"), py::arg("seed"), py::arg("packed") = c10::nullopt);
   m.def("seg_postprocess", &seg_postprocess);
+  m.def("upsample_bilinear_fwd", &upsample_bilinear_fwd);
+  m.def("upsample_bilinear_bwd", &upsample_bilinear_bwd);
   m.def("gemm_nt", &gemm_nt, py::arg("a"), py::arg("b"), py::arg("splitk") = 0);
   m.def("dwconv3x3_fwd", &dwconv3x3_fwd, py::arg("x"), py::arg("w"),
         py::arg("bias") = c10::nullopt, py::arg("flip") = false);

@@ -229,8 +229,9 @@ class SegformerDecodeHead(nn.Module):
             x = proj(feat.flatten(2).transpose(1, 2))
             x = x.transpose(1, 2).reshape(b, -1, h, w)
             if x.shape[2:] != size:
-                x = F.interpolate(x.float(), size=size, mode="bilinear",
-                                  align_corners=False).to(x.dtype)
+                # integer-scale bilinear on the gather-backward HIP kernel
+                # (upsample.hip); torch fp32 fallback otherwise
+                x = ops.upsample_bilinear(x, size)
             ups.append(x)
         x = self.linear_fuse(torch.cat(ups[::-1], dim=1))
         x = self.batch_norm(x.float()).to(x.dtype)

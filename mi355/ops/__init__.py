@@ -284,6 +284,36 @@ def dwconv3x3(x: torch.Tensor, w: torch.Tensor, bias=None) -> torch.Tensor:
     return torch.nn.functional.conv2d(x, w, bias, padding=1, groups=x.shape[1])
 
 
+class _UpsampleBilinear(torch.autograd.Function):
+    """Bilinear upsample (align_corners=False) with a GATHER backward
+    (upsample.hip) — replaces the decode head's F.interpolate whose torch
+    NHWC backward scatters with atomics (515 us/stage at the reference
+    config)."""
+
+    @staticmethod
+    def forward(ctx, x, Ho, Wo):
+        ctx.in_hw = (x.shape[2], x.shape[3])
+        return _load_ext().upsample_bilinear_fwd(x, Ho, Wo)
+
+    @staticmethod
+    def backward(ctx, dy):
+        hi, wi = ctx.in_hw
+        return _load_ext().upsample_bilinear_bwd(dy.contiguous(), hi, wi), None, None
+
+
+def upsample_bilinear(x: torch.Tensor, size) -> torch.Tensor:
+    """align_corners=False bilinear resize to ``size``; HIP path for the
+    integer-scale upsamples the SegFormer decode head produces, torch (in
+    fp32, matching HF semantics) otherwise."""
+    Ho, Wo = int(size[0]), int(size[1])
+    hi, wi = x.shape[2], x.shape[3]
+    if _use_hip(x) and Ho % hi == 0 and Wo % wi == 0 and Ho >= hi and Wo >= wi:
+        return _UpsampleBilinear.apply(x, Ho, Wo)
+    return torch.nn.functional.interpolate(
+        x.float(), size=(Ho, Wo), mode="bilinear", align_corners=False
+    ).to(x.dtype)
+
+
 class _PatchEmbedS1(torch.autograd.Function):
     """SegFormer stage-1 overlapping patch-embed conv (7x7 s4 p3, 3->32) on
     the hand-written tiled kernel (K13).  dX is not produced — stage 1

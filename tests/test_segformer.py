@@ -220,3 +220,30 @@ def test_seg_ce_fused_matches_torch(shape):
     gerr = ((got_grad.float() - l32.grad).abs().max() /
             (l32.grad.abs().max() + 1e-9)).item()
     assert gerr < 2e-2, gerr
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape,size", [((2, 256, 64, 64), (128, 128)),
+                                        ((2, 64, 32, 32), (128, 128)),
+                                        ((1, 16, 16, 16), (128, 128)),
+                                        ((2, 8, 16, 24), (64, 96))])
+def test_upsample_bilinear_matches_torch(shape, size):
+    """Gather-backward bilinear upsample (K16) vs torch fp32 interpolate."""
+    import torch.nn.functional as F
+    from mi355 import ops
+
+    torch.manual_seed(3)
+    x = torch.randn(*shape, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = ops.upsample_bilinear(x, size)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    x32 = x.detach().float().requires_grad_(True)
+    y32 = F.interpolate(x32, size=size, mode="bilinear", align_corners=False)
+    y32.backward(gy.float())
+
+    def relerr(a, ref):
+        return (a.float() - ref).norm() / ref.norm().clamp(min=1e-6)
+
+    assert relerr(y, y32) < 2e-2
+    assert relerr(x.grad, x32.grad) < 2e-2
