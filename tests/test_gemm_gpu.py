@@ -146,3 +146,41 @@ def test_gemm_nt_skinny_matches_fp32_ref(shape, monkeypatch):
     got = ops.gemm_nt(a, b)
     ref = a.float() @ b.float().t()
     assert _relerr(got, ref) < 3e-2
+
+
+# ---------------------------------------------------------------------- #
+# CPU-checkable dispatch-policy invariants (no GPU needed)                #
+# ---------------------------------------------------------------------- #
+def test_gemm_policy_dims_and_env(monkeypatch):
+    from mi355.ops.gemm_table import _WINS, gemm_policy
+
+    monkeypatch.delenv("MI355_GEMM", raising=False)
+    # every tabled win satisfies the kernel's dimension constraints and
+    # dispatches to the HIP kernel in auto mode
+    for op, rows, cols, kc in _WINS:
+        assert rows % 128 == 0 and cols % 128 == 0 and kc % 64 == 0, (op, rows, cols, kc)
+        assert gemm_policy(op, rows, cols, kc)
+    # a shape NOT in the table stays on the library in auto mode
+    assert not gemm_policy("nt", 1024, 768, 768)
+    # dims the kernels cannot run are never dispatched, even with =hip
+    monkeypatch.setenv("MI355_GEMM", "hip")
+    assert not gemm_policy("nt", 100, 768, 768)
+    monkeypatch.setenv("MI355_GEMM", "blaslt")
+    for op, rows, cols, kc in _WINS:
+        assert not gemm_policy(op, rows, cols, kc)
+
+
+def test_upsample_bilinear_cpu_fallback_matches_interpolate():
+    import torch.nn.functional as F
+
+    torch.manual_seed(0)
+    x = torch.randn(2, 3, 16, 24)
+    # CPU (and non-integer-scale) inputs take the torch fp32 path verbatim
+    y = ops.upsample_bilinear(x, (64, 96))
+    ref = F.interpolate(x.float(), size=(64, 96), mode="bilinear",
+                        align_corners=False).to(x.dtype)
+    assert torch.equal(y, ref)
+    y2 = ops.upsample_bilinear(x, (40, 50))  # non-integer scale
+    ref2 = F.interpolate(x.float(), size=(40, 50), mode="bilinear",
+                         align_corners=False).to(x.dtype)
+    assert torch.equal(y2, ref2)
