@@ -1,5 +1,7 @@
-from mi355.serve import (Deployment, PredictorDeployment, get_deployment,
-                         json_request, pandas_read_json, run, shutdown)
+from mi355.serve import (Deployment, PredictorDeployment, deployment,
+                         get_deployment, json_request, pandas_read_json,
+                         run, shutdown)
 
 __all__ = ["run", "shutdown", "Deployment", "PredictorDeployment",
-           "pandas_read_json", "json_request", "get_deployment"]
+           "deployment", "pandas_read_json", "json_request",
+           "get_deployment"]

@@ -83,3 +83,18 @@ def test_reference_import_forms():
     from ray.tune.schedulers.async_hyperband import ASHAScheduler  # noqa: F401
     from ray.tune.tuner import TuneConfig  # noqa: F401
     from ray.util.actor_pool import ActorPool  # noqa: F401
+
+
+def test_raylet_objectref_and_serve_deployment_resolve():
+    """Overview_of_Ray.ipynb cell-34 shows futures as ray._raylet.ObjectRef;
+    Introduction_to_Ray_AI_Runtime.ipynb uses @serve.deployment."""
+    import ray
+    import ray._raylet
+    from ray import serve
+
+    ref = None
+    assert ray._raylet.ObjectRef is not None
+    assert callable(serve.deployment)
+    # the runtime's actual future type IS the advertised one
+    from mi355.runtime.core import ObjectRef
+    assert ray._raylet.ObjectRef is ObjectRef
