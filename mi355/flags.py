@@ -61,6 +61,11 @@ FLAGS = [
     Flag("MI355_SPILL_THRESHOLD_MB", "<unset>",
          "Objects larger than this spill straight to disk; unset = only on "
          "a full /dev/shm."),
+    Flag("MI355_SHM_BUDGET_MB", "<unset>",
+         "Per-store owned-shm footprint cap: beyond it (or on a full "
+         "/dev/shm) the least-recently-read sealed segments are moved to "
+         "the spill dir and unlinked; readers fall through to the spill "
+         "file."),
     Flag("MI355_GEMM", "auto",
          "Projection-GEMM dispatch: auto = per-shape table from the "
          "graph-captured A/B (profiles/r02_gemm_ab.md), hip = always the "

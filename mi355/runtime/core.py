@@ -123,6 +123,18 @@ class ObjectStore:
         thr = os.environ.get("MI355_SPILL_THRESHOLD_MB")
         self._spill_threshold = int(float(thr) * 1024 * 1024) if thr else None
         self._spilled: list[str] = []
+        # LRU eviction of SEALED owned segments (r2 ROADMAP item delivered
+        # early): when the owned-shm footprint exceeds MI355_SHM_BUDGET_MB,
+        # or a put hits a full /dev/shm, the least-recently-read owned
+        # segments are copied to the spill dir and unlinked.  Readers in
+        # other processes keep working: live mmaps survive the unlink
+        # (POSIX) and NEW readers fall through to the spill file, which
+        # get()/ready() already consult after a shm miss.
+        bud = os.environ.get("MI355_SHM_BUDGET_MB")
+        self._shm_budget = int(float(bud) * 1024 * 1024) if bud else None
+        from collections import OrderedDict
+        self._owned_sizes: "OrderedDict[str, int]" = OrderedDict()
+        self._shm_bytes = 0
 
     # -- disk spill ------------------------------------------------------ #
     def _spill_path(self, name: str) -> str:
@@ -143,12 +155,44 @@ class ObjectStore:
         except FileNotFoundError:
             return None
 
+    def _evict_lru(self) -> bool:
+        """Move the least-recently-read owned sealed segment to disk and
+        unlink its shm. Returns False when nothing evictable remains."""
+        while self._owned_sizes:
+            name, sz = self._owned_sizes.popitem(last=False)
+            self._shm_bytes -= sz
+            try:
+                buf = self._native.attach(name)
+            except Exception:
+                buf = None
+            if buf is None:
+                continue  # unsealed or already gone — not evictable
+            try:
+                self._spill_put(bytes(buf), name)
+            except OSError:
+                return False  # disk full too: stop evicting
+            try:
+                self._native.unlink(name)
+                self._native.close(name)
+            except Exception:
+                pass
+            if name in self._owned:
+                self._owned.remove(name)
+            return True
+        return False
+
     # -- native-core fast paths ----------------------------------------- #
     def _nput(self, data: bytes, name: str):
         mv = self._native.create(name, len(data))
         mv[:len(data)] = data
         self._native.seal(name, len(data))
         self._owned.append(name)
+        self._owned_sizes[name] = len(data)
+        self._shm_bytes += len(data)
+        if self._shm_budget is not None:
+            while self._shm_bytes > self._shm_budget and len(self._owned_sizes) > 1:
+                if not self._evict_lru():
+                    break
 
     def _nget(self, name):
         return self._native.attach(name)  # None until sealed
@@ -164,8 +208,18 @@ class ObjectStore:
             try:
                 self._nput(data, name)
             except (OSError, MemoryError, RuntimeError):
-                # /dev/shm full: spill to disk instead of failing the put
-                self._spill_put(data, name)
+                # /dev/shm full: evict LRU owned segments to disk and retry;
+                # only if nothing is evictable does the NEW object spill
+                stored = False
+                while self._evict_lru():
+                    try:
+                        self._nput(data, name)
+                        stored = True
+                        break
+                    except (OSError, MemoryError, RuntimeError):
+                        continue
+                if not stored:
+                    self._spill_put(data, name)
             return ref
         seg = shared_memory.SharedMemory(name=name, create=True, size=len(data) + 8)
         _untrack(name)
@@ -186,6 +240,8 @@ class ObjectStore:
 
     def get(self, ref: ObjectRef, timeout: float | None = None):
         name = _shm_name(ref)
+        if name in self._owned_sizes:  # LRU recency bump for owned objects
+            self._owned_sizes.move_to_end(name)
         deadline = None if timeout is None else time.monotonic() + timeout
         delay = 0.0005
         while True:
