@@ -164,15 +164,27 @@ class Dataset:
                 files.extend(sorted(_glob.glob(os.path.join(p, "*.parquet"))))
             else:
                 files.append(p)
+        from .. import runtime
+
+        if runtime.is_initialized() and files:
+            # STREAMING read (r2 ROADMAP item): one task per file, read on a
+            # worker, block stays in the object store as a ref — the driver
+            # never holds the file bytes; downstream map_batches chains
+            # pipeline the refs straight through.
+            task = runtime.remote(_read_parquet_file)
+            return Dataset([task.remote(f) for f in files])
         return Dataset([pq.read_table(f) for f in files])
 
     @staticmethod
     def read_csv(paths) -> "Dataset":
-        from pyarrow import csv as pacsv
-
         if isinstance(paths, str):
             paths = [paths]
-        return Dataset([pacsv.read_csv(p) for p in paths])
+        from .. import runtime
+
+        if runtime.is_initialized() and paths:
+            task = runtime.remote(_read_csv_file)
+            return Dataset([task.remote(p) for p in paths])
+        return Dataset([_read_csv_file(p) for p in paths])
 
     # ------------------------------------------------------------------ #
     # inspection                                                          #
@@ -527,3 +539,14 @@ def _apply_batch_fn(fn_bytes, tbl, batch_format, batch_size=None):
                  for t in _slice_batches(tbl, batch_size)]
         return pa.concat_tables(parts, promote_options="default")
     return _to_table(fn(_from_table(tbl, batch_format)))
+
+
+def _read_parquet_file(path):
+    """Worker-side file read for the streaming read_parquet path."""
+    return pq.read_table(path)
+
+
+def _read_csv_file(path):
+    from pyarrow import csv as pacsv
+
+    return pacsv.read_csv(path)
