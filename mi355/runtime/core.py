@@ -149,10 +149,16 @@ class ObjectStore:
         self._spilled.append(name)
 
     def _spill_get(self, name: str):
+        # mmap readback: deserialized numpy/arrow buffers alias the mapping
+        # (the returned memoryview keeps the mmap alive), so a spilled-block
+        # get never copies the file through driver heap
+        import mmap as _mmap
+
         try:
             with open(self._spill_path(name), "rb") as f:
-                return f.read()
-        except FileNotFoundError:
+                mm = _mmap.mmap(f.fileno(), 0, access=_mmap.ACCESS_READ)
+            return memoryview(mm)
+        except (FileNotFoundError, ValueError):
             return None
 
     def _evict_lru(self) -> bool:
@@ -251,7 +257,7 @@ class ObjectStore:
                     return _deserialize(memoryview(buf))
                 data = self._spill_get(name)
                 if data is not None:
-                    return _deserialize(memoryview(data))
+                    return _deserialize(data)
             else:
                 try:
                     seg = self._attach(name)
@@ -259,7 +265,9 @@ class ObjectStore:
                     if n > 0:
                         return _deserialize(memoryview(seg.buf)[8:8 + n])
                 except FileNotFoundError:
-                    pass
+                    data = self._spill_get(name)
+                    if data is not None:
+                        return _deserialize(data)
             if deadline is not None and time.monotonic() > deadline:
                 raise TimeoutError(f"object {ref} not available after {timeout}s")
             time.sleep(delay)
