@@ -1,4 +1,5 @@
-from .dataset import ActorPoolStrategy, Dataset, GroupedData
+from .dataset import (ActorPoolStrategy, Dataset, DatasetPipeline,
+                      GroupedData)
 from .preprocessor import (BatchMapper, Chain, LabelEncoder, MinMaxScaler,
                            PowerTransformer, Preprocessor, StandardScaler)
 
@@ -11,7 +12,7 @@ read_parquet = Dataset.read_parquet
 read_csv = Dataset.read_csv
 
 __all__ = [
-    "Dataset", "GroupedData", "ActorPoolStrategy", "Preprocessor",
+    "Dataset", "DatasetPipeline", "GroupedData", "ActorPoolStrategy", "Preprocessor",
     "BatchMapper", "MinMaxScaler", "StandardScaler", "PowerTransformer",
     "LabelEncoder", "Chain", "from_items", "from_pandas", "from_numpy",
     "from_arrow", "from_huggingface", "read_parquet", "read_csv",

@@ -108,6 +108,12 @@ class Dataset:
     def materialize(self) -> "Dataset":
         return Dataset(list(self._tables()))
 
+    def window(self, *, blocks_per_window: int = 10) -> "DatasetPipeline":
+        k = max(1, int(blocks_per_window))
+        wins = [Dataset(self._blocks[i:i + k])
+                for i in range(0, max(1, len(self._blocks)), k)]
+        return DatasetPipeline(wins)
+
     def is_fully_materialized(self) -> bool:
         return all(isinstance(b, pa.Table) for b in self._blocks)
 
@@ -482,6 +488,64 @@ class Dataset:
         for o in others:
             blocks.extend(o._blocks)
         return Dataset(blocks)
+
+
+class DatasetPipeline:
+    """Windowed streaming execution (Ray's ``Dataset.window`` /
+    DatasetPipeline): transforms are recorded as stages and applied
+    window-by-window at iteration time, so peak memory is one window's
+    blocks (which are themselves object-store refs under the runtime).
+    The reference notebooks do not use pipelines; this completes the C6
+    dataset surface."""
+
+    def __init__(self, windows: list["Dataset"], stages=None, epochs: int = 1):
+        self._windows = windows
+        self._stages = list(stages or [])
+        self._epochs = epochs
+
+    def num_windows(self) -> int:
+        return len(self._windows) * self._epochs
+
+    def map_batches(self, fn, **kwargs) -> "DatasetPipeline":
+        return DatasetPipeline(self._windows,
+                               self._stages + [("map_batches", fn, kwargs)],
+                               self._epochs)
+
+    def repeat(self, times: int = 1) -> "DatasetPipeline":
+        return DatasetPipeline(self._windows, self._stages,
+                               self._epochs * max(1, times))
+
+    def _transformed(self):
+        for _ in range(self._epochs):
+            for w in self._windows:
+                ds = w
+                for kind, fn, kwargs in self._stages:
+                    ds = getattr(ds, kind)(fn, **kwargs)
+                yield ds
+
+    def iter_batches(self, **kwargs):
+        for ds in self._transformed():
+            yield from ds.iter_batches(**kwargs)
+
+    def iter_rows(self):
+        for ds in self._transformed():
+            for tbl in ds._tables():
+                yield from tbl.to_pylist()
+
+    def count(self) -> int:
+        return sum(ds.count() for ds in self._transformed())
+
+    def take(self, n: int = 20):
+        out = []
+        for row in self.iter_rows():
+            out.append(row)
+            if len(out) >= n:
+                break
+        return out
+
+    def __repr__(self):
+        return (f"DatasetPipeline(windows={len(self._windows)}, "
+                f"stages={len(self._stages)}, epochs={self._epochs})")
 
 
 class GroupedData:

@@ -319,3 +319,27 @@ def test_read_parquet_streams_through_object_store(tmp_path):
     ds2 = Dataset.read_parquet(str(tmp_path))
     assert ds2.is_fully_materialized()
     assert ds2.count() == 30
+
+
+def test_window_pipeline():
+    """Dataset.window() -> DatasetPipeline: stages apply per window at
+    iteration time; repeat() multiplies epochs."""
+    import pandas as pd
+
+    from mi355.data import Dataset
+
+    ds = Dataset.from_items([{"x": i} for i in range(40)], parallelism=8)
+    pipe = ds.window(blocks_per_window=2)
+    assert pipe.num_windows() == 4
+    out = pipe.map_batches(lambda df: pd.DataFrame({"x": df["x"] * 10}),
+                           batch_format="pandas", batch_size=None)
+    assert out.count() == 40
+    assert sorted(r["x"] for r in out.iter_rows()) == [i * 10 for i in range(40)]
+    got = []
+    for b in out.iter_batches(batch_size=7, batch_format="pandas"):
+        assert len(b) <= 7
+        got.extend(b["x"].tolist())
+    assert sorted(got) == [i * 10 for i in range(40)]
+    assert out.take(3) == [{"x": 0}, {"x": 10}, {"x": 20}]
+    rep = out.repeat(3)
+    assert rep.num_windows() == 12 and rep.count() == 120
