@@ -88,9 +88,16 @@ def _runtime_state():
                                          and a._proc.is_alive())})
         except Exception:
             continue
+    store = {}
+    st = getattr(rt, "store", None)
+    if st is not None:
+        store = {"owned_segments": len(getattr(st, "_owned_sizes", ())),
+                 "owned_bytes": getattr(st, "_shm_bytes", 0),
+                 "spilled": len(getattr(st, "_spilled", ())),
+                 "budget_bytes": getattr(st, "_shm_budget", None)}
     return {"initialized": True, "num_gpus": rt.num_gpus,
             "workers": workers, "actors": actors,
-            "tasks_done": len(rt._events)}
+            "tasks_done": len(rt._events), "object_store": store}
 
 
 def _runs_state(base=None):
@@ -147,9 +154,13 @@ def _render_html(st, t0):
                      for w in rt["workers"]])
         at = _table(["actor", "pid", "alive"],
                     [(a["name"], a["pid"], a["alive"]) for a in rt["actors"]])
+        os_ = rt.get("object_store") or {}
+        store_line = (f" &middot; store: {os_.get('owned_segments', 0)} segs "
+                      f"/ {os_.get('owned_bytes', 0) / 1e6:.1f} MB shm, "
+                      f"{os_.get('spilled', 0)} spilled") if os_ else ""
         runtime = (f"<p>{len(rt['workers'])} workers &middot; "
                    f"{rt['num_gpus']} GPU(s) &middot; "
-                   f"{rt['tasks_done']} tasks completed</p>" + wt +
+                   f"{rt['tasks_done']} tasks completed{store_line}</p>" + wt +
                    "<h2>Actors</h2>" + at)
     else:
         runtime = "<p>runtime not initialized in this process</p>"
