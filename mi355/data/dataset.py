@@ -523,8 +523,33 @@ class DatasetPipeline:
                     ds = getattr(ds, kind)(fn, **kwargs)
                 yield ds
 
-    def iter_batches(self, **kwargs):
-        for ds in self._transformed():
+    def iter_batches(self, prefetch_windows: int = 1, **kwargs):
+        """Iterate batches window-by-window.  With ``prefetch_windows`` > 0
+        a background thread runs the NEXT window's stages (remote-task
+        submission under the runtime) while the current window is being
+        consumed — pipeline-parallel windows."""
+        if prefetch_windows <= 0:
+            for ds in self._transformed():
+                yield from ds.iter_batches(**kwargs)
+            return
+        import queue as _q
+        import threading
+
+        q: "_q.Queue" = _q.Queue(maxsize=max(1, prefetch_windows))
+        sentinel = object()
+
+        def producer():
+            try:
+                for ds in self._transformed():
+                    q.put(ds)
+            finally:
+                q.put(sentinel)
+
+        threading.Thread(target=producer, daemon=True).start()
+        while True:
+            ds = q.get()
+            if ds is sentinel:
+                return
             yield from ds.iter_batches(**kwargs)
 
     def iter_rows(self):

@@ -343,3 +343,23 @@ def test_window_pipeline():
     assert out.take(3) == [{"x": 0}, {"x": 10}, {"x": 20}]
     rep = out.repeat(3)
     assert rep.num_windows() == 12 and rep.count() == 120
+
+
+def test_window_pipeline_prefetch_under_runtime():
+    """Prefetched windows still yield every batch exactly once and in
+    window order, with the stages running as remote tasks."""
+    import pandas as pd
+
+    import mi355
+    from mi355.data import Dataset
+
+    mi355.init(num_workers=2)
+    ds = Dataset.from_items([{"x": i} for i in range(60)], parallelism=6)
+    pipe = ds.window(blocks_per_window=2).map_batches(
+        lambda df: pd.DataFrame({"x": df["x"] + 100}),
+        batch_format="pandas", batch_size=None)
+    got = []
+    for b in pipe.iter_batches(prefetch_windows=2, batch_size=10,
+                               batch_format="pandas"):
+        got.extend(b["x"].tolist())
+    assert got == [i + 100 for i in range(60)]
