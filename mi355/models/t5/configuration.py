@@ -67,6 +67,15 @@ T5_PRESETS = {
     "flan-t5-large": dict(
         d_model=1024, d_kv=64, d_ff=2816, num_layers=24, num_decoder_layers=24, num_heads=16
     ),
+    # scale-up presets (google/flan-t5-xl, -xxl): same T5-v1.1 topology;
+    # xxl is ~11B params = ~44 GB of bf16 master+optimizer state, well
+    # inside one MI355X's 288 GB HBM3E for single-GPU fine-tuning
+    "flan-t5-xl": dict(
+        d_model=2048, d_kv=64, d_ff=5120, num_layers=24, num_decoder_layers=24, num_heads=32
+    ),
+    "flan-t5-xxl": dict(
+        d_model=4096, d_kv=64, d_ff=10240, num_layers=24, num_decoder_layers=24, num_heads=64
+    ),
 }
 
 

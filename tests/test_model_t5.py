@@ -126,3 +126,20 @@ def test_model_gpu_generate_matches_cpu():
     # argmax over near-equal logits can differ on a few tokens; demand majority
     agree = (cpu_gen == gpu_gen.cpu()).float().mean().item()
     assert agree >= 0.6, f"CPU/GPU generation diverged badly: {agree}"
+
+
+def test_xl_presets_build_on_meta():
+    """flan-t5-xl / -xxl presets construct with the published parameter
+    counts (meta device — no memory; the xxl fine-tune state fits one
+    MI355X's 288 GB HBM)."""
+    import torch
+
+    from mi355.models.t5.configuration import t5_config
+    from mi355.models.t5.modeling import T5ForConditionalGeneration
+
+    expect = {"flan-t5-xl": (2.7e9, 3.1e9), "flan-t5-xxl": (10.5e9, 11.5e9)}
+    for name, (lo, hi) in expect.items():
+        with torch.device("meta"):
+            m = T5ForConditionalGeneration(t5_config(name))
+        n = sum(p.numel() for p in m.parameters())
+        assert lo < n < hi, (name, n)
