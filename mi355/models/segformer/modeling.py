@@ -52,6 +52,19 @@ class SegformerConfig:
         known = set(cls.__dataclass_fields__)
         return cls(**{k: v for k, v in d.items() if k in known})
 
+    @classmethod
+    def from_preset(cls, name: str, **overrides) -> "SegformerConfig":
+        """MiT-b0..b5 encoder presets (nvidia/mit-bN; the reference uses b0
+        — Scaling_model_training.ipynb cell-16 — the rest complete the
+        published family for scale-up)."""
+        key = name.split("/")[-1].replace("mit-", "").replace("segformer-", "")
+        if key not in SEGFORMER_PRESETS:
+            raise KeyError(f"unknown SegFormer preset {name!r}; "
+                           f"have {sorted(SEGFORMER_PRESETS)}")
+        kw = dict(SEGFORMER_PRESETS[key])
+        kw.update(overrides)
+        return cls(**kw)
+
     def save_pretrained(self, path):
         os.makedirs(path, exist_ok=True)
         with open(os.path.join(path, "config.json"), "w") as f:
@@ -237,6 +250,23 @@ class SegformerDecodeHead(nn.Module):
         x = self.batch_norm(x.float()).to(x.dtype)
         x = F.relu(x)
         return self.classifier(self.dropout(x))
+
+
+SEGFORMER_PRESETS = {
+    # depths, hidden sizes, decoder width per the published MiT family
+    "b0": dict(depths=[2, 2, 2, 2], hidden_sizes=[32, 64, 160, 256],
+               decoder_hidden_size=256),
+    "b1": dict(depths=[2, 2, 2, 2], hidden_sizes=[64, 128, 320, 512],
+               decoder_hidden_size=256),
+    "b2": dict(depths=[3, 4, 6, 3], hidden_sizes=[64, 128, 320, 512],
+               decoder_hidden_size=768),
+    "b3": dict(depths=[3, 4, 18, 3], hidden_sizes=[64, 128, 320, 512],
+               decoder_hidden_size=768),
+    "b4": dict(depths=[3, 8, 27, 3], hidden_sizes=[64, 128, 320, 512],
+               decoder_hidden_size=768),
+    "b5": dict(depths=[3, 6, 40, 3], hidden_sizes=[64, 128, 320, 512],
+               decoder_hidden_size=768),
+}
 
 
 @dataclass

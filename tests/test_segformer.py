@@ -247,3 +247,27 @@ def test_upsample_bilinear_matches_torch(shape, size):
 
     assert relerr(y, y32) < 2e-2
     assert relerr(x.grad, x32.grad) < 2e-2
+
+
+def test_mit_family_presets():
+    """MiT-b0..b5 presets match the published parameter counts (meta
+    device), and a non-b0 config runs forward+loss on CPU (the custom-
+    kernel guards fall back cleanly off its shapes)."""
+    from mi355.models.segformer.modeling import (
+        SegformerConfig, SegformerForSemanticSegmentation)
+
+    expect = {"b0": (3.4e6, 4.2e6), "b1": (13e6, 15e6), "b2": (24e6, 30e6),
+              "b3": (44e6, 50e6), "b4": (60e6, 67e6), "b5": (81e6, 90e6)}
+    for name, (lo, hi) in expect.items():
+        with torch.device("meta"):
+            m = SegformerForSemanticSegmentation(SegformerConfig.from_preset(name))
+        n = sum(p.numel() for p in m.parameters())
+        assert lo < n < hi, (name, n)
+
+    m = SegformerForSemanticSegmentation(
+        SegformerConfig.from_preset("b1", num_labels=19))
+    x = torch.randn(1, 3, 64, 64)
+    labels = torch.randint(0, 19, (1, 64, 64))
+    out = m(pixel_values=x, labels=labels)
+    assert out.loss.isfinite()
+    out.loss.backward()
